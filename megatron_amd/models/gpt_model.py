@@ -114,6 +114,10 @@ class GPTModel(MegatronModule):
             )
         if self._language_model_key in state_dict:
             state_dict = state_dict[self._language_model_key]
+        elif any(k.startswith("language_model.") for k in state_dict):
+            # plain nn.Module state dict (flat keys)
+            torch.nn.Module.load_state_dict(self, state_dict, strict=strict)
+            return
         self.language_model.load_state_dict(state_dict, strict=strict)
 
 

@@ -1,0 +1,79 @@
+// Python bindings for the MI355X ops extension.
+
+#include <torch/extension.h>
+
+#include <vector>
+
+// norms.hip
+std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor weight,
+                                       double eps);
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                       torch::Tensor weight,
+                                       torch::Tensor inv);
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor weight,
+                                         torch::Tensor bias, double eps);
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor weight,
+                                         torch::Tensor mean,
+                                         torch::Tensor inv);
+
+// softmax.hip
+torch::Tensor scaled_masked_softmax_fwd(torch::Tensor x,
+                                        c10::optional<torch::Tensor> mask,
+                                        double scale, bool causal);
+torch::Tensor scaled_masked_softmax_bwd(torch::Tensor dy, torch::Tensor y,
+                                        double scale);
+
+// elementwise.hip
+torch::Tensor glu_fwd(torch::Tensor x, int64_t mode);
+torch::Tensor glu_bwd(torch::Tensor dy, torch::Tensor x, int64_t mode);
+torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor cosT,
+                       torch::Tensor sinT);
+torch::Tensor rope_bwd(torch::Tensor dy, torch::Tensor cosT,
+                       torch::Tensor sinT);
+std::vector<torch::Tensor> bias_dropout_add_fwd(
+    torch::Tensor x, c10::optional<torch::Tensor> bias,
+    torch::Tensor residual, double p, int64_t seed, int64_t offset);
+torch::Tensor dropout_bwd(torch::Tensor dy, torch::Tensor mask, double p);
+
+// adam.hip
+void fused_adam(std::vector<torch::Tensor> params,
+                std::vector<torch::Tensor> grads,
+                std::vector<torch::Tensor> exp_avgs,
+                std::vector<torch::Tensor> exp_avg_sqs, double lr,
+                double beta1, double beta2, double eps, double wd,
+                int64_t step, int64_t adam_w_mode);
+
+// wgrad.hip
+void wgrad_gemm_accum_fp32(torch::Tensor input, torch::Tensor grad_output,
+                           torch::Tensor main_grad);
+
+// flash_attn.hip
+std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
+                                          torch::Tensor v, bool causal,
+                                          double softmax_scale,
+                                          int64_t window_size);
+std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
+                                          torch::Tensor k, torch::Tensor v,
+                                          torch::Tensor out, torch::Tensor lse,
+                                          bool causal, double softmax_scale,
+                                          int64_t window_size);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("layernorm_fwd", &layernorm_fwd);
+  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("scaled_masked_softmax_fwd", &scaled_masked_softmax_fwd);
+  m.def("scaled_masked_softmax_bwd", &scaled_masked_softmax_bwd);
+  m.def("glu_fwd", &glu_fwd);
+  m.def("glu_bwd", &glu_bwd);
+  m.def("rope_fwd", &rope_fwd);
+  m.def("rope_bwd", &rope_bwd);
+  m.def("bias_dropout_add_fwd", &bias_dropout_add_fwd);
+  m.def("dropout_bwd", &dropout_bwd);
+  m.def("fused_adam", &fused_adam);
+  m.def("wgrad_gemm_accum_fp32", &wgrad_gemm_accum_fp32);
+  m.def("flash_attn_fwd", &flash_attn_fwd);
+  m.def("flash_attn_bwd", &flash_attn_bwd);
+}

@@ -56,7 +56,8 @@ def load(required: Optional[bool] = None) -> Optional[ModuleType]:
             )
         return None
     try:
-        spec = importlib.util.spec_from_file_location("megatron_amd_ops_C", so_path)
+        # module name must match TORCH_EXTENSION_NAME (PyInit__C)
+        spec = importlib.util.spec_from_file_location("_C", so_path)
         mod = importlib.util.module_from_spec(spec)
         spec.loader.exec_module(mod)
         _EXT = mod

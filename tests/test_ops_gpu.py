@@ -1,0 +1,285 @@
+"""GPU numerics: every HIP kernel vs a plain PyTorch fp32 reference.
+
+Mirrors the reference's fused_kernels/tests/test_fused_kernels.py strategy:
+load the extension, compare fused outputs against unfused torch math.
+"""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def _ext():
+    from megatron_amd.ops import ext
+
+    return ext.load(required=True)
+
+
+def rel_err(a, b):
+    a = a.float()
+    b = b.float()
+    denom = b.abs().max().clamp(min=1e-6)
+    return ((a - b).abs().max() / denom).item()
+
+
+@pytest.fixture(autouse=True)
+def _seed():
+    torch.manual_seed(1234)
+    torch.cuda.manual_seed(1234)
+
+
+class TestRMSNorm:
+    @pytest.mark.parametrize("H", [4096, 1000, 8192])
+    @pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+    def test_fwd_bwd(self, H, dtype):
+        ext = _ext()
+        rows = 512
+        x = torch.randn(rows, H, device="cuda", dtype=dtype)
+        w = (1 + 0.1 * torch.randn(H, device="cuda", dtype=dtype))
+        eps = 1e-5
+
+        y, inv = ext.rmsnorm_fwd(x, w, eps)
+        xf = x.float()
+        ref_inv = torch.rsqrt(xf.pow(2).mean(-1) + eps)
+        ref_y = xf * ref_inv.unsqueeze(-1) * w.float()
+        assert rel_err(y, ref_y) < (5e-2 if dtype == torch.bfloat16 else 1e-5)
+        assert rel_err(inv, ref_inv) < 1e-4
+
+        dy = torch.randn_like(x)
+        dx, dw = ext.rmsnorm_bwd(dy, x, w, inv)
+        xr = xf.clone().requires_grad_(True)
+        wr = w.float().clone().requires_grad_(True)
+        yr = xr * torch.rsqrt(xr.pow(2).mean(-1, keepdim=True) + eps) * wr
+        yr.backward(dy.float())
+        tol = 5e-2 if dtype == torch.bfloat16 else 1e-4
+        assert rel_err(dx, xr.grad) < tol
+        assert rel_err(dw, wr.grad) < tol
+
+
+class TestLayerNorm:
+    @pytest.mark.parametrize("H", [4096, 1024])
+    def test_fwd_bwd(self, H):
+        ext = _ext()
+        rows = 256
+        dtype = torch.bfloat16
+        x = torch.randn(rows, H, device="cuda", dtype=dtype)
+        w = 1 + 0.1 * torch.randn(H, device="cuda", dtype=dtype)
+        b = 0.1 * torch.randn(H, device="cuda", dtype=dtype)
+        eps = 1e-5
+        y, mean, inv = ext.layernorm_fwd(x, w, b, eps)
+        ref = torch.nn.functional.layer_norm(
+            x.float(), (H,), w.float(), b.float(), eps
+        )
+        assert rel_err(y, ref) < 5e-2
+
+        dy = torch.randn_like(x)
+        dx, dw, db = ext.layernorm_bwd(dy, x, w, mean, inv)
+        xr = x.float().clone().requires_grad_(True)
+        wr = w.float().clone().requires_grad_(True)
+        br = b.float().clone().requires_grad_(True)
+        yr = torch.nn.functional.layer_norm(xr, (H,), wr, br, eps)
+        yr.backward(dy.float())
+        assert rel_err(dx, xr.grad) < 5e-2
+        assert rel_err(dw, wr.grad) < 5e-2
+        assert rel_err(db, br.grad) < 5e-2
+
+
+class TestSoftmax:
+    @pytest.mark.parametrize("causal", [True, False])
+    @pytest.mark.parametrize("sk", [128, 1024, 2048])
+    def test_fwd_bwd(self, causal, sk):
+        ext = _ext()
+        b, np_, sq = 2, 4, sk
+        x = torch.randn(b, np_, sq, sk, device="cuda", dtype=torch.bfloat16)
+        scale = 0.5
+        y = ext.scaled_masked_softmax_fwd(x, None, scale, causal)
+        xf = x.float() * scale
+        if causal:
+            cm = torch.ones(sq, sk, dtype=torch.bool, device="cuda").triu(1)
+            xf = xf.masked_fill(cm, float("-inf"))
+        ref = torch.softmax(xf, dim=-1)
+        assert rel_err(y, ref) < 5e-2
+
+        dy = torch.randn_like(x)
+        dx = ext.scaled_masked_softmax_bwd(dy, y, scale)
+        xr = (x.float() * 1.0).requires_grad_(True)
+        xs = xr * scale
+        if causal:
+            xs = xs.masked_fill(cm, float("-inf"))
+        yr = torch.softmax(xs, dim=-1)
+        yr.backward(dy.float())
+        assert rel_err(dx, xr.grad) < 5e-2
+
+    def test_bool_mask(self):
+        ext = _ext()
+        b, np_, sq, sk = 2, 2, 64, 64
+        x = torch.randn(b, np_, sq, sk, device="cuda", dtype=torch.bfloat16)
+        mask = torch.rand(b, 1, sq, sk, device="cuda") > 0.8
+        y = ext.scaled_masked_softmax_fwd(x, mask, 1.0, False)
+        ref = torch.softmax(
+            x.float().masked_fill(mask, float("-inf")), dim=-1
+        ).nan_to_num(0.0)
+        assert rel_err(y, ref) < 5e-2
+
+
+class TestGLU:
+    @pytest.mark.parametrize("mode,fn", [
+        (0, lambda t: t),
+        (1, torch.nn.functional.gelu),
+        (2, torch.relu),
+        (3, torch.nn.functional.silu),
+    ])
+    def test_fwd_bwd(self, mode, fn):
+        ext = _ext()
+        rows, F = 512, 1024
+        x = torch.randn(rows, 2 * F, device="cuda", dtype=torch.bfloat16)
+        y = ext.glu_fwd(x, mode)
+        x1, x2 = x.float().chunk(2, dim=-1)
+        ref = x1 * fn(x2)
+        assert rel_err(y, ref) < 5e-2
+
+        dy = torch.randn_like(y)
+        dx = ext.glu_bwd(dy, x, mode)
+        xr = x.float().clone().requires_grad_(True)
+        a, c = xr.chunk(2, dim=-1)
+        yr = a * fn(c)
+        yr.backward(dy.float())
+        assert rel_err(dx, xr.grad) < 5e-2
+
+
+class TestRoPE:
+    def test_fwd_bwd(self):
+        from megatron_amd.models.rope import precompute_freqs
+
+        ext = _ext()
+        s, b, n, h = 128, 2, 4, 128
+        x = torch.randn(s, b, n, h, device="cuda", dtype=torch.bfloat16)
+        cos, sin = precompute_freqs(h, s, device="cuda")
+        y = ext.rope_fwd(x, cos, sin)
+
+        xf = x.float()
+        x1, x2 = xf[..., 0::2], xf[..., 1::2]
+        c = cos.view(s, 1, 1, -1)
+        sn = sin.view(s, 1, 1, -1)
+        ref = torch.stack([x1 * c - x2 * sn, x2 * c + x1 * sn], -1).flatten(-2)
+        assert rel_err(y, ref) < 5e-2
+
+        dy = torch.randn_like(x)
+        dx = ext.rope_bwd(dy, cos, sin)
+        g1, g2 = dy.float()[..., 0::2], dy.float()[..., 1::2]
+        refdx = torch.stack([g1 * c + g2 * sn, g2 * c - g1 * sn], -1).flatten(-2)
+        assert rel_err(dx, refdx) < 5e-2
+
+
+class TestDropoutAdd:
+    def test_statistics_and_bwd(self):
+        ext = _ext()
+        x = torch.randn(4096, 1024, device="cuda", dtype=torch.bfloat16)
+        res = torch.randn_like(x)
+        bias = torch.randn(1024, device="cuda", dtype=torch.bfloat16)
+        p = 0.3
+        y, mask = ext.bias_dropout_add_fwd(x, bias, res, p, 12345, 0)
+        keep_frac = mask.float().mean().item()
+        assert abs(keep_frac - (1 - p)) < 0.01
+        ref = (
+            (x.float() + bias.float()) * mask.float() / (1 - p) + res.float()
+        )
+        assert rel_err(y, ref) < 5e-2
+
+        dy = torch.randn_like(x)
+        dx = ext.dropout_bwd(dy, mask, p)
+        refdx = dy.float() * mask.float() / (1 - p)
+        assert rel_err(dx, refdx) < 5e-2
+
+
+class TestAdam:
+    def test_matches_torch_adamw(self):
+        ext = _ext()
+        n = 12345
+        p = torch.randn(n, device="cuda")
+        g = torch.randn(n, device="cuda")
+        m = torch.zeros(n, device="cuda")
+        v = torch.zeros(n, device="cuda")
+        p_ref = p.clone().requires_grad_(True)
+        opt = torch.optim.AdamW([p_ref], lr=1e-3, betas=(0.9, 0.999),
+                                eps=1e-8, weight_decay=0.01)
+        for step in range(1, 4):
+            ext.fused_adam([p], [g], [m], [v], 1e-3, 0.9, 0.999, 1e-8, 0.01,
+                           step, 1)
+            p_ref.grad = g.clone()
+            opt.step()
+        assert rel_err(p, p_ref.detach()) < 1e-4
+
+
+class TestWgrad:
+    def test_accumulate(self):
+        ext = _ext()
+        K, in_dim, out_dim = 512, 256, 128
+        inp = torch.randn(K, in_dim, device="cuda", dtype=torch.bfloat16)
+        gout = torch.randn(K, out_dim, device="cuda", dtype=torch.bfloat16)
+        main_grad = torch.randn(out_dim, in_dim, device="cuda")
+        expected = main_grad + gout.float().t() @ inp.float()
+        ext.wgrad_gemm_accum_fp32(inp, gout, main_grad)
+        assert rel_err(main_grad, expected) < 1e-2
+
+
+class TestFlashAttention:
+    @pytest.mark.parametrize("sq,sk", [(256, 256), (128, 384), (333, 333)])
+    @pytest.mark.parametrize("causal", [True, False])
+    @pytest.mark.parametrize("gqa", [1, 4])
+    @pytest.mark.parametrize("d", [128, 64])
+    def test_fwd_vs_fp32_reference(self, sq, sk, causal, gqa, d):
+        from megatron_amd.ops.functional import _sdpa_reference
+
+        ext = _ext()
+        b, n = 2, 4
+        nkv = n // gqa
+        q = torch.randn(b, sq, n, d, device="cuda", dtype=torch.bfloat16)
+        k = torch.randn(b, sk, nkv, d, device="cuda", dtype=torch.bfloat16)
+        v = torch.randn(b, sk, nkv, d, device="cuda", dtype=torch.bfloat16)
+        scale = 1.0 / math.sqrt(d)
+        out, lse = ext.flash_attn_fwd(q, k, v, causal, scale, -1)
+        ref, ref_lse = _sdpa_reference(q, k, v, causal, scale, None)
+        assert rel_err(out, ref) < 6e-2
+        assert (lse - ref_lse).abs().max().item() < 1e-2
+
+    def test_sliding_window(self):
+        from megatron_amd.ops.functional import _sdpa_reference
+
+        ext = _ext()
+        b, s, n, d = 1, 512, 2, 128
+        w = 128
+        q = torch.randn(b, s, n, d, device="cuda", dtype=torch.bfloat16)
+        k = torch.randn(b, s, n, d, device="cuda", dtype=torch.bfloat16)
+        v = torch.randn(b, s, n, d, device="cuda", dtype=torch.bfloat16)
+        scale = 1.0 / math.sqrt(d)
+        out, lse = ext.flash_attn_fwd(q, k, v, True, scale, w)
+        ref, _ = _sdpa_reference(q, k, v, True, scale, w)
+        assert rel_err(out, ref) < 6e-2
+
+    @pytest.mark.parametrize("causal", [True, False])
+    @pytest.mark.parametrize("gqa", [1, 4])
+    def test_bwd_vs_fp32_reference(self, causal, gqa):
+        from megatron_amd.ops.functional import (
+            _sdpa_reference, _sdpa_reference_bwd,
+        )
+
+        ext = _ext()
+        b, s, n, d = 2, 256, 4, 128
+        nkv = n // gqa
+        q = torch.randn(b, s, n, d, device="cuda", dtype=torch.bfloat16)
+        k = torch.randn(b, s, nkv, d, device="cuda", dtype=torch.bfloat16)
+        v = torch.randn(b, s, nkv, d, device="cuda", dtype=torch.bfloat16)
+        scale = 1.0 / math.sqrt(d)
+        out, lse = ext.flash_attn_fwd(q, k, v, causal, scale, -1)
+        dout = torch.randn_like(out)
+        dq, dk, dv = ext.flash_attn_bwd(dout, q, k, v, out, lse, causal,
+                                        scale, -1)
+        rdq, rdk, rdv = _sdpa_reference_bwd(dout, q, k, v, out, lse, causal,
+                                            scale, None)
+        assert rel_err(dv, rdv) < 8e-2
+        assert rel_err(dk, rdk) < 8e-2
+        assert rel_err(dq, rdq) < 8e-2
