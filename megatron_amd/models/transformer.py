@@ -306,8 +306,9 @@ class ParallelAttention(MegatronModule):
 
         # rotary embedding
         if self.rope_cos is not None:
-            cos = self.rope_cos.to(query.device)
-            sin = self.rope_sin.to(query.device)
+            # tables stay fp32 regardless of module dtype conversion
+            cos = self.rope_cos.to(device=query.device, dtype=torch.float32)
+            sin = self.rope_sin.to(device=query.device, dtype=torch.float32)
             if inference_params is not None:
                 offset = inference_params.sequence_len_offset
                 if position_ids is None:

@@ -217,6 +217,27 @@ class DistributedOptimizer(MixedPrecisionOptimizer):
             self.model_gbuf_ranges
         )
 
+        # Contiguous param buffers mirroring the grad buffer layout; params
+        # are re-pointed into them FIRST so every shard view built below
+        # references the final storage (the DP all-gather then lands directly
+        # in param storage).
+        self.param_buffers = []
+        for model_index, model in enumerate(self.models):
+            current_param_buffers = {}
+            for dtype, grad_buffer in model._grad_buffers.items():
+                buf = torch.empty(
+                    grad_buffer.numel_padded, dtype=params_dtype,
+                    device=grad_buffer.data.device,
+                )
+                for param, (start, end) in (
+                    model._grad_buffer_param_index_map[dtype].items()
+                ):
+                    view = buf[start:end].view(param.data.shape)
+                    view.detach().copy_(param.data)
+                    param.data = view
+                current_param_buffers[dtype] = buf
+            self.param_buffers.append(current_param_buffers)
+
         self.opt_group_ranges = self.build_optimizer_group_ranges(
             self.optimizer.param_groups, self.model_gbuf_ranges
         )
@@ -230,54 +251,9 @@ class DistributedOptimizer(MixedPrecisionOptimizer):
             self.opt_group_ranges,
         )
 
-        # contiguous param buffers mirroring grad buffer layout; params are
-        # re-pointed into them so the DP all-gather lands in param storage
-        self.param_buffers = []
-        for model_index, model in enumerate(self.models):
-            current_param_buffers = {}
-            for dtype, grad_buffer in model._grad_buffers.items():
-                param_dtype = params_dtype
-                buf = torch.empty(
-                    grad_buffer.numel_padded, dtype=param_dtype,
-                    device=grad_buffer.data.device,
-                )
-                for param, (start, end) in (
-                    model._grad_buffer_param_index_map[dtype].items()
-                ):
-                    view = buf[start:end].view(param.data.shape)
-                    view.detach().copy_(param.data)
-                    param.data = view
-                current_param_buffers[dtype] = buf
-            self.param_buffers.append(current_param_buffers)
-
-        # shard model params must re-reference the new storage
-        self._rebuild_shard_model_param_views()
-
         self.optimizer.param_groups = [
             g["orig_group"] for g in self.opt_group_ranges
         ]
-
-    def _rebuild_shard_model_param_views(self):
-        for group_index, group_range in enumerate(self.opt_group_ranges):
-            f16_i = 0
-            fp32_i = 0
-            for param in group_range["params"]:
-                model_index, dtype = self.model_param_gbuf_map[param]
-                param_range = self.model_gbuf_ranges[model_index][dtype][
-                    "param_map"
-                ][param]["param"]
-                if param.data.dtype in (torch.half, torch.bfloat16):
-                    self.shard_float16_groups[group_index][f16_i] = (
-                        param.data.detach().view(-1)[
-                            param_range.start : param_range.end
-                        ]
-                    )
-                    f16_i += 1
-                elif param.data.dtype == torch.float:
-                    self.shard_fp32_groups[group_index][fp32_i] = (
-                        param.data.view(-1)[param_range.start : param_range.end]
-                    )
-                    fp32_i += 1
 
     def get_model_param_range_map(self, param):
         model_index, dtype = self.model_param_gbuf_map[param]

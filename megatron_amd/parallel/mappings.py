@@ -23,6 +23,10 @@ from .utils import split_tensor_along_last_dim
 def _reduce(input_: torch.Tensor) -> torch.Tensor:
     if ps.get_tensor_model_parallel_world_size() == 1:
         return input_
+    # collectives need dense storage; autograd can hand us stride-0
+    # expanded grads (e.g. from .sum().backward())
+    if not input_.is_contiguous():
+        input_ = input_.contiguous()
     dist.all_reduce(input_, group=ps.get_tensor_model_parallel_group())
     return input_
 
