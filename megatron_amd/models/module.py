@@ -84,6 +84,8 @@ def conversion_helper(val, conversion):
 
 def fp32_to_float16(val, float16_convertor):
     def half_conversion(val):
+        if val is None or not torch.is_tensor(val):
+            return val
         val_typecheck = val
         if isinstance(val_typecheck, (Parameter, Variable)):
             val_typecheck = val.data
