@@ -4,24 +4,30 @@
 // (transformer.py:9,369,528-552): causal + sliding-window attention with
 // native GQA/MQA head broadcast (no K/V expansion), forward + backward.
 //
-// Design (v1, correctness-first MFMA structure — tuned variants iterate on
-// this skeleton):
+// v2 design notes:
 //  - mfma_f32_16x16x32_bf16 tiles; 256-thread blocks = 4 waves;
 //    BLOCK_M = BLOCK_N = 64; head dims 64 / 128.
-//  - Operand layouts chosen so Q/K A- and B-fragments load as contiguous
-//    16-byte chunks straight from HBM ([S][D] rows match the fragment's
-//    8-element k-runs); only P (QK^T output -> PV input) round-trips
-//    through a padded LDS strip, and V/dO/Q B-fragments that need the
-//    d-major orientation are staged in LDS tiles.
+//  - Q/K A- and B-fragments whose 8-element k-runs lie along the head dim
+//    load as contiguous 16-byte chunks straight from HBM ([S][D] layout).
+//  - Fragments that need the d-major orientation (V in PV, dO/Q in dK/dV,
+//    K in dQ) are staged in LDS "tr images" of [row/4][D/16][4][16] tiles
+//    and read with the gfx950 hardware transpose read ds_read_b64_tr_b16
+//    (lane-linear per-16-lane-group addresses; mapping verified empirically
+//    with ops/csrc/probe_tr.hip on MI355X: with per-lane address
+//    base + (lane&15)*8B, lane l elem j reads lds16[base + (l&15) + 16*j]).
+//    This replaces v1's 8 scalar ds_read_u16 per fragment with 2 tr reads
+//    (the dominant non-MFMA cost in v1's inner loop).
+//  - s_setprio(1) brackets the MFMA clusters (technique T5: the 4-wave
+//    schedule has load/compute role diversity for the scheduler to
+//    arbitrate).
 //  - online softmax in fp32 registers; cross-lane row reductions are
 //    __shfl_xor within 16-lane groups (wave64).
 //  - backward is split k-parallel (dK/dV, GQA-group loop => register
 //    accumulation, no atomics) and q-parallel (dQ) plus a small
-//    delta = rowsum(dO*O) preprocess kernel, using the stored
-//    logsumexp as in FlashAttention-2.
+//    delta = rowsum(dO*O) preprocess kernel, using the stored logsumexp.
 //
-// Fragment maps (gfx950 mfma_f32_16x16x32_bf16, verified by the
-// tests/gpu parity suite):
+// Fragment maps (gfx950 mfma_f32_16x16x32_bf16, verified by the GPU parity
+// suite):
 //   A[i][k]: lane l holds A[l%16][(l/16)*8 + j], j = 0..7
 //   B[k][j]: lane l holds B[(l/16)*8 + j][l%16]
 //   C[r][c]: lane l, reg r holds C[(l/16)*4 + r][l%16]
@@ -37,14 +43,14 @@ namespace {
 
 using frag_b16 = __attribute__((ext_vector_type(8))) short;
 using frag_f32 = __attribute__((ext_vector_type(4))) float;
+using v4s = __attribute__((ext_vector_type(4))) short;
 
 constexpr int kBlockM = 64;
 constexpr int kBlockN = 64;
 constexpr int kThreads = 256;
-constexpr int kStrip = 72;  // padded LDS strip stride (bf16 elems) per 64 cols
+constexpr int kStrip = 72;  // padded LDS strip stride (bf16 elems), P/dS tiles
 
 __device__ __forceinline__ frag_b16 load_frag_global(const __hip_bfloat16* p) {
-  // 8 contiguous bf16 = 16 B
   uint4 u = *reinterpret_cast<const uint4*>(p);
   union {
     uint4 u;
@@ -70,6 +76,64 @@ __device__ __forceinline__ float group16_sum(float v) {
   return v;
 }
 
+// --- tr-image helpers ------------------------------------------------------
+// image layout: rows x D stored as [row/4][D/16] tiles of [4][16] row-major.
+
+template <int D>
+__device__ __forceinline__ int tr_off(int row, int col) {
+  return ((row >> 2) * (D / 16) + (col >> 4)) * 64 + (row & 3) * 16 +
+         (col & 15);
+}
+
+// B-fragment B[row0 + (l>>4)*8 + jj][dtile*16 + (l&15)], jj = 0..7.
+template <int D>
+__device__ __forceinline__ frag_b16 tr_bfrag(const __hip_bfloat16* img,
+                                             int row0, int dtile, int lane) {
+  const int c = lane & 15;
+  const int ktile = (row0 >> 2) + (lane >> 4) * 2;
+  const int b1 = (ktile * (D / 16) + dtile) * 64;
+  const int b2 = ((ktile + 1) * (D / 16) + dtile) * 64;
+  const __attribute__((address_space(3))) v4s* p1 =
+      (const __attribute__((address_space(3))) v4s*)&img[b1] + c;
+  const __attribute__((address_space(3))) v4s* p2 =
+      (const __attribute__((address_space(3))) v4s*)&img[b2] + c;
+  v4s lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+      (__attribute__((address_space(3))) v4s*)p1);
+  v4s hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+      (__attribute__((address_space(3))) v4s*)p2);
+  frag_b16 f;
+  f[0] = lo[0];
+  f[1] = lo[1];
+  f[2] = lo[2];
+  f[3] = lo[3];
+  f[4] = hi[0];
+  f[5] = hi[1];
+  f[6] = hi[2];
+  f[7] = hi[3];
+  return f;
+}
+
+// cooperative stage of a [rows x D] bf16 tile from global into a tr image
+template <int D, int ROWS>
+__device__ __forceinline__ void stage_tr_image(
+    __hip_bfloat16* img, const __hip_bfloat16* src_base, long row_stride,
+    int first_row, int max_row) {
+  const int vec_per_row = D / 8;
+  for (int idx = threadIdx.x; idx < ROWS * vec_per_row; idx += kThreads) {
+    int r = idx / vec_per_row;
+    int c8 = (idx % vec_per_row) * 8;
+    int row = first_row + r;
+    uint4 val;
+    if (row < max_row) {
+      val = *reinterpret_cast<const uint4*>(src_base + (long)row * row_stride +
+                                            c8);
+    } else {
+      val = make_uint4(0, 0, 0, 0);
+    }
+    *reinterpret_cast<uint4*>(&img[tr_off<D>(r, c8)]) = val;
+  }
+}
+
 // ---------------------------------------------------------------------------
 // forward
 
@@ -79,17 +143,16 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
     const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ out,
     float* __restrict__ lse, int B, int Sq, int Sk, int Hq, int Hkv,
     float scale, int causal, int window) {
-  constexpr int KFRAGS = D / 32;   // A/B fragment K-steps
-  constexpr int DTILES = D / 16;   // output col tiles
-  constexpr int VPAD = D + 8;
+  constexpr int KFRAGS = D / 32;
+  constexpr int DTILES = D / 16;
 
   __shared__ __hip_bfloat16 p_lds[4][16 * kStrip];
-  __shared__ __hip_bfloat16 v_lds[kBlockN * VPAD];
+  __shared__ __hip_bfloat16 v_img[kBlockN * D];
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int row_in_tile = lane & 15;    // A-frag row / C col
-  const int kgroup = lane >> 4;         // 0..3
+  const int row_in_tile = lane & 15;
+  const int kgroup = lane >> 4;
 
   const int qb = blockIdx.x;
   const int h = blockIdx.y;
@@ -100,13 +163,12 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
   const long k_base = ((long)b * Sk * Hkv + hkv) * D;
   const long v_base = k_base;
   const long o_base = q_base;
-  const int rq = Hq * D;   // row stride in q/out
-  const int rk = Hkv * D;  // row stride in k/v
+  const int rq = Hq * D;
+  const int rk = Hkv * D;
 
-  const int qrow0 = qb * kBlockM + wave * 16;  // wave's first q row
+  const int qrow0 = qb * kBlockM + wave * 16;
   const int skq = Sk - Sq;
 
-  // Q A-fragments (row = qrow0 + lane%16, k-run = kgroup*8)
   frag_b16 qf[KFRAGS];
   {
     int qrow = qrow0 + row_in_tile;
@@ -128,7 +190,6 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
     l_run[r] = 0.f;
   }
 
-  // key-block range for the whole block (waves mask individually)
   int kb_end = (Sk + kBlockN - 1) / kBlockN;
   if (causal) {
     int max_qrow = qb * kBlockM + kBlockM - 1;
@@ -145,27 +206,10 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
   for (int kb = kb_start; kb < kb_end; ++kb) {
     const int kstart = kb * kBlockN;
 
-    // stage V tile cooperatively: 64 rows x D cols
-    {
-      const int vec_per_row = D / 8;  // uint4 count per row
-      for (int idx = threadIdx.x; idx < kBlockN * vec_per_row;
-           idx += kThreads) {
-        int krow = idx / vec_per_row;
-        int c8 = (idx % vec_per_row) * 8;
-        int key = kstart + krow;
-        uint4 val;
-        if (key < Sk) {
-          val = *reinterpret_cast<const uint4*>(v + v_base + (long)key * rk +
-                                                c8);
-        } else {
-          val = make_uint4(0, 0, 0, 0);
-        }
-        *reinterpret_cast<uint4*>(&v_lds[krow * VPAD + c8]) = val;
-      }
-    }
+    stage_tr_image<D, kBlockN>(v_img, v + v_base, rk, kstart, Sk);
     __syncthreads();
 
-    // S = Q K^T for this wave's 16 rows x 64 cols
+    // S = Q K^T
     frag_f32 st[4];
 #pragma unroll
     for (int t = 0; t < 4; ++t) st[t] = frag_f32{0.f, 0.f, 0.f, 0.f};
@@ -173,16 +217,21 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
     for (int t = 0; t < 4; ++t) {
       int key = kstart + t * 16 + row_in_tile;
       int kr = key < Sk ? key : Sk - 1;
+      frag_b16 bf[KFRAGS];
 #pragma unroll
       for (int kk = 0; kk < KFRAGS; ++kk) {
-        frag_b16 bf = load_frag_global(k + k_base + (long)kr * rk + kk * 32 +
-                                       kgroup * 8);
-        st[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[kk], bf, st[t], 0,
-                                                        0, 0);
+        bf[kk] = load_frag_global(k + k_base + (long)kr * rk + kk * 32 +
+                                  kgroup * 8);
       }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int kk = 0; kk < KFRAGS; ++kk) {
+        st[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[kk], bf[kk],
+                                                        st[t], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_setprio(0);
     }
 
-    // mask + scale into s[t][r]; track masked-ness
     float s_val[4][4];
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
@@ -197,8 +246,6 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
       }
     }
 
-    // NOTE: C layout row = kgroup*4 + r with col = row_in_tile; the row-wise
-    // reduction is over col => over the 16 lanes of the row_in_tile group.
     float m_new[4], alpha[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
@@ -211,27 +258,24 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
     }
 
     float p_val[4][4];
-    float rowsum[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       float acc = 0.f;
 #pragma unroll
       for (int t = 0; t < 4; ++t) {
-        float pv = (s_val[t][r] < -1e29f)
-                       ? 0.f
-                       : __expf(s_val[t][r] - m_new[r]);
+        float pv = (s_val[t][r] < -1e29f) ? 0.f
+                                          : __expf(s_val[t][r] - m_new[r]);
         p_val[t][r] = pv;
         acc += pv;
       }
-      rowsum[r] = group16_sum(acc);
-      l_run[r] = l_run[r] * alpha[r] + rowsum[r];
+      float rowsum = group16_sum(acc);
+      l_run[r] = l_run[r] * alpha[r] + rowsum;
 #pragma unroll
       for (int t = 0; t < DTILES; ++t) {
         o_acc[t][r] *= alpha[r];
       }
     }
 
-    // P -> LDS strip [row 16][col 64] stride kStrip
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
 #pragma unroll
@@ -241,42 +285,38 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
         p_lds[wave][row * kStrip + col] = __float2bfloat16(p_val[t][r]);
       }
     }
-    __syncthreads();  // also covers v_lds reuse across iterations
+    __syncthreads();
 
     // O += P V
 #pragma unroll
     for (int kk2 = 0; kk2 < 2; ++kk2) {
-      // A-frag: P rows (lane%16), k-run = kk2*32 + kgroup*8
       frag_b16 pf;
       {
         const __hip_bfloat16* src =
             &p_lds[wave][row_in_tile * kStrip + kk2 * 32 + kgroup * 8];
-        uint4 u = *reinterpret_cast<const uint4*>(src);
         union {
           uint4 u;
           frag_b16 f;
         } cvt;
-        cvt.u = u;
+        cvt.u = *reinterpret_cast<const uint4*>(src);
         pf = cvt.f;
       }
+      frag_b16 vf[DTILES];
 #pragma unroll
       for (int t = 0; t < DTILES; ++t) {
-        // B-frag: V[key = kk2*32 + kgroup*8 + j][d = t*16 + lane%16]
-        frag_b16 vf;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          int key = kk2 * 32 + kgroup * 8 + j;
-          vf[j] = __bfloat16_as_short(
-              v_lds[key * VPAD + t * 16 + row_in_tile]);
-        }
-        o_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vf, o_acc[t],
-                                                           0, 0, 0);
+        vf[t] = tr_bfrag<D>(v_img, kk2 * 32, t, lane);
       }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int t = 0; t < DTILES; ++t) {
+        o_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vf[t],
+                                                           o_acc[t], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_setprio(0);
     }
     __syncthreads();
   }
 
-  // epilogue: normalize + store O (bf16) and lse (fp32)
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     int row = qrow0 + kgroup * 4 + r;
@@ -296,14 +336,14 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// backward preprocess: delta[b,h,s] = sum_d dO * O  (fp32)
+// backward preprocess: delta[b,h,s] = sum_d dO * O (fp32)
 
 template <int D>
 __global__ void fa_bwd_delta_kernel(const __hip_bfloat16* __restrict__ dout,
                                     const __hip_bfloat16* __restrict__ out,
                                     float* __restrict__ delta, int B, int Sq,
                                     int Hq) {
-  const long row = blockIdx.x;  // b*Sq*Hq rows? use (b*Hq+h)*Sq + s layout
+  const long row = blockIdx.x;
   const int s = row % Sq;
   const long bh = row / Sq;
   const int b = bh / Hq;
@@ -318,8 +358,7 @@ __global__ void fa_bwd_delta_kernel(const __hip_bfloat16* __restrict__ dout,
 }
 
 // ---------------------------------------------------------------------------
-// backward dK/dV (k-parallel; loops over q heads of the GQA group and q
-// blocks; accumulates dK/dV in registers — no atomics)
+// backward dK/dV (k-parallel)
 
 template <int D>
 __global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
@@ -331,12 +370,11 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
     float scale, int causal, int window) {
   constexpr int KFRAGS = D / 32;
   constexpr int DTILES = D / 16;
-  constexpr int TPAD = D + 8;
 
-  __shared__ __hip_bfloat16 pt_lds[4][16 * kStrip];   // P^T strips
-  __shared__ __hip_bfloat16 dst_lds[4][16 * kStrip];  // dS^T strips
-  __shared__ __hip_bfloat16 do_lds[kBlockM * TPAD];   // dO tile
-  __shared__ __hip_bfloat16 q_tile_lds[kBlockM * TPAD];
+  __shared__ __hip_bfloat16 pt_lds[4][16 * kStrip];
+  __shared__ __hip_bfloat16 dst_lds[4][16 * kStrip];
+  __shared__ __hip_bfloat16 do_img[kBlockM * D];
+  __shared__ __hip_bfloat16 q_img[kBlockM * D];
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -353,9 +391,8 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
   const int rq = Hq * D;
   const int skq = Sk - Sq;
 
-  const int key0 = kb * kBlockN + wave * 16;  // wave's first key
+  const int key0 = kb * kBlockN + wave * 16;
 
-  // K and V A-fragments for this wave's 16 keys
   frag_b16 ka[KFRAGS], va[KFRAGS];
   {
     int key = key0 + row_in_tile;
@@ -376,16 +413,14 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
     dk_acc[t] = frag_f32{0.f, 0.f, 0.f, 0.f};
   }
 
-  // q-block range for this k-block
   int qb_start = 0;
   if (causal) {
     int min_key = kb * kBlockN;
-    int min_qrow = min_key - skq;  // first q row that can attend min_key
+    int min_qrow = min_key - skq;
     if (min_qrow > 0) qb_start = min_qrow / kBlockM;
   }
   int qb_end = (Sq + kBlockM - 1) / kBlockM;
   if (window > 0) {
-    // q rows beyond key + window - 1 - skq can't see this block
     int max_key = kb * kBlockN + kBlockN - 1;
     int max_qrow = max_key + window - 1 - skq;
     qb_end = min(qb_end, max_qrow / kBlockM + 1);
@@ -399,31 +434,10 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
     for (int qb = qb_start; qb < qb_end; ++qb) {
       const int qstart = qb * kBlockM;
 
-      // stage dO and Q tiles cooperatively
-      {
-        const int vec_per_row = D / 8;
-        for (int idx = threadIdx.x; idx < kBlockM * vec_per_row;
-             idx += kThreads) {
-          int qrow = idx / vec_per_row;
-          int c8 = (idx % vec_per_row) * 8;
-          int sq_idx = qstart + qrow;
-          uint4 dval, qval;
-          if (sq_idx < Sq) {
-            dval = *reinterpret_cast<const uint4*>(dout + q_base +
-                                                   (long)sq_idx * rq + c8);
-            qval = *reinterpret_cast<const uint4*>(q + q_base +
-                                                   (long)sq_idx * rq + c8);
-          } else {
-            dval = make_uint4(0, 0, 0, 0);
-            qval = make_uint4(0, 0, 0, 0);
-          }
-          *reinterpret_cast<uint4*>(&do_lds[qrow * TPAD + c8]) = dval;
-          *reinterpret_cast<uint4*>(&q_tile_lds[qrow * TPAD + c8]) = qval;
-        }
-      }
+      stage_tr_image<D, kBlockM>(do_img, dout + q_base, rq, qstart, Sq);
+      stage_tr_image<D, kBlockM>(q_img, q + q_base, rq, qstart, Sq);
       __syncthreads();
 
-      // S^T = K Q^T : rows = keys (this wave's 16), cols = 64 q rows
       frag_f32 stt[4], dpt[4];
 #pragma unroll
       for (int t = 0; t < 4; ++t) {
@@ -434,22 +448,25 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
       for (int t = 0; t < 4; ++t) {
         int qrow = qstart + t * 16 + row_in_tile;
         int qr = qrow < Sq ? qrow : Sq - 1;
+        frag_b16 qbf[KFRAGS], dbf[KFRAGS];
 #pragma unroll
         for (int kk = 0; kk < KFRAGS; ++kk) {
-          // B-frag Q^T: Q[qrow l%16][d-run]
-          frag_b16 qbf = load_frag_global(q + q_base + (long)qr * rq +
-                                          kk * 32 + kgroup * 8);
-          stt[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ka[kk], qbf,
+          qbf[kk] = load_frag_global(q + q_base + (long)qr * rq + kk * 32 +
+                                     kgroup * 8);
+          dbf[kk] = load_frag_global(dout + q_base + (long)qr * rq + kk * 32 +
+                                     kgroup * 8);
+        }
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int kk = 0; kk < KFRAGS; ++kk) {
+          stt[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ka[kk], qbf[kk],
                                                            stt[t], 0, 0, 0);
-          // B-frag dO^T for dP^T = V dO^T
-          frag_b16 dbf = load_frag_global(dout + q_base + (long)qr * rq +
-                                          kk * 32 + kgroup * 8);
-          dpt[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(va[kk], dbf,
+          dpt[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(va[kk], dbf[kk],
                                                            dpt[t], 0, 0, 0);
         }
+        __builtin_amdgcn_s_setprio(0);
       }
 
-      // P^T = exp(S^T*scale - lse), dS^T = P^T*(dP^T - delta)*scale
 #pragma unroll
       for (int t = 0; t < 4; ++t) {
 #pragma unroll
@@ -465,7 +482,7 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
             pt = __expf(stt[t][r] * scale - l);
             dst = pt * (dpt[t][r] - delta_h[qrow]) * scale;
           }
-          int lrow = kgroup * 4 + r;    // key within wave strip
+          int lrow = kgroup * 4 + r;
           int lcol = t * 16 + row_in_tile;
           pt_lds[wave][lrow * kStrip + lcol] = __float2bfloat16(pt);
           dst_lds[wave][lrow * kStrip + lcol] = __float2bfloat16(dst);
@@ -491,30 +508,26 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
           ptf = c1.f;
           dstf = c2.f;
         }
+        frag_b16 dof[DTILES], qtf[DTILES];
 #pragma unroll
         for (int t = 0; t < DTILES; ++t) {
-          frag_b16 dof, qtf;
-#pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            int qrow = kk2 * 32 + kgroup * 8 + j;
-            dof[j] = __bfloat16_as_short(
-                do_lds[qrow * TPAD + t * 16 + row_in_tile]);
-            qtf[j] = __bfloat16_as_short(
-                q_tile_lds[qrow * TPAD + t * 16 + row_in_tile]);
-          }
-          dv_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ptf, dof,
-                                                              dv_acc[t], 0,
-                                                              0, 0);
-          dk_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dstf, qtf,
-                                                              dk_acc[t], 0,
-                                                              0, 0);
+          dof[t] = tr_bfrag<D>(do_img, kk2 * 32, t, lane);
+          qtf[t] = tr_bfrag<D>(q_img, kk2 * 32, t, lane);
         }
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int t = 0; t < DTILES; ++t) {
+          dv_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              ptf, dof[t], dv_acc[t], 0, 0, 0);
+          dk_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              dstf, qtf[t], dk_acc[t], 0, 0, 0);
+        }
+        __builtin_amdgcn_s_setprio(0);
       }
       __syncthreads();
     }
   }
 
-  // store dK/dV (C layout: row = key kgroup*4+r, col = d)
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     int key = kb * kBlockN + wave * 16 + kgroup * 4 + r;
@@ -542,10 +555,9 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dq_kernel(
     int Sq, int Sk, int Hq, int Hkv, float scale, int causal, int window) {
   constexpr int KFRAGS = D / 32;
   constexpr int DTILES = D / 16;
-  constexpr int TPAD = D + 8;
 
   __shared__ __hip_bfloat16 ds_lds[4][16 * kStrip];
-  __shared__ __hip_bfloat16 k_tile_lds[kBlockN * TPAD];
+  __shared__ __hip_bfloat16 k_img[kBlockN * D];
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -567,7 +579,6 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dq_kernel(
   const float* lse_h = lse + ((long)b * Hq + h) * Sq;
   const float* delta_h = delta + ((long)b * Hq + h) * Sq;
 
-  // Q and dO A-fragments for this wave's rows
   frag_b16 qf[KFRAGS], dof[KFRAGS];
   {
     int qrow = qrow0 + row_in_tile;
@@ -607,27 +618,9 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dq_kernel(
   for (int kb = kb_start; kb < kb_end; ++kb) {
     const int kstart = kb * kBlockN;
 
-    // stage K tile
-    {
-      const int vec_per_row = D / 8;
-      for (int idx = threadIdx.x; idx < kBlockN * vec_per_row;
-           idx += kThreads) {
-        int krow = idx / vec_per_row;
-        int c8 = (idx % vec_per_row) * 8;
-        int key = kstart + krow;
-        uint4 val;
-        if (key < Sk) {
-          val = *reinterpret_cast<const uint4*>(k + k_base + (long)key * rk +
-                                                c8);
-        } else {
-          val = make_uint4(0, 0, 0, 0);
-        }
-        *reinterpret_cast<uint4*>(&k_tile_lds[krow * TPAD + c8]) = val;
-      }
-    }
+    stage_tr_image<D, kBlockN>(k_img, k + k_base, rk, kstart, Sk);
     __syncthreads();
 
-    // S and dP tiles
     frag_f32 st[4], dp[4];
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
@@ -638,20 +631,25 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dq_kernel(
     for (int t = 0; t < 4; ++t) {
       int key = kstart + t * 16 + row_in_tile;
       int kr = key < Sk ? key : Sk - 1;
+      frag_b16 kbf[KFRAGS], vbf[KFRAGS];
 #pragma unroll
       for (int kk = 0; kk < KFRAGS; ++kk) {
-        frag_b16 kbf = load_frag_global(k + k_base + (long)kr * rk + kk * 32 +
-                                        kgroup * 8);
-        st[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[kk], kbf, st[t], 0,
-                                                        0, 0);
-        frag_b16 vbf = load_frag_global(v + k_base + (long)kr * rk + kk * 32 +
-                                        kgroup * 8);
-        dp[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dof[kk], vbf, dp[t],
-                                                        0, 0, 0);
+        kbf[kk] = load_frag_global(k + k_base + (long)kr * rk + kk * 32 +
+                                   kgroup * 8);
+        vbf[kk] = load_frag_global(v + k_base + (long)kr * rk + kk * 32 +
+                                   kgroup * 8);
       }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int kk = 0; kk < KFRAGS; ++kk) {
+        st[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[kk], kbf[kk],
+                                                        st[t], 0, 0, 0);
+        dp[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dof[kk], vbf[kk],
+                                                        dp[t], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_setprio(0);
     }
 
-    // dS = P*(dP - delta)*scale -> LDS strip
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
 #pragma unroll
@@ -687,24 +685,23 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dq_kernel(
         cvt.u = *reinterpret_cast<const uint4*>(src);
         dsf = cvt.f;
       }
+      frag_b16 ktf[DTILES];
 #pragma unroll
       for (int t = 0; t < DTILES; ++t) {
-        frag_b16 ktf;
+        ktf[t] = tr_bfrag<D>(k_img, kk2 * 32, t, lane);
+      }
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          int key = kk2 * 32 + kgroup * 8 + j;
-          ktf[j] = __bfloat16_as_short(
-              k_tile_lds[key * TPAD + t * 16 + row_in_tile]);
-        }
-        dq_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, ktf,
+      for (int t = 0; t < DTILES; ++t) {
+        dq_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, ktf[t],
                                                             dq_acc[t], 0, 0,
                                                             0);
       }
+      __builtin_amdgcn_s_setprio(0);
     }
     __syncthreads();
   }
 
-  // store dQ
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     int row = qrow0 + kgroup * 4 + r;
@@ -735,8 +732,7 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
   TORCH_CHECK(Hq % Hkv == 0);
 
   auto out = torch::empty_like(q);
-  auto lse = torch::empty({B, Hq, Sq},
-                          q.options().dtype(torch::kFloat32));
+  auto lse = torch::empty({B, Hq, Sq}, q.options().dtype(torch::kFloat32));
   dim3 grid((Sq + kBlockM - 1) / kBlockM, Hq, B);
   auto stream = c10::hip::getCurrentHIPStream();
   int win = window_size > 0 ? (int)window_size : 0;
