@@ -11,7 +11,9 @@ from .language_model import (  # noqa: F401
     parallel_lm_logits,
     scaled_init_method_normal,
 )
+from .bert_model import BertModel  # noqa: F401
 from .module import Float16Module, MegatronModule  # noqa: F401
+from .t5_model import T5Model  # noqa: F401
 from .norms import LayerNorm, RMSNorm  # noqa: F401
 from .transformer import (  # noqa: F401
     CoreAttention,

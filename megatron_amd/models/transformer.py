@@ -222,12 +222,7 @@ class ParallelAttention(MegatronModule):
             // self.num_attention_heads_kv_per_partition
         )
 
-        assert attention_type == AttnType.self_attn
-        qkv_out = cfg.kv_channels * (
-            cfg.num_attention_heads + 2 * cfg.num_attention_heads_kv
-        )
-        self.query_key_value = mpu.ColumnParallelLinear(
-            cfg.hidden_size, qkv_out,
+        linear_kwargs = dict(
             bias=cfg.use_bias, gather_output=False, init_method=init_method,
             params_dtype=cfg.params_dtype,
             use_cpu_initialization=cfg.use_cpu_initialization,
@@ -240,6 +235,24 @@ class ParallelAttention(MegatronModule):
             ),
             world_size=world_size,
         )
+        if attention_type == AttnType.self_attn:
+            qkv_out = cfg.kv_channels * (
+                cfg.num_attention_heads + 2 * cfg.num_attention_heads_kv
+            )
+            self.query_key_value = mpu.ColumnParallelLinear(
+                cfg.hidden_size, qkv_out, **linear_kwargs
+            )
+        else:
+            # cross attention (T5 decoder): separate q and fused kv
+            assert cfg.num_attention_heads == cfg.num_attention_heads_kv, (
+                "cross attention does not support GQA"
+            )
+            self.query = mpu.ColumnParallelLinear(
+                cfg.hidden_size, projection_size, **linear_kwargs
+            )
+            self.key_value = mpu.ColumnParallelLinear(
+                cfg.hidden_size, 2 * projection_size, **linear_kwargs
+            )
 
         self.use_flash_attn = cfg.use_flash_attn
         if self.use_flash_attn:
@@ -284,25 +297,39 @@ class ParallelAttention(MegatronModule):
         )
 
     def forward(self, hidden_states, attention_mask, position_ids=None,
-                inference_params=None):
+                inference_params=None, encoder_output=None):
         # hidden_states [s, b, h]
         sq, b = hidden_states.shape[0], hidden_states.shape[1]
         np_ = self.num_attention_heads_per_partition
         nkv = self.num_attention_heads_kv_per_partition
         hn = self.hidden_size_per_attention_head
 
-        mixed, _ = self.query_key_value(hidden_states)
-        if self.sequence_parallel:
-            sq = mixed.shape[0]
-        # per-group layout [nq+2, hn] within each kv group
-        ngroups = nkv
-        nq_per_group = np_ // nkv
-        mixed = mixed.view(sq, b, ngroups, (nq_per_group + 2) * hn)
-        query = mixed[..., : nq_per_group * hn].reshape(sq, b, np_, hn)
-        key = mixed[..., nq_per_group * hn : (nq_per_group + 1) * hn].reshape(
-            sq, b, nkv, hn
-        )
-        value = mixed[..., (nq_per_group + 1) * hn :].reshape(sq, b, nkv, hn)
+        if self.attention_type == AttnType.self_attn:
+            mixed, _ = self.query_key_value(hidden_states)
+            if self.sequence_parallel:
+                sq = mixed.shape[0]
+            # per-group layout [nq+2, hn] within each kv group
+            ngroups = nkv
+            nq_per_group = np_ // nkv
+            mixed = mixed.view(sq, b, ngroups, (nq_per_group + 2) * hn)
+            query = mixed[..., : nq_per_group * hn].reshape(sq, b, np_, hn)
+            key = mixed[
+                ..., nq_per_group * hn : (nq_per_group + 1) * hn
+            ].reshape(sq, b, nkv, hn)
+            value = mixed[..., (nq_per_group + 1) * hn :].reshape(
+                sq, b, nkv, hn
+            )
+        else:
+            # cross attention: q from decoder states, kv from encoder output
+            q_out, _ = self.query(hidden_states)
+            kv_out, _ = self.key_value(encoder_output)
+            if self.sequence_parallel:
+                sq = q_out.shape[0]
+            sk = kv_out.shape[0]
+            query = q_out.reshape(sq, b, np_, hn)
+            kv = kv_out.view(sk, b, np_, 2 * hn)
+            key = kv[..., :hn].contiguous()
+            value = kv[..., hn:].contiguous()
 
         # rotary embedding
         if self.rope_cos is not None:
@@ -343,8 +370,10 @@ class ParallelAttention(MegatronModule):
             key = k_cache[: start + sq]
             value = v_cache[: start + sq]
 
-        use_flash = self.use_flash_attn and (
-            inference_params is None or sq > 1
+        use_flash = (
+            self.use_flash_attn
+            and self.attention_type == AttnType.self_attn
+            and (inference_params is None or sq > 1)
         )
         if use_flash:
             if self.n_rep > 1 and not ops_f._ext.available() and not query.is_cuda:
@@ -392,11 +421,20 @@ class ParallelTransformerLayer(MegatronModule):
             self.mlp_layernorm = get_norm(cfg)
         if not self.parallel_attn:
             self.post_attention_layernorm = get_norm(cfg)
+        if self.layer_type == LayerType.decoder:
+            # T5-style cross attention over the encoder output
+            self.inter_attention = ParallelAttention(
+                cfg, init_method, output_layer_init_method, layer_number,
+                attention_type=AttnType.cross_attn,
+                attn_mask_type=AttnMaskType.padding, world_size=world_size,
+            )
+            self.post_inter_attention_layernorm = get_norm(cfg)
         self.mlp = ParallelMLP(cfg, init_method, output_layer_init_method,
                                world_size=world_size)
 
     def forward(self, hidden_states, attention_mask, position_ids=None,
-                inference_params=None):
+                inference_params=None, encoder_output=None,
+                enc_dec_attn_mask=None):
         # [s, b, h]
         residual = hidden_states
 
@@ -432,7 +470,18 @@ class ParallelTransformerLayer(MegatronModule):
             attn_out, attn_bias, residual, self.hidden_dropout, self.training
         )
 
-        ln2_out = self.post_attention_layernorm(attn_res)
+        if self.layer_type == LayerType.decoder:
+            ln_cross = self.post_attention_layernorm(attn_res)
+            cross_out, cross_bias = self.inter_attention(
+                ln_cross, enc_dec_attn_mask, encoder_output=encoder_output
+            )
+            attn_res = ops_f.bias_dropout_add(
+                cross_out, cross_bias, attn_res, self.hidden_dropout,
+                self.training,
+            )
+            ln2_out = self.post_inter_attention_layernorm(attn_res)
+        else:
+            ln2_out = self.post_attention_layernorm(attn_res)
         if self.apply_residual_connection_post_layernorm:
             residual2 = ln2_out
         else:
@@ -510,13 +559,17 @@ class ParallelTransformer(MegatronModule):
     def set_input_tensor(self, input_tensor):
         self.input_tensor = input_tensor
 
-    def _checkpointed_forward(self, hidden_states, attention_mask, position_ids):
+    def _checkpointed_forward(self, hidden_states, attention_mask,
+                              position_ids, encoder_output=None,
+                              enc_dec_attn_mask=None):
         def custom(start, end):
             def custom_forward(*args):
                 x = args[0]
                 for index in range(start, end):
                     layer = self._get_layer(index)
-                    x = layer(x, attention_mask, position_ids=position_ids)
+                    x = layer(x, attention_mask, position_ids=position_ids,
+                              encoder_output=encoder_output,
+                              enc_dec_attn_mask=enc_dec_attn_mask)
                 return x
 
             return custom_forward
@@ -544,7 +597,8 @@ class ParallelTransformer(MegatronModule):
         return hidden_states
 
     def forward(self, hidden_states, attention_mask, position_ids=None,
-                inference_params=None):
+                inference_params=None, encoder_output=None,
+                enc_dec_attn_mask=None):
         if not self.pre_process:
             hidden_states = self.input_tensor
 
@@ -556,7 +610,8 @@ class ParallelTransformer(MegatronModule):
         with rng_context:
             if self.recompute_granularity == "full" and self.training:
                 hidden_states = self._checkpointed_forward(
-                    hidden_states, attention_mask, position_ids
+                    hidden_states, attention_mask, position_ids,
+                    encoder_output, enc_dec_attn_mask,
                 )
             else:
                 for index in range(self.num_layers):
@@ -565,6 +620,8 @@ class ParallelTransformer(MegatronModule):
                         hidden_states, attention_mask,
                         position_ids=position_ids,
                         inference_params=inference_params,
+                        encoder_output=encoder_output,
+                        enc_dec_attn_mask=enc_dec_attn_mask,
                     )
 
         if self.post_process:

@@ -43,6 +43,14 @@ void fused_adam(std::vector<torch::Tensor> params,
                 std::vector<torch::Tensor> exp_avg_sqs, double lr,
                 double beta1, double beta2, double eps, double wd,
                 int64_t step, int64_t adam_w_mode);
+void fused_adam_with_model_copy(std::vector<torch::Tensor> params,
+                                std::vector<torch::Tensor> grads,
+                                std::vector<torch::Tensor> exp_avgs,
+                                std::vector<torch::Tensor> exp_avg_sqs,
+                                std::vector<torch::Tensor> model_params,
+                                double lr, double beta1, double beta2,
+                                double eps, double wd, int64_t step,
+                                int64_t adam_w_mode);
 
 // wgrad.hip
 void wgrad_gemm_accum_fp32(torch::Tensor input, torch::Tensor grad_output,
@@ -73,6 +81,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bias_dropout_add_fwd", &bias_dropout_add_fwd);
   m.def("dropout_bwd", &dropout_bwd);
   m.def("fused_adam", &fused_adam);
+  m.def("fused_adam_with_model_copy", &fused_adam_with_model_copy);
   m.def("wgrad_gemm_accum_fp32", &wgrad_gemm_accum_fp32);
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_bwd", &flash_attn_bwd);

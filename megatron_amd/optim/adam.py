@@ -24,6 +24,7 @@ class FusedAdam(torch.optim.Optimizer):
 
     @torch.no_grad()
     def step(self, closure=None):
+        self.wrote_model_params = False
         loss = None
         if closure is not None:
             with torch.enable_grad():
@@ -62,11 +63,25 @@ class FusedAdam(torch.optim.Optimizer):
                 if hasattr(mod, "fused_adam"):
                     ext = mod
             if ext is not None:
-                ext.fused_adam(
-                    params, grads, exp_avgs, exp_avg_sqs,
-                    lr, beta1, beta2, eps, wd, step,
-                    1 if self.adam_w_mode else 0,
-                )
+                model_outs = [getattr(p, "model_out", None) for p in params]
+                if all(m is not None for m in model_outs):
+                    # fused master update + bf16 model-param write-back
+                    ext.fused_adam_with_model_copy(
+                        [p.view(-1) for p in params],
+                        [g.reshape(-1) for g in grads],
+                        [m.view(-1) for m in exp_avgs],
+                        [v.view(-1) for v in exp_avg_sqs],
+                        [m.view(-1) for m in model_outs],
+                        lr, beta1, beta2, eps, wd, step,
+                        1 if self.adam_w_mode else 0,
+                    )
+                    self.wrote_model_params = True
+                else:
+                    ext.fused_adam(
+                        params, grads, exp_avgs, exp_avg_sqs,
+                        lr, beta1, beta2, eps, wd, step,
+                        1 if self.adam_w_mode else 0,
+                    )
                 continue
 
             if self.adam_w_mode and wd != 0.0:

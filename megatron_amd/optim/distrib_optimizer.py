@@ -172,6 +172,7 @@ class DistributedOptimizer(MixedPrecisionOptimizer):
                         if hasattr(param, attr):
                             setattr(shard_model_param, attr, getattr(param, attr))
                             setattr(shard_main_param, attr, getattr(param, attr))
+                    shard_main_param.model_out = shard_model_param
                     model_float16_params_this_group.append(param)
                     shard_float16_params_this_group.append(shard_model_param)
                     shard_fp32_from_float16_params_this_group.append(shard_main_param)

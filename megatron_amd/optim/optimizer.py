@@ -316,7 +316,8 @@ class MixedPrecisionOptimizer(MegatronOptimizer):
         _t("optimizer-inner-step", False)
 
         _t("optimizer-copy-main-to-model-params", True)
-        self._copy_main_params_to_model_params()
+        if not getattr(self.optimizer, "wrote_model_params", False):
+            self._copy_main_params_to_model_params()
         _t("optimizer-copy-main-to-model-params", False)
 
         return True, grad_norm, num_zeros_in_grad
@@ -358,6 +359,9 @@ class Float16OptimizerWithFloat16Params(MixedPrecisionOptimizer):
                         if hasattr(param, attr):
                             setattr(main_param, attr, getattr(param, attr))
                     param.main_param = main_param
+                    # lets the fused Adam kernel write the bf16/fp16 model
+                    # copy in the same pass (ops/csrc/adam.hip)
+                    main_param.model_out = param
                     param_group["params"][i] = main_param
                     fp32_from_float16_params_this_group.append(main_param)
                     if param in self.optimizer.state:
