@@ -49,6 +49,59 @@ __device__ __forceinline__ float act_grad(int mode, float v) {
   }
 }
 
+// vectorized bf16 GLU (F % 8 == 0): 16-B loads of both halves
+__global__ void glu_fwd_kernel_bf16v(const __hip_bfloat16* __restrict__ x,
+                                     __hip_bfloat16* __restrict__ y,
+                                     long rows, int F, int mode) {
+  const int FV = F / 8;
+  const long total = rows * (long)FV;
+  for (long idx = (long)blockIdx.x * kBlock + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * kBlock) {
+    const long r = idx / FV;
+    const int j = idx % FV;
+    const uint4* xr = reinterpret_cast<const uint4*>(x + r * (long)(2 * F));
+    Bf16x8 v1, v2, vo;
+    v1.u = xr[j];
+    v2.u = xr[FV + j];
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      vo.h[k] = __float2bfloat16(
+          __bfloat162float(v1.h[k]) *
+          act_eval(mode, __bfloat162float(v2.h[k])));
+    }
+    reinterpret_cast<uint4*>(y + r * (long)F)[j] = vo.u;
+  }
+}
+
+__global__ void glu_bwd_kernel_bf16v(const __hip_bfloat16* __restrict__ dy,
+                                     const __hip_bfloat16* __restrict__ x,
+                                     __hip_bfloat16* __restrict__ dx,
+                                     long rows, int F, int mode) {
+  const int FV = F / 8;
+  const long total = rows * (long)FV;
+  for (long idx = (long)blockIdx.x * kBlock + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * kBlock) {
+    const long r = idx / FV;
+    const int j = idx % FV;
+    const uint4* xr = reinterpret_cast<const uint4*>(x + r * (long)(2 * F));
+    uint4* dxr = reinterpret_cast<uint4*>(dx + r * (long)(2 * F));
+    Bf16x8 vg, v1, v2, d1, d2;
+    vg.u = reinterpret_cast<const uint4*>(dy + r * (long)F)[j];
+    v1.u = xr[j];
+    v2.u = xr[FV + j];
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      float g = __bfloat162float(vg.h[k]);
+      float x1 = __bfloat162float(v1.h[k]);
+      float x2 = __bfloat162float(v2.h[k]);
+      d1.h[k] = __float2bfloat16(g * act_eval(mode, x2));
+      d2.h[k] = __float2bfloat16(g * x1 * act_grad(mode, x2));
+    }
+    dxr[j] = d1.u;
+    dxr[FV + j] = d2.u;
+  }
+}
+
 // y[i, j] = x1[i, j] * act(x2[i, j]),  x = [x1 | x2] along last dim
 template <typename T>
 __global__ void glu_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
@@ -177,6 +230,13 @@ torch::Tensor glu_fwd(torch::Tensor x, int64_t mode) {
   auto y = torch::empty({rows, F}, x.options());
   auto stream = c10::hip::getCurrentHIPStream();
   long total = rows * (long)F;
+  if (x.scalar_type() == torch::kBFloat16 && F % 8 == 0) {
+    hipLaunchKernelGGL(glu_fwd_kernel_bf16v, dim3(grid_for(total / 8)),
+                       dim3(kBlock), 0, stream,
+                       (const __hip_bfloat16*)x.data_ptr(),
+                       (__hip_bfloat16*)y.data_ptr(), rows, F, (int)mode);
+    return y;
+  }
 #define LAUNCH_GLU_F(T)                                                      \
   hipLaunchKernelGGL((glu_fwd_kernel<T>), dim3(grid_for(total)),             \
                      dim3(kBlock), 0, stream, (const T*)x.data_ptr(),        \
@@ -192,6 +252,14 @@ torch::Tensor glu_bwd(torch::Tensor dy, torch::Tensor x, int64_t mode) {
   auto dx = torch::empty_like(x);
   auto stream = c10::hip::getCurrentHIPStream();
   long total = rows * (long)F;
+  if (x.scalar_type() == torch::kBFloat16 && F % 8 == 0) {
+    hipLaunchKernelGGL(glu_bwd_kernel_bf16v, dim3(grid_for(total / 8)),
+                       dim3(kBlock), 0, stream,
+                       (const __hip_bfloat16*)dy.data_ptr(),
+                       (const __hip_bfloat16*)x.data_ptr(),
+                       (__hip_bfloat16*)dx.data_ptr(), rows, F, (int)mode);
+    return dx;
+  }
 #define LAUNCH_GLU_B(T)                                                      \
   hipLaunchKernelGGL((glu_bwd_kernel<T>), dim3(grid_for(total)),             \
                      dim3(kBlock), 0, stream, (const T*)dy.data_ptr(),       \

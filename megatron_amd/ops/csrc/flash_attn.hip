@@ -147,7 +147,7 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
   constexpr int DTILES = D / 16;
 
   __shared__ __hip_bfloat16 p_lds[4][16 * kStrip];
-  __shared__ __hip_bfloat16 v_img[kBlockN * D];
+  __shared__ __hip_bfloat16 v_img[2][kBlockN * D];
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -203,11 +203,21 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
     if (min_key > 0) kb_start = min_key / kBlockN;
   }
 
+  // prologue: stage first V tile
+  stage_tr_image<D, kBlockN>(v_img[kb_start & 1], v + v_base, rk,
+                             kb_start * kBlockN, Sk);
+  __syncthreads();
+
   for (int kb = kb_start; kb < kb_end; ++kb) {
     const int kstart = kb * kBlockN;
-
-    stage_tr_image<D, kBlockN>(v_img, v + v_base, rk, kstart, Sk);
-    __syncthreads();
+    const __hip_bfloat16* v_cur = v_img[kb & 1];
+    // prefetch next V tile into the other buffer (loads overlap this
+    // iteration's compute; the single end-of-iteration barrier both
+    // publishes it and retires this buffer's readers)
+    if (kb + 1 < kb_end) {
+      stage_tr_image<D, kBlockN>(v_img[(kb + 1) & 1], v + v_base, rk,
+                                 (kb + 1) * kBlockN, Sk);
+    }
 
     // S = Q K^T
     frag_f32 st[4];
@@ -217,19 +227,13 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
     for (int t = 0; t < 4; ++t) {
       int key = kstart + t * 16 + row_in_tile;
       int kr = key < Sk ? key : Sk - 1;
-      frag_b16 bf[KFRAGS];
 #pragma unroll
       for (int kk = 0; kk < KFRAGS; ++kk) {
-        bf[kk] = load_frag_global(k + k_base + (long)kr * rk + kk * 32 +
-                                  kgroup * 8);
+        frag_b16 bf = load_frag_global(k + k_base + (long)kr * rk + kk * 32 +
+                                       kgroup * 8);
+        st[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[kk], bf, st[t], 0,
+                                                        0, 0);
       }
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int kk = 0; kk < KFRAGS; ++kk) {
-        st[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[kk], bf[kk],
-                                                        st[t], 0, 0, 0);
-      }
-      __builtin_amdgcn_s_setprio(0);
     }
 
     float s_val[4][4];
@@ -285,7 +289,7 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
         p_lds[wave][row * kStrip + col] = __float2bfloat16(p_val[t][r]);
       }
     }
-    __syncthreads();
+    // strips are wave-private: lgkmcnt ordering suffices, no barrier
 
     // O += P V
 #pragma unroll
@@ -301,18 +305,12 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
         cvt.u = *reinterpret_cast<const uint4*>(src);
         pf = cvt.f;
       }
-      frag_b16 vf[DTILES];
 #pragma unroll
       for (int t = 0; t < DTILES; ++t) {
-        vf[t] = tr_bfrag<D>(v_img, kk2 * 32, t, lane);
+        frag_b16 vf = tr_bfrag<D>(v_cur, kk2 * 32, t, lane);
+        o_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vf, o_acc[t],
+                                                           0, 0, 0);
       }
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int t = 0; t < DTILES; ++t) {
-        o_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vf[t],
-                                                           o_acc[t], 0, 0, 0);
-      }
-      __builtin_amdgcn_s_setprio(0);
     }
     __syncthreads();
   }
@@ -448,23 +446,17 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
       for (int t = 0; t < 4; ++t) {
         int qrow = qstart + t * 16 + row_in_tile;
         int qr = qrow < Sq ? qrow : Sq - 1;
-        frag_b16 qbf[KFRAGS], dbf[KFRAGS];
 #pragma unroll
         for (int kk = 0; kk < KFRAGS; ++kk) {
-          qbf[kk] = load_frag_global(q + q_base + (long)qr * rq + kk * 32 +
-                                     kgroup * 8);
-          dbf[kk] = load_frag_global(dout + q_base + (long)qr * rq + kk * 32 +
-                                     kgroup * 8);
-        }
-        __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-        for (int kk = 0; kk < KFRAGS; ++kk) {
-          stt[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ka[kk], qbf[kk],
+          frag_b16 qbf = load_frag_global(q + q_base + (long)qr * rq +
+                                          kk * 32 + kgroup * 8);
+          stt[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ka[kk], qbf,
                                                            stt[t], 0, 0, 0);
-          dpt[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(va[kk], dbf[kk],
+          frag_b16 dbf = load_frag_global(dout + q_base + (long)qr * rq +
+                                          kk * 32 + kgroup * 8);
+          dpt[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(va[kk], dbf,
                                                            dpt[t], 0, 0, 0);
         }
-        __builtin_amdgcn_s_setprio(0);
       }
 
 #pragma unroll
@@ -508,21 +500,15 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
           ptf = c1.f;
           dstf = c2.f;
         }
-        frag_b16 dof[DTILES], qtf[DTILES];
 #pragma unroll
         for (int t = 0; t < DTILES; ++t) {
-          dof[t] = tr_bfrag<D>(do_img, kk2 * 32, t, lane);
-          qtf[t] = tr_bfrag<D>(q_img, kk2 * 32, t, lane);
-        }
-        __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-        for (int t = 0; t < DTILES; ++t) {
+          frag_b16 dof = tr_bfrag<D>(do_img, kk2 * 32, t, lane);
           dv_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              ptf, dof[t], dv_acc[t], 0, 0, 0);
+              ptf, dof, dv_acc[t], 0, 0, 0);
+          frag_b16 qtf = tr_bfrag<D>(q_img, kk2 * 32, t, lane);
           dk_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              dstf, qtf[t], dk_acc[t], 0, 0, 0);
+              dstf, qtf, dk_acc[t], 0, 0, 0);
         }
-        __builtin_amdgcn_s_setprio(0);
       }
       __syncthreads();
     }
@@ -557,7 +543,7 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dq_kernel(
   constexpr int DTILES = D / 16;
 
   __shared__ __hip_bfloat16 ds_lds[4][16 * kStrip];
-  __shared__ __hip_bfloat16 k_img[kBlockN * D];
+  __shared__ __hip_bfloat16 k_img[2][kBlockN * D];
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -615,11 +601,17 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dq_kernel(
     delta_r[r] = row < Sq ? delta_h[row] : 0.f;
   }
 
+  stage_tr_image<D, kBlockN>(k_img[kb_start & 1], k + k_base, rk,
+                             kb_start * kBlockN, Sk);
+  __syncthreads();
+
   for (int kb = kb_start; kb < kb_end; ++kb) {
     const int kstart = kb * kBlockN;
-
-    stage_tr_image<D, kBlockN>(k_img, k + k_base, rk, kstart, Sk);
-    __syncthreads();
+    const __hip_bfloat16* k_cur = k_img[kb & 1];
+    if (kb + 1 < kb_end) {
+      stage_tr_image<D, kBlockN>(k_img[(kb + 1) & 1], k + k_base, rk,
+                                 (kb + 1) * kBlockN, Sk);
+    }
 
     frag_f32 st[4], dp[4];
 #pragma unroll
@@ -631,23 +623,17 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dq_kernel(
     for (int t = 0; t < 4; ++t) {
       int key = kstart + t * 16 + row_in_tile;
       int kr = key < Sk ? key : Sk - 1;
-      frag_b16 kbf[KFRAGS], vbf[KFRAGS];
 #pragma unroll
       for (int kk = 0; kk < KFRAGS; ++kk) {
-        kbf[kk] = load_frag_global(k + k_base + (long)kr * rk + kk * 32 +
-                                   kgroup * 8);
-        vbf[kk] = load_frag_global(v + k_base + (long)kr * rk + kk * 32 +
-                                   kgroup * 8);
+        frag_b16 kbf = load_frag_global(k + k_base + (long)kr * rk + kk * 32 +
+                                        kgroup * 8);
+        st[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[kk], kbf, st[t],
+                                                        0, 0, 0);
+        frag_b16 vbf = load_frag_global(v + k_base + (long)kr * rk + kk * 32 +
+                                        kgroup * 8);
+        dp[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dof[kk], vbf, dp[t],
+                                                        0, 0, 0);
       }
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int kk = 0; kk < KFRAGS; ++kk) {
-        st[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[kk], kbf[kk],
-                                                        st[t], 0, 0, 0);
-        dp[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dof[kk], vbf[kk],
-                                                        dp[t], 0, 0, 0);
-      }
-      __builtin_amdgcn_s_setprio(0);
     }
 
 #pragma unroll
@@ -669,7 +655,7 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dq_kernel(
         ds_lds[wave][lrow * kStrip + lcol] = __float2bfloat16(ds);
       }
     }
-    __syncthreads();
+    // wave-private strip: no barrier needed
 
     // dQ += dS K
 #pragma unroll
@@ -685,19 +671,13 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dq_kernel(
         cvt.u = *reinterpret_cast<const uint4*>(src);
         dsf = cvt.f;
       }
-      frag_b16 ktf[DTILES];
 #pragma unroll
       for (int t = 0; t < DTILES; ++t) {
-        ktf[t] = tr_bfrag<D>(k_img, kk2 * 32, t, lane);
-      }
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int t = 0; t < DTILES; ++t) {
-        dq_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, ktf[t],
+        frag_b16 ktf = tr_bfrag<D>(k_cur, kk2 * 32, t, lane);
+        dq_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, ktf,
                                                             dq_acc[t], 0, 0,
                                                             0);
       }
-      __builtin_amdgcn_s_setprio(0);
     }
     __syncthreads();
   }

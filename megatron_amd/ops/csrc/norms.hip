@@ -124,6 +124,103 @@ __global__ void norm_fwd_kernel_bf16v(const __hip_bfloat16* __restrict__ x,
   }
 }
 
+// Vectorized bf16 RMSNorm backward (H % 8 == 0, H <= BLOCK*8*MAX_CV):
+// 16-B loads/stores, dweight accumulated in registers per block and folded
+// with one atomicAdd per element at the end.
+template <bool RMS, int BLOCK, int MAX_CV>
+__global__ void norm_bwd_kernel_bf16v(const __hip_bfloat16* __restrict__ dy,
+                                      const __hip_bfloat16* __restrict__ x,
+                                      const __hip_bfloat16* __restrict__ weight,
+                                      const float* __restrict__ mean,
+                                      const float* __restrict__ inv,
+                                      __hip_bfloat16* __restrict__ dx,
+                                      float* __restrict__ dweight,
+                                      float* __restrict__ dbias, long rows,
+                                      int H) {
+  __shared__ float lds[BLOCK / WAVE_SIZE];
+  const int HV = H / 8;
+  const int CV = (HV + BLOCK - 1) / BLOCK;  // <= MAX_CV vec-chunks per thread
+
+  float acc_dw[MAX_CV][8];
+  float acc_db[MAX_CV][8];
+#pragma unroll
+  for (int c = 0; c < MAX_CV; ++c) {
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      acc_dw[c][k] = 0.f;
+      acc_db[c][k] = 0.f;
+    }
+  }
+
+  const uint4* wv = reinterpret_cast<const uint4*>(weight);
+
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const uint4* dyr = reinterpret_cast<const uint4*>(dy + row * (long)H);
+    const uint4* xr = reinterpret_cast<const uint4*>(x + row * (long)H);
+    uint4* dxr = reinterpret_cast<uint4*>(dx + row * (long)H);
+    const float m = RMS ? 0.f : mean[row];
+    const float r = inv[row];
+
+    float dot = 0.f, gsum = 0.f;
+    for (int i = threadIdx.x; i < HV; i += BLOCK) {
+      Bf16x8 vg, vx, vw;
+      vg.u = dyr[i];
+      vx.u = xr[i];
+      vw.u = wv[i];
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float g = __bfloat162float(vg.h[k]);
+        float xv = __bfloat162float(vx.h[k]);
+        float w = __bfloat162float(vw.h[k]);
+        float xhat = (xv - m) * r;
+        float gw = g * w;
+        dot += gw * xhat;
+        if (!RMS) gsum += gw;
+      }
+    }
+    dot = block_reduce_sum<BLOCK>(dot, lds);
+    if (!RMS) {
+      __syncthreads();
+      gsum = block_reduce_sum<BLOCK>(gsum, lds);
+    }
+    const float inv_H = 1.0f / H;
+
+    for (int i = threadIdx.x, c = 0; i < HV; i += BLOCK, ++c) {
+      Bf16x8 vg, vx, vw, vo;
+      vg.u = dyr[i];
+      vx.u = xr[i];
+      vw.u = wv[i];
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float g = __bfloat162float(vg.h[k]);
+        float xv = __bfloat162float(vx.h[k]);
+        float w = __bfloat162float(vw.h[k]);
+        float xhat = (xv - m) * r;
+        float gw = g * w;
+        float d;
+        if (RMS) {
+          d = (gw - xhat * dot * inv_H) * r;
+        } else {
+          d = (gw - gsum * inv_H - xhat * dot * inv_H) * r;
+        }
+        vo.h[k] = __float2bfloat16(d);
+        acc_dw[c][k] += g * xhat;
+        if (!RMS) acc_db[c][k] += g;
+      }
+      dxr[i] = vo.u;
+    }
+    __syncthreads();
+  }
+
+  for (int i = threadIdx.x, c = 0; i < HV; i += BLOCK, ++c) {
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      atomicAdd(&dweight[i * 8 + k], acc_dw[c][k]);
+      if (!RMS) atomicAdd(&dbias[i * 8 + k], acc_db[c][k]);
+    }
+  }
+}
+
 // Backward. MAX_ACC register accumulators per thread for dweight/dbias.
 template <typename T, bool RMS, int BLOCK, int MAX_ACC>
 __global__ void norm_bwd_kernel(const T* __restrict__ dy,
@@ -253,6 +350,21 @@ void norm_bwd_launch(const torch::Tensor& dy, const torch::Tensor& x,
               kBlock * kMaxAcc, ")");
   int grid = (int)std::min<long>(rows, 2048);
   auto stream = c10::hip::getCurrentHIPStream();
+  if constexpr (std::is_same_v<T, __hip_bfloat16>) {
+    if (H % 8 == 0 && H <= kBlock * 8 * 4) {
+      hipLaunchKernelGGL((norm_bwd_kernel_bf16v<RMS, kBlock, 4>), dim3(grid),
+                         dim3(kBlock), 0, stream,
+                         (const __hip_bfloat16*)dy.data_ptr(),
+                         (const __hip_bfloat16*)x.data_ptr(),
+                         (const __hip_bfloat16*)w.data_ptr(),
+                         RMS ? nullptr : mean->data_ptr<float>(),
+                         inv.data_ptr<float>(),
+                         (__hip_bfloat16*)dx.data_ptr(),
+                         dw.data_ptr<float>(),
+                         RMS ? nullptr : db->data_ptr<float>(), rows, H);
+      return;
+    }
+  }
   hipLaunchKernelGGL((norm_bwd_kernel<T, RMS, kBlock, kMaxAcc>), dim3(grid),
                      dim3(kBlock), 0, stream, (const T*)dy.data_ptr(),
                      (const T*)x.data_ptr(), (const T*)w.data_ptr(),
