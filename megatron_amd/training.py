@@ -413,6 +413,13 @@ def _train(forward_step_func, model, optimizer, opt_param_scheduler,
                 iteration, cfg, process_non_loss_data_func, False,
             )
 
+        # cluster preemption (ADLR autoresume) polling
+        if iteration % 50 == 0:
+            from .utils import check_adlr_autoresume_termination
+
+            check_adlr_autoresume_termination(iteration, model, optimizer,
+                                              opt_param_scheduler, cfg)
+
         saved_checkpoint = False
         if cfg.exit_signal_handler:
             signal_handler = global_state.get_signal_handler()

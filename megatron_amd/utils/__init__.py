@@ -138,3 +138,51 @@ def print_rank_last(message):
             print(message, flush=True)
     else:
         print(message, flush=True)
+
+
+# --- ADLR autoresume (cluster preemption) hook (reference utils.py:117-134) --
+
+_ADLR_AUTORESUME = None
+
+
+def get_adlr_autoresume():
+    """Lazily import the site-specific AutoResume module if present."""
+    global _ADLR_AUTORESUME
+    if _ADLR_AUTORESUME is None:
+        import os
+        import sys
+
+        sys.path.append(os.environ.get("SUBMIT_SCRIPTS", "."))
+        try:
+            from userlib.auto_resume import AutoResume  # site package
+
+            AutoResume.init()
+            _ADLR_AUTORESUME = AutoResume
+        except ImportError:
+            _ADLR_AUTORESUME = False
+    return _ADLR_AUTORESUME or None
+
+
+def check_adlr_autoresume_termination(iteration, model, optimizer,
+                                      opt_param_scheduler, cfg):
+    """Save a checkpoint and request resume when the cluster signals
+    preemption."""
+    import sys
+
+    import torch
+
+    from ..checkpointing import save_checkpoint
+
+    autoresume = get_adlr_autoresume()
+    if autoresume is None:
+        return
+    torch.distributed.barrier()
+    if autoresume.termination_requested():
+        if cfg.save:
+            save_checkpoint(iteration, model, optimizer, opt_param_scheduler,
+                            cfg)
+        print_rank_0(">>> autoresume termination request found!")
+        if torch.distributed.get_rank() == 0:
+            autoresume.request_resume()
+        print_rank_0(">>> training terminated. Returning")
+        sys.exit(0)
