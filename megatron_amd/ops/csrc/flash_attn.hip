@@ -4,27 +4,31 @@
 // (transformer.py:9,369,528-552): causal + sliding-window attention with
 // native GQA/MQA head broadcast (no K/V expansion), forward + backward.
 //
-// v2 design notes:
+// v4 design (measured on MI355X; see profiles/):
 //  - mfma_f32_16x16x32_bf16 tiles; 256-thread blocks = 4 waves;
 //    BLOCK_M = BLOCK_N = 64; head dims 64 / 128.
-//  - Q/K A- and B-fragments whose 8-element k-runs lie along the head dim
-//    load as contiguous 16-byte chunks straight from HBM ([S][D] layout).
-//  - Fragments that need the d-major orientation (V in PV, dO/Q in dK/dV,
-//    K in dQ) are staged in LDS "tr images" of [row/4][D/16][4][16] tiles
-//    and read with the gfx950 hardware transpose read ds_read_b64_tr_b16
-//    (lane-linear per-16-lane-group addresses; mapping verified empirically
-//    with ops/csrc/probe_tr.hip on MI355X: with per-lane address
-//    base + (lane&15)*8B, lane l elem j reads lds16[base + (l&15) + 16*j]).
-//    This replaces v1's 8 scalar ds_read_u16 per fragment with 2 tr reads
-//    (the dominant non-MFMA cost in v1's inner loop).
-//  - s_setprio(1) brackets the MFMA clusters (technique T5: the 4-wave
-//    schedule has load/compute role diversity for the scheduler to
-//    arbitrate).
-//  - online softmax in fp32 registers; cross-lane row reductions are
-//    __shfl_xor within 16-lane groups (wave64).
-//  - backward is split k-parallel (dK/dV, GQA-group loop => register
-//    accumulation, no atomics) and q-parallel (dQ) plus a small
-//    delta = rowsum(dO*O) preprocess kernel, using the stored logsumexp.
+//  - every K/V/dO/Q tile is staged ONCE per block into a padded LDS
+//    "tr image" ([row/4] tile-rows of (D/16) [4][16] segments, +8-element
+//    row pad) that serves BOTH fragment orientations conflict-free:
+//      * d-run A/B fragments (8 contiguous head-dims of one row) as single
+//        16-byte ds_read_b128 — lanes' 4-dword footprints interleave across
+//        the 64 banks thanks to the +8 pad;
+//      * row-run B fragments (8 contiguous rows at one head-dim) via the
+//        gfx950 hardware transpose read ds_read_b64_tr_b16 (lane mapping
+//        verified empirically with ops/csrc/probe_tr.hip: per-lane address
+//        base + (lane&15)*8B delivers lds16[base + (lane&15) + 16*j]).
+//    The inner loop therefore touches HBM only through the cooperative
+//    staging copies, which are double-buffered so the next tile's loads
+//    overlap this tile's MFMAs with ONE barrier per iteration (v2/v3's
+//    per-wave global B-fragment loads serialized a load->MFMA chain on
+//    every k-step — ~70% SQ_WAIT_ANY in PMC, profiles/).
+//  - online softmax in fp32 registers; row reductions are __shfl_xor within
+//    16-lane groups (wave64); P/dS round-trip through wave-private padded
+//    strips (no barrier).
+//  - backward splits k-parallel dK/dV (flattened (gqa-head, q-block) loop =>
+//    register accumulation, no atomics) and q-parallel dQ, plus a
+//    delta = rowsum(dO*O) preprocess, using the stored logsumexp
+//    (FlashAttention-2 scheme).
 //
 // Fragment maps (gfx950 mfma_f32_16x16x32_bf16, verified by the GPU parity
 // suite):
@@ -48,17 +52,7 @@ using v4s = __attribute__((ext_vector_type(4))) short;
 constexpr int kBlockM = 64;
 constexpr int kBlockN = 64;
 constexpr int kThreads = 256;
-constexpr int kStrip = 72;  // padded LDS strip stride (bf16 elems), P/dS tiles
-
-__device__ __forceinline__ frag_b16 load_frag_global(const __hip_bfloat16* p) {
-  uint4 u = *reinterpret_cast<const uint4*>(p);
-  union {
-    uint4 u;
-    frag_b16 f;
-  } cvt;
-  cvt.u = u;
-  return cvt.f;
-}
+constexpr int kStrip = 72;  // padded strip stride (bf16), P/dS tiles
 
 __device__ __forceinline__ float group16_max(float v) {
 #pragma unroll
@@ -76,23 +70,59 @@ __device__ __forceinline__ float group16_sum(float v) {
   return v;
 }
 
-// --- tr-image helpers ------------------------------------------------------
-// image layout: rows x D stored as [row/4][D/16] tiles of [4][16] row-major.
+// --- padded tr-image -------------------------------------------------------
+
+template <int D>
+constexpr int tr_stride() {
+  return D * 4 + 8;  // one 4-row tile-row + 8-element pad
+}
+
+template <int D>
+constexpr int tr_elems(int rows) {
+  return (rows / 4) * tr_stride<D>();
+}
 
 template <int D>
 __device__ __forceinline__ int tr_off(int row, int col) {
-  return ((row >> 2) * (D / 16) + (col >> 4)) * 64 + (row & 3) * 16 +
+  return (row >> 2) * tr_stride<D>() + (col >> 4) * 64 + (row & 3) * 16 +
          (col & 15);
 }
 
-// B-fragment B[row0 + (l>>4)*8 + jj][dtile*16 + (l&15)], jj = 0..7.
+__device__ __forceinline__ frag_b16 lds_read16(const __hip_bfloat16* p) {
+  union {
+    uint4 u;
+    frag_b16 f;
+  } cvt;
+  cvt.u = *reinterpret_cast<const uint4*>(p);
+  return cvt.f;
+}
+
+__device__ __forceinline__ frag_b16 global_read16(const __hip_bfloat16* p) {
+  union {
+    uint4 u;
+    frag_b16 f;
+  } cvt;
+  cvt.u = *reinterpret_cast<const uint4*>(p);
+  return cvt.f;
+}
+
+// d-run fragment: 8 contiguous head-dims of row `row`, starting at d0
+// (d0 % 16 in {0, 8}).
+template <int D>
+__device__ __forceinline__ frag_b16 img_dfrag(const __hip_bfloat16* img,
+                                              int row, int d0) {
+  return lds_read16(&img[tr_off<D>(row, d0)]);
+}
+
+// row-run B fragment: B[row0 + (l>>4)*8 + jj][dtile*16 + (l&15)] via two
+// hardware transpose reads.
 template <int D>
 __device__ __forceinline__ frag_b16 tr_bfrag(const __hip_bfloat16* img,
                                              int row0, int dtile, int lane) {
   const int c = lane & 15;
   const int ktile = (row0 >> 2) + (lane >> 4) * 2;
-  const int b1 = (ktile * (D / 16) + dtile) * 64;
-  const int b2 = ((ktile + 1) * (D / 16) + dtile) * 64;
+  const int b1 = ktile * tr_stride<D>() + dtile * 64;
+  const int b2 = b1 + tr_stride<D>();
   const __attribute__((address_space(3))) v4s* p1 =
       (const __attribute__((address_space(3))) v4s*)&img[b1] + c;
   const __attribute__((address_space(3))) v4s* p2 =
@@ -145,9 +175,11 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
     float scale, int causal, int window) {
   constexpr int KFRAGS = D / 32;
   constexpr int DTILES = D / 16;
+  constexpr int IMG = tr_elems<D>(kBlockN);
 
   __shared__ __hip_bfloat16 p_lds[4][16 * kStrip];
-  __shared__ __hip_bfloat16 v_img[2][kBlockN * D];
+  __shared__ __hip_bfloat16 k_img[2][IMG];
+  __shared__ __hip_bfloat16 v_img[2][IMG];
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -175,8 +207,8 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
     int qr = qrow < Sq ? qrow : Sq - 1;
 #pragma unroll
     for (int kk = 0; kk < KFRAGS; ++kk) {
-      qf[kk] = load_frag_global(q + q_base + (long)qr * rq + kk * 32 +
-                                kgroup * 8);
+      qf[kk] = global_read16(q + q_base + (long)qr * rq + kk * 32 +
+                             kgroup * 8);
     }
   }
 
@@ -203,34 +235,35 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
     if (min_key > 0) kb_start = min_key / kBlockN;
   }
 
-  // prologue: stage first V tile
+  stage_tr_image<D, kBlockN>(k_img[kb_start & 1], k + k_base, rk,
+                             kb_start * kBlockN, Sk);
   stage_tr_image<D, kBlockN>(v_img[kb_start & 1], v + v_base, rk,
                              kb_start * kBlockN, Sk);
   __syncthreads();
 
   for (int kb = kb_start; kb < kb_end; ++kb) {
     const int kstart = kb * kBlockN;
+    const __hip_bfloat16* k_cur = k_img[kb & 1];
     const __hip_bfloat16* v_cur = v_img[kb & 1];
-    // prefetch next V tile into the other buffer (loads overlap this
-    // iteration's compute; the single end-of-iteration barrier both
-    // publishes it and retires this buffer's readers)
+    // issue next tile's staging loads; they land behind this iteration's
+    // compute and are published by the single end-of-iteration barrier
     if (kb + 1 < kb_end) {
+      stage_tr_image<D, kBlockN>(k_img[(kb + 1) & 1], k + k_base, rk,
+                                 (kb + 1) * kBlockN, Sk);
       stage_tr_image<D, kBlockN>(v_img[(kb + 1) & 1], v + v_base, rk,
                                  (kb + 1) * kBlockN, Sk);
     }
 
-    // S = Q K^T
+    // S = Q K^T : B-fragments = K d-runs from LDS
     frag_f32 st[4];
 #pragma unroll
     for (int t = 0; t < 4; ++t) st[t] = frag_f32{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-    for (int t = 0; t < 4; ++t) {
-      int key = kstart + t * 16 + row_in_tile;
-      int kr = key < Sk ? key : Sk - 1;
+    for (int kk = 0; kk < KFRAGS; ++kk) {
 #pragma unroll
-      for (int kk = 0; kk < KFRAGS; ++kk) {
-        frag_b16 bf = load_frag_global(k + k_base + (long)kr * rk + kk * 32 +
-                                       kgroup * 8);
+      for (int t = 0; t < 4; ++t) {
+        frag_b16 bf = img_dfrag<D>(k_cur, t * 16 + row_in_tile,
+                                   kk * 32 + kgroup * 8);
         st[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[kk], bf, st[t], 0,
                                                         0, 0);
       }
@@ -291,20 +324,11 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
     }
     // strips are wave-private: lgkmcnt ordering suffices, no barrier
 
-    // O += P V
+    // O += P V : A = P strip, B = V row-runs (transpose reads)
 #pragma unroll
     for (int kk2 = 0; kk2 < 2; ++kk2) {
-      frag_b16 pf;
-      {
-        const __hip_bfloat16* src =
-            &p_lds[wave][row_in_tile * kStrip + kk2 * 32 + kgroup * 8];
-        union {
-          uint4 u;
-          frag_b16 f;
-        } cvt;
-        cvt.u = *reinterpret_cast<const uint4*>(src);
-        pf = cvt.f;
-      }
+      frag_b16 pf = lds_read16(
+          &p_lds[wave][row_in_tile * kStrip + kk2 * 32 + kgroup * 8]);
 #pragma unroll
       for (int t = 0; t < DTILES; ++t) {
         frag_b16 vf = tr_bfrag<D>(v_cur, kk2 * 32, t, lane);
@@ -356,7 +380,8 @@ __global__ void fa_bwd_delta_kernel(const __hip_bfloat16* __restrict__ dout,
 }
 
 // ---------------------------------------------------------------------------
-// backward dK/dV (k-parallel)
+// backward dK/dV (k-parallel; flattened (gqa head, q-block) loop with
+// double-buffered dO/Q images)
 
 template <int D>
 __global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
@@ -368,11 +393,12 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
     float scale, int causal, int window) {
   constexpr int KFRAGS = D / 32;
   constexpr int DTILES = D / 16;
+  constexpr int IMG = tr_elems<D>(kBlockM);
 
   __shared__ __hip_bfloat16 pt_lds[4][16 * kStrip];
   __shared__ __hip_bfloat16 dst_lds[4][16 * kStrip];
-  __shared__ __hip_bfloat16 do_img[kBlockM * D];
-  __shared__ __hip_bfloat16 q_img[kBlockM * D];
+  __shared__ __hip_bfloat16 do_img[2][IMG];
+  __shared__ __hip_bfloat16 q_img[2][IMG];
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -397,10 +423,10 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
     int kr = key < Sk ? key : Sk - 1;
 #pragma unroll
     for (int kk = 0; kk < KFRAGS; ++kk) {
-      ka[kk] = load_frag_global(k + k_base + (long)kr * rk + kk * 32 +
-                                kgroup * 8);
-      va[kk] = load_frag_global(v + k_base + (long)kr * rk + kk * 32 +
-                                kgroup * 8);
+      ka[kk] = global_read16(k + k_base + (long)kr * rk + kk * 32 +
+                             kgroup * 8);
+      va[kk] = global_read16(v + k_base + (long)kr * rk + kk * 32 +
+                             kgroup * 8);
     }
   }
 
@@ -424,94 +450,102 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
     qb_end = min(qb_end, max_qrow / kBlockM + 1);
   }
 
-  for (int hq = hkv * gqa; hq < (hkv + 1) * gqa; ++hq) {
-    const long q_base = ((long)b * Sq * Hq + hq) * D;
+  const int nqb = qb_end - qb_start;
+  const int iters = gqa * (nqb > 0 ? nqb : 0);
+
+  auto stage_iter = [&](int it, int buf) {
+    int hq = hkv * gqa + it / nqb;
+    int qbx = qb_start + it % nqb;
+    const long qb_base = ((long)b * Sq * Hq + hq) * D;
+    stage_tr_image<D, kBlockM>(do_img[buf], dout + qb_base, rq,
+                               qbx * kBlockM, Sq);
+    stage_tr_image<D, kBlockM>(q_img[buf], q + qb_base, rq, qbx * kBlockM,
+                               Sq);
+  };
+
+  if (iters > 0) {
+    stage_iter(0, 0);
+    __syncthreads();
+  }
+
+  for (int it = 0; it < iters; ++it) {
+    const int hq = hkv * gqa + it / nqb;
+    const int qbx = qb_start + it % nqb;
+    const int qstart = qbx * kBlockM;
+    const __hip_bfloat16* do_cur = do_img[it & 1];
+    const __hip_bfloat16* q_cur = q_img[it & 1];
     const float* lse_h = lse + ((long)b * Hq + hq) * Sq;
     const float* delta_h = delta + ((long)b * Hq + hq) * Sq;
 
-    for (int qb = qb_start; qb < qb_end; ++qb) {
-      const int qstart = qb * kBlockM;
-
-      stage_tr_image<D, kBlockM>(do_img, dout + q_base, rq, qstart, Sq);
-      stage_tr_image<D, kBlockM>(q_img, q + q_base, rq, qstart, Sq);
-      __syncthreads();
-
-      frag_f32 stt[4], dpt[4];
-#pragma unroll
-      for (int t = 0; t < 4; ++t) {
-        stt[t] = frag_f32{0.f, 0.f, 0.f, 0.f};
-        dpt[t] = frag_f32{0.f, 0.f, 0.f, 0.f};
-      }
-#pragma unroll
-      for (int t = 0; t < 4; ++t) {
-        int qrow = qstart + t * 16 + row_in_tile;
-        int qr = qrow < Sq ? qrow : Sq - 1;
-#pragma unroll
-        for (int kk = 0; kk < KFRAGS; ++kk) {
-          frag_b16 qbf = load_frag_global(q + q_base + (long)qr * rq +
-                                          kk * 32 + kgroup * 8);
-          stt[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ka[kk], qbf,
-                                                           stt[t], 0, 0, 0);
-          frag_b16 dbf = load_frag_global(dout + q_base + (long)qr * rq +
-                                          kk * 32 + kgroup * 8);
-          dpt[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(va[kk], dbf,
-                                                           dpt[t], 0, 0, 0);
-        }
-      }
-
-#pragma unroll
-      for (int t = 0; t < 4; ++t) {
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          int key = kb * kBlockN + wave * 16 + kgroup * 4 + r;
-          int qrow = qstart + t * 16 + row_in_tile;
-          bool masked = (key >= Sk) || (qrow >= Sq);
-          if (causal && key > qrow + skq) masked = true;
-          if (window > 0 && key < qrow + skq - window + 1) masked = true;
-          float pt = 0.f, dst = 0.f;
-          if (!masked) {
-            float l = lse_h[qrow];
-            pt = __expf(stt[t][r] * scale - l);
-            dst = pt * (dpt[t][r] - delta_h[qrow]) * scale;
-          }
-          int lrow = kgroup * 4 + r;
-          int lcol = t * 16 + row_in_tile;
-          pt_lds[wave][lrow * kStrip + lcol] = __float2bfloat16(pt);
-          dst_lds[wave][lrow * kStrip + lcol] = __float2bfloat16(dst);
-        }
-      }
-      __syncthreads();
-
-      // dV += P^T dO ; dK += dS^T Q
-#pragma unroll
-      for (int kk2 = 0; kk2 < 2; ++kk2) {
-        frag_b16 ptf, dstf;
-        {
-          const __hip_bfloat16* src1 =
-              &pt_lds[wave][row_in_tile * kStrip + kk2 * 32 + kgroup * 8];
-          const __hip_bfloat16* src2 =
-              &dst_lds[wave][row_in_tile * kStrip + kk2 * 32 + kgroup * 8];
-          union {
-            uint4 u;
-            frag_b16 f;
-          } c1, c2;
-          c1.u = *reinterpret_cast<const uint4*>(src1);
-          c2.u = *reinterpret_cast<const uint4*>(src2);
-          ptf = c1.f;
-          dstf = c2.f;
-        }
-#pragma unroll
-        for (int t = 0; t < DTILES; ++t) {
-          frag_b16 dof = tr_bfrag<D>(do_img, kk2 * 32, t, lane);
-          dv_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              ptf, dof, dv_acc[t], 0, 0, 0);
-          frag_b16 qtf = tr_bfrag<D>(q_img, kk2 * 32, t, lane);
-          dk_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              dstf, qtf, dk_acc[t], 0, 0, 0);
-        }
-      }
-      __syncthreads();
+    if (it + 1 < iters) {
+      stage_iter(it + 1, (it + 1) & 1);
     }
+
+    // S^T = K Q^T ; dP^T = V dO^T : B-fragments are Q/dO d-runs from LDS
+    frag_f32 stt[4], dpt[4];
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+      stt[t] = frag_f32{0.f, 0.f, 0.f, 0.f};
+      dpt[t] = frag_f32{0.f, 0.f, 0.f, 0.f};
+    }
+#pragma unroll
+    for (int kk = 0; kk < KFRAGS; ++kk) {
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        frag_b16 qbf = img_dfrag<D>(q_cur, t * 16 + row_in_tile,
+                                    kk * 32 + kgroup * 8);
+        stt[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ka[kk], qbf, stt[t],
+                                                         0, 0, 0);
+        frag_b16 dbf = img_dfrag<D>(do_cur, t * 16 + row_in_tile,
+                                    kk * 32 + kgroup * 8);
+        dpt[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(va[kk], dbf, dpt[t],
+                                                         0, 0, 0);
+      }
+    }
+
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int key = kb * kBlockN + wave * 16 + kgroup * 4 + r;
+        int qrow = qstart + t * 16 + row_in_tile;
+        bool masked = (key >= Sk) || (qrow >= Sq);
+        if (causal && key > qrow + skq) masked = true;
+        if (window > 0 && key < qrow + skq - window + 1) masked = true;
+        float pt = 0.f, dst = 0.f;
+        if (!masked) {
+          float l = lse_h[qrow];
+          pt = __expf(stt[t][r] * scale - l);
+          dst = pt * (dpt[t][r] - delta_h[qrow]) * scale;
+        }
+        int lrow = kgroup * 4 + r;
+        int lcol = t * 16 + row_in_tile;
+        pt_lds[wave][lrow * kStrip + lcol] = __float2bfloat16(pt);
+        dst_lds[wave][lrow * kStrip + lcol] = __float2bfloat16(dst);
+      }
+    }
+    // strips are wave-private: no barrier
+
+    // dV += P^T dO ; dK += dS^T Q  (B = dO/Q row-runs: transpose reads)
+#pragma unroll
+    for (int kk2 = 0; kk2 < 2; ++kk2) {
+      frag_b16 ptf = lds_read16(
+          &pt_lds[wave][row_in_tile * kStrip + kk2 * 32 + kgroup * 8]);
+      frag_b16 dstf = lds_read16(
+          &dst_lds[wave][row_in_tile * kStrip + kk2 * 32 + kgroup * 8]);
+#pragma unroll
+      for (int t = 0; t < DTILES; ++t) {
+        frag_b16 dof = tr_bfrag<D>(do_cur, kk2 * 32, t, lane);
+        dv_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ptf, dof,
+                                                            dv_acc[t], 0, 0,
+                                                            0);
+        frag_b16 qtf = tr_bfrag<D>(q_cur, kk2 * 32, t, lane);
+        dk_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dstf, qtf,
+                                                            dk_acc[t], 0, 0,
+                                                            0);
+      }
+    }
+    __syncthreads();
   }
 
 #pragma unroll
@@ -541,9 +575,11 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dq_kernel(
     int Sq, int Sk, int Hq, int Hkv, float scale, int causal, int window) {
   constexpr int KFRAGS = D / 32;
   constexpr int DTILES = D / 16;
+  constexpr int IMG = tr_elems<D>(kBlockN);
 
   __shared__ __hip_bfloat16 ds_lds[4][16 * kStrip];
-  __shared__ __hip_bfloat16 k_img[2][kBlockN * D];
+  __shared__ __hip_bfloat16 k_img[2][IMG];
+  __shared__ __hip_bfloat16 v_img[2][IMG];
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -571,10 +607,10 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dq_kernel(
     int qr = qrow < Sq ? qrow : Sq - 1;
 #pragma unroll
     for (int kk = 0; kk < KFRAGS; ++kk) {
-      qf[kk] = load_frag_global(q + q_base + (long)qr * rq + kk * 32 +
-                                kgroup * 8);
-      dof[kk] = load_frag_global(dout + q_base + (long)qr * rq + kk * 32 +
-                                 kgroup * 8);
+      qf[kk] = global_read16(q + q_base + (long)qr * rq + kk * 32 +
+                             kgroup * 8);
+      dof[kk] = global_read16(dout + q_base + (long)qr * rq + kk * 32 +
+                              kgroup * 8);
     }
   }
 
@@ -603,16 +639,22 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dq_kernel(
 
   stage_tr_image<D, kBlockN>(k_img[kb_start & 1], k + k_base, rk,
                              kb_start * kBlockN, Sk);
+  stage_tr_image<D, kBlockN>(v_img[kb_start & 1], v + k_base, rk,
+                             kb_start * kBlockN, Sk);
   __syncthreads();
 
   for (int kb = kb_start; kb < kb_end; ++kb) {
     const int kstart = kb * kBlockN;
     const __hip_bfloat16* k_cur = k_img[kb & 1];
+    const __hip_bfloat16* v_cur = v_img[kb & 1];
     if (kb + 1 < kb_end) {
       stage_tr_image<D, kBlockN>(k_img[(kb + 1) & 1], k + k_base, rk,
                                  (kb + 1) * kBlockN, Sk);
+      stage_tr_image<D, kBlockN>(v_img[(kb + 1) & 1], v + k_base, rk,
+                                 (kb + 1) * kBlockN, Sk);
     }
 
+    // S = Q K^T ; dP = dO V^T : B-fragments are K/V d-runs from LDS
     frag_f32 st[4], dp[4];
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
@@ -620,17 +662,15 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dq_kernel(
       dp[t] = frag_f32{0.f, 0.f, 0.f, 0.f};
     }
 #pragma unroll
-    for (int t = 0; t < 4; ++t) {
-      int key = kstart + t * 16 + row_in_tile;
-      int kr = key < Sk ? key : Sk - 1;
+    for (int kk = 0; kk < KFRAGS; ++kk) {
 #pragma unroll
-      for (int kk = 0; kk < KFRAGS; ++kk) {
-        frag_b16 kbf = load_frag_global(k + k_base + (long)kr * rk + kk * 32 +
-                                        kgroup * 8);
+      for (int t = 0; t < 4; ++t) {
+        frag_b16 kbf = img_dfrag<D>(k_cur, t * 16 + row_in_tile,
+                                    kk * 32 + kgroup * 8);
         st[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[kk], kbf, st[t],
                                                         0, 0, 0);
-        frag_b16 vbf = load_frag_global(v + k_base + (long)kr * rk + kk * 32 +
-                                        kgroup * 8);
+        frag_b16 vbf = img_dfrag<D>(v_cur, t * 16 + row_in_tile,
+                                    kk * 32 + kgroup * 8);
         dp[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dof[kk], vbf, dp[t],
                                                         0, 0, 0);
       }
@@ -657,20 +697,11 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dq_kernel(
     }
     // wave-private strip: no barrier needed
 
-    // dQ += dS K
+    // dQ += dS K  (B = K row-runs: transpose reads)
 #pragma unroll
     for (int kk2 = 0; kk2 < 2; ++kk2) {
-      frag_b16 dsf;
-      {
-        const __hip_bfloat16* src =
-            &ds_lds[wave][row_in_tile * kStrip + kk2 * 32 + kgroup * 8];
-        union {
-          uint4 u;
-          frag_b16 f;
-        } cvt;
-        cvt.u = *reinterpret_cast<const uint4*>(src);
-        dsf = cvt.f;
-      }
+      frag_b16 dsf = lds_read16(
+          &ds_lds[wave][row_in_tile * kStrip + kk2 * 32 + kgroup * 8]);
 #pragma unroll
       for (int t = 0; t < DTILES; ++t) {
         frag_b16 ktf = tr_bfrag<D>(k_cur, kk2 * 32, t, lane);
