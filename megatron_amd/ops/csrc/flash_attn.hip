@@ -144,12 +144,12 @@ __device__ __forceinline__ frag_b16 tr_bfrag(const __hip_bfloat16* img,
 }
 
 // cooperative stage of a [rows x D] bf16 tile from global into a tr image
-template <int D, int ROWS>
+template <int D, int ROWS, int NT = kThreads>
 __device__ __forceinline__ void stage_tr_image(
     __hip_bfloat16* img, const __hip_bfloat16* src_base, long row_stride,
     int first_row, int max_row) {
   const int vec_per_row = D / 8;
-  for (int idx = threadIdx.x; idx < ROWS * vec_per_row; idx += kThreads) {
+  for (int idx = threadIdx.x; idx < ROWS * vec_per_row; idx += NT) {
     int r = idx / vec_per_row;
     int c8 = (idx % vec_per_row) * 8;
     int row = first_row + r;
@@ -167,8 +167,8 @@ __device__ __forceinline__ void stage_tr_image(
 // ---------------------------------------------------------------------------
 // forward
 
-template <int D>
-__global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
+template <int D, int NW>
+__global__ __launch_bounds__(NW * 64) void fa_fwd_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ out,
     float* __restrict__ lse, int B, int Sq, int Sk, int Hq, int Hkv,
@@ -176,8 +176,10 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
   constexpr int KFRAGS = D / 32;
   constexpr int DTILES = D / 16;
   constexpr int IMG = tr_elems<D>(kBlockN);
+  constexpr int BM = NW * 16;       // q rows per workgroup
+  constexpr int NT = NW * 64;       // threads
 
-  __shared__ __hip_bfloat16 p_lds[4][16 * kStrip];
+  __shared__ __hip_bfloat16 p_lds[NW][16 * kStrip];
   __shared__ __hip_bfloat16 k_img[2][IMG];
   __shared__ __hip_bfloat16 v_img[2][IMG];
 
@@ -198,7 +200,7 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
   const int rq = Hq * D;
   const int rk = Hkv * D;
 
-  const int qrow0 = qb * kBlockM + wave * 16;
+  const int qrow0 = qb * BM + wave * 16;
   const int skq = Sk - Sq;
 
   frag_b16 qf[KFRAGS];
@@ -224,21 +226,21 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
 
   int kb_end = (Sk + kBlockN - 1) / kBlockN;
   if (causal) {
-    int max_qrow = qb * kBlockM + kBlockM - 1;
+    int max_qrow = qb * BM + BM - 1;
     int max_key = max_qrow + skq;
     kb_end = min(kb_end, max_key / kBlockN + 1);
   }
   int kb_start = 0;
   if (window > 0) {
-    int min_qrow = qb * kBlockM;
+    int min_qrow = qb * BM;
     int min_key = min_qrow + skq - window + 1;
     if (min_key > 0) kb_start = min_key / kBlockN;
   }
 
-  stage_tr_image<D, kBlockN>(k_img[kb_start & 1], k + k_base, rk,
-                             kb_start * kBlockN, Sk);
-  stage_tr_image<D, kBlockN>(v_img[kb_start & 1], v + v_base, rk,
-                             kb_start * kBlockN, Sk);
+  stage_tr_image<D, kBlockN, NT>(k_img[kb_start & 1], k + k_base, rk,
+                                 kb_start * kBlockN, Sk);
+  stage_tr_image<D, kBlockN, NT>(v_img[kb_start & 1], v + v_base, rk,
+                                 kb_start * kBlockN, Sk);
   __syncthreads();
 
   for (int kb = kb_start; kb < kb_end; ++kb) {
@@ -248,10 +250,10 @@ __global__ __launch_bounds__(kThreads) void fa_fwd_kernel(
     // issue next tile's staging loads; they land behind this iteration's
     // compute and are published by the single end-of-iteration barrier
     if (kb + 1 < kb_end) {
-      stage_tr_image<D, kBlockN>(k_img[(kb + 1) & 1], k + k_base, rk,
-                                 (kb + 1) * kBlockN, Sk);
-      stage_tr_image<D, kBlockN>(v_img[(kb + 1) & 1], v + v_base, rk,
-                                 (kb + 1) * kBlockN, Sk);
+      stage_tr_image<D, kBlockN, NT>(k_img[(kb + 1) & 1], k + k_base, rk,
+                                     (kb + 1) * kBlockN, Sk);
+      stage_tr_image<D, kBlockN, NT>(v_img[(kb + 1) & 1], v + v_base, rk,
+                                     (kb + 1) * kBlockN, Sk);
     }
 
     // S = Q K^T : B-fragments = K d-runs from LDS
@@ -744,26 +746,21 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
 
   auto out = torch::empty_like(q);
   auto lse = torch::empty({B, Hq, Sq}, q.options().dtype(torch::kFloat32));
-  dim3 grid((Sq + kBlockM - 1) / kBlockM, Hq, B);
   auto stream = c10::hip::getCurrentHIPStream();
   int win = window_size > 0 ? (int)window_size : 0;
+  static const int nw_env = []() {
+    const char* e = getenv("MEGATRON_AMD_FA_FWD_WAVES");
+    return e ? atoi(e) : 12;
+  }();
+#define LAUNCH_FWD(DD, NW)                                                      do {                                                                            dim3 grid((Sq + NW * 16 - 1) / (NW * 16), Hq, B);                             hipLaunchKernelGGL((fa_fwd_kernel<DD, NW>), grid, dim3(NW * 64), 0,                              stream, (const __hip_bfloat16*)q.data_ptr(),                                  (const __hip_bfloat16*)k.data_ptr(),                                          (const __hip_bfloat16*)v.data_ptr(),                                          (__hip_bfloat16*)out.data_ptr(),                                              lse.data_ptr<float>(), B, Sq, Sk, Hq, Hkv,                                    (float)softmax_scale, causal ? 1 : 0, win);              } while (0)
   if (D == 128) {
-    hipLaunchKernelGGL((fa_fwd_kernel<128>), grid, dim3(kThreads), 0, stream,
-                       (const __hip_bfloat16*)q.data_ptr(),
-                       (const __hip_bfloat16*)k.data_ptr(),
-                       (const __hip_bfloat16*)v.data_ptr(),
-                       (__hip_bfloat16*)out.data_ptr(),
-                       lse.data_ptr<float>(), B, Sq, Sk, Hq, Hkv,
-                       (float)softmax_scale, causal ? 1 : 0, win);
+    if (nw_env >= 12) LAUNCH_FWD(128, 12);
+    else LAUNCH_FWD(128, 4);
   } else {
-    hipLaunchKernelGGL((fa_fwd_kernel<64>), grid, dim3(kThreads), 0, stream,
-                       (const __hip_bfloat16*)q.data_ptr(),
-                       (const __hip_bfloat16*)k.data_ptr(),
-                       (const __hip_bfloat16*)v.data_ptr(),
-                       (__hip_bfloat16*)out.data_ptr(),
-                       lse.data_ptr<float>(), B, Sq, Sk, Hq, Hkv,
-                       (float)softmax_scale, causal ? 1 : 0, win);
+    if (nw_env >= 12) LAUNCH_FWD(64, 12);
+    else LAUNCH_FWD(64, 4);
   }
+#undef LAUNCH_FWD
   return {out, lse};
 }
 

@@ -212,13 +212,26 @@ __global__ void norm_bwd_kernel_bf16v(const __hip_bfloat16* __restrict__ dy,
     __syncthreads();
   }
 
+  // per-block partial rows (no atomics: fp32 atomic contention on H hot
+  // addresses dominated the kernel) — reduced by norm_bwd_reduce_kernel
   for (int i = threadIdx.x, c = 0; i < HV; i += BLOCK, ++c) {
 #pragma unroll
     for (int k = 0; k < 8; ++k) {
-      atomicAdd(&dweight[i * 8 + k], acc_dw[c][k]);
-      if (!RMS) atomicAdd(&dbias[i * 8 + k], acc_db[c][k]);
+      dweight[(long)blockIdx.x * H + i * 8 + k] = acc_dw[c][k];
+      if (!RMS) dbias[(long)blockIdx.x * H + i * 8 + k] = acc_db[c][k];
     }
   }
+}
+
+// column-sum of the [grid][H] partial buffers
+__global__ void norm_bwd_reduce_kernel(const float* __restrict__ partial,
+                                       float* __restrict__ out, int nblocks,
+                                       int H) {
+  int i = blockIdx.x * 256 + threadIdx.x;
+  if (i >= H) return;
+  float acc = 0.f;
+  for (int b = 0; b < nblocks; ++b) acc += partial[(long)b * H + i];
+  out[i] = acc;
 }
 
 // Backward. MAX_ACC register accumulators per thread for dweight/dbias.
@@ -352,6 +365,11 @@ void norm_bwd_launch(const torch::Tensor& dy, const torch::Tensor& x,
   auto stream = c10::hip::getCurrentHIPStream();
   if constexpr (std::is_same_v<T, __hip_bfloat16>) {
     if (H % 8 == 0 && H <= kBlock * 8 * 4) {
+      grid = (int)std::min<long>(rows, 608);  // ~2-3 blocks per CU worth
+      auto opts = dy.options().dtype(torch::kFloat32);
+      auto dw_part = torch::empty({grid, H}, opts);
+      torch::Tensor db_part;
+      if (!RMS) db_part = torch::empty({grid, H}, opts);
       hipLaunchKernelGGL((norm_bwd_kernel_bf16v<RMS, kBlock, 4>), dim3(grid),
                          dim3(kBlock), 0, stream,
                          (const __hip_bfloat16*)dy.data_ptr(),
@@ -360,8 +378,16 @@ void norm_bwd_launch(const torch::Tensor& dy, const torch::Tensor& x,
                          RMS ? nullptr : mean->data_ptr<float>(),
                          inv.data_ptr<float>(),
                          (__hip_bfloat16*)dx.data_ptr(),
-                         dw.data_ptr<float>(),
-                         RMS ? nullptr : db->data_ptr<float>(), rows, H);
+                         dw_part.data_ptr<float>(),
+                         RMS ? nullptr : db_part.data_ptr<float>(), rows, H);
+      hipLaunchKernelGGL(norm_bwd_reduce_kernel, dim3((H + 255) / 256),
+                         dim3(256), 0, stream, dw_part.data_ptr<float>(),
+                         dw.data_ptr<float>(), grid, H);
+      if (!RMS) {
+        hipLaunchKernelGGL(norm_bwd_reduce_kernel, dim3((H + 255) / 256),
+                           dim3(256), 0, stream, db_part.data_ptr<float>(),
+                           db->data_ptr<float>(), grid, H);
+      }
       return;
     }
   }
