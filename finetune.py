@@ -41,12 +41,17 @@ def model_provider(pre_process=True, post_process=True):
 
 
 def get_batch(data_iterator):
-    """(reference finetune.py:103-166)"""
+    """(reference finetune.py:103-166). For instruction data the loss covers
+    only assistant tokens, with --scalar_loss_mask weighting the rest."""
     cfg = get_config()
-    keys = ["text"]
+    instruction = cfg.model_type == "instruction"
+    keys = ["text"] if not instruction else ["text", "assistant_mask",
+                                             "pad_mask"]
     datatype = torch.int64
     if data_iterator is not None:
         data = next(data_iterator)
+        if instruction and data is not None:
+            data = {k: v.long() for k, v in data.items() if k in keys}
     else:
         data = None
     data_b = broadcast_data(keys, data, datatype)
@@ -63,6 +68,11 @@ def get_batch(data_iterator):
         tokens, tokenizer_eod, cfg.reset_position_ids,
         cfg.reset_attention_mask, cfg.eod_mask_loss,
     )
+    if instruction:
+        # label at position i is token i+1 -> use its role/pad bits
+        assistant = data_b["assistant_mask"][:, 1:].contiguous().float()
+        pad = data_b["pad_mask"][:, 1:].contiguous().float()
+        loss_mask = assistant + cfg.scalar_loss_mask * (pad - assistant)
     return tokens, labels, loss_mask, attention_mask, position_ids
 
 
