@@ -385,8 +385,8 @@ __global__ void fa_bwd_delta_kernel(const __hip_bfloat16* __restrict__ dout,
 // backward dK/dV (k-parallel; flattened (gqa head, q-block) loop with
 // double-buffered dO/Q images)
 
-template <int D>
-__global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
+template <int D, int NW>
+__global__ __launch_bounds__(NW * 64) void fa_bwd_dkdv_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v,
     const __hip_bfloat16* __restrict__ dout, const float* __restrict__ lse,
@@ -396,9 +396,11 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
   constexpr int KFRAGS = D / 32;
   constexpr int DTILES = D / 16;
   constexpr int IMG = tr_elems<D>(kBlockM);
+  constexpr int BN = NW * 16;  // keys per workgroup
+  constexpr int NT = NW * 64;
 
-  __shared__ __hip_bfloat16 pt_lds[4][16 * kStrip];
-  __shared__ __hip_bfloat16 dst_lds[4][16 * kStrip];
+  __shared__ __hip_bfloat16 pt_lds[NW][16 * kStrip];
+  __shared__ __hip_bfloat16 dst_lds[NW][16 * kStrip];
   __shared__ __hip_bfloat16 do_img[2][IMG];
   __shared__ __hip_bfloat16 q_img[2][IMG];
 
@@ -417,7 +419,7 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
   const int rq = Hq * D;
   const int skq = Sk - Sq;
 
-  const int key0 = kb * kBlockN + wave * 16;
+  const int key0 = kb * BN + wave * 16;
 
   frag_b16 ka[KFRAGS], va[KFRAGS];
   {
@@ -441,13 +443,13 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
 
   int qb_start = 0;
   if (causal) {
-    int min_key = kb * kBlockN;
+    int min_key = kb * BN;
     int min_qrow = min_key - skq;
     if (min_qrow > 0) qb_start = min_qrow / kBlockM;
   }
   int qb_end = (Sq + kBlockM - 1) / kBlockM;
   if (window > 0) {
-    int max_key = kb * kBlockN + kBlockN - 1;
+    int max_key = kb * BN + BN - 1;
     int max_qrow = max_key + window - 1 - skq;
     qb_end = min(qb_end, max_qrow / kBlockM + 1);
   }
@@ -459,10 +461,10 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
     int hq = hkv * gqa + it / nqb;
     int qbx = qb_start + it % nqb;
     const long qb_base = ((long)b * Sq * Hq + hq) * D;
-    stage_tr_image<D, kBlockM>(do_img[buf], dout + qb_base, rq,
-                               qbx * kBlockM, Sq);
-    stage_tr_image<D, kBlockM>(q_img[buf], q + qb_base, rq, qbx * kBlockM,
-                               Sq);
+    stage_tr_image<D, kBlockM, NT>(do_img[buf], dout + qb_base, rq,
+                                   qbx * kBlockM, Sq);
+    stage_tr_image<D, kBlockM, NT>(q_img[buf], q + qb_base, rq,
+                                   qbx * kBlockM, Sq);
   };
 
   if (iters > 0) {
@@ -509,7 +511,7 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
     for (int t = 0; t < 4; ++t) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        int key = kb * kBlockN + wave * 16 + kgroup * 4 + r;
+        int key = kb * BN + wave * 16 + kgroup * 4 + r;
         int qrow = qstart + t * 16 + row_in_tile;
         bool masked = (key >= Sk) || (qrow >= Sq);
         if (causal && key > qrow + skq) masked = true;
@@ -552,7 +554,7 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
 
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
-    int key = kb * kBlockN + wave * 16 + kgroup * 4 + r;
+    int key = kb * BN + wave * 16 + kgroup * 4 + r;
     if (key < Sk) {
 #pragma unroll
       for (int t = 0; t < DTILES; ++t) {
@@ -568,8 +570,8 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dkdv_kernel(
 // ---------------------------------------------------------------------------
 // backward dQ (q-parallel)
 
-template <int D>
-__global__ __launch_bounds__(kThreads) void fa_bwd_dq_kernel(
+template <int D, int NW>
+__global__ __launch_bounds__(NW * 64) void fa_bwd_dq_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v,
     const __hip_bfloat16* __restrict__ dout, const float* __restrict__ lse,
@@ -578,8 +580,10 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dq_kernel(
   constexpr int KFRAGS = D / 32;
   constexpr int DTILES = D / 16;
   constexpr int IMG = tr_elems<D>(kBlockN);
+  constexpr int BM = NW * 16;
+  constexpr int NT = NW * 64;
 
-  __shared__ __hip_bfloat16 ds_lds[4][16 * kStrip];
+  __shared__ __hip_bfloat16 ds_lds[NW][16 * kStrip];
   __shared__ __hip_bfloat16 k_img[2][IMG];
   __shared__ __hip_bfloat16 v_img[2][IMG];
 
@@ -598,7 +602,7 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dq_kernel(
   const int rq = Hq * D;
   const int rk = Hkv * D;
   const int skq = Sk - Sq;
-  const int qrow0 = qb * kBlockM + wave * 16;
+  const int qrow0 = qb * BM + wave * 16;
 
   const float* lse_h = lse + ((long)b * Hq + h) * Sq;
   const float* delta_h = delta + ((long)b * Hq + h) * Sq;
@@ -622,12 +626,12 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dq_kernel(
 
   int kb_end = (Sk + kBlockN - 1) / kBlockN;
   if (causal) {
-    int max_qrow = qb * kBlockM + kBlockM - 1;
+    int max_qrow = qb * BM + BM - 1;
     kb_end = min(kb_end, (max_qrow + skq) / kBlockN + 1);
   }
   int kb_start = 0;
   if (window > 0) {
-    int min_key = qb * kBlockM + skq - window + 1;
+    int min_key = qb * BM + skq - window + 1;
     if (min_key > 0) kb_start = min_key / kBlockN;
   }
 
@@ -639,10 +643,10 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dq_kernel(
     delta_r[r] = row < Sq ? delta_h[row] : 0.f;
   }
 
-  stage_tr_image<D, kBlockN>(k_img[kb_start & 1], k + k_base, rk,
-                             kb_start * kBlockN, Sk);
-  stage_tr_image<D, kBlockN>(v_img[kb_start & 1], v + k_base, rk,
-                             kb_start * kBlockN, Sk);
+  stage_tr_image<D, kBlockN, NT>(k_img[kb_start & 1], k + k_base, rk,
+                                 kb_start * kBlockN, Sk);
+  stage_tr_image<D, kBlockN, NT>(v_img[kb_start & 1], v + k_base, rk,
+                                 kb_start * kBlockN, Sk);
   __syncthreads();
 
   for (int kb = kb_start; kb < kb_end; ++kb) {
@@ -650,10 +654,10 @@ __global__ __launch_bounds__(kThreads) void fa_bwd_dq_kernel(
     const __hip_bfloat16* k_cur = k_img[kb & 1];
     const __hip_bfloat16* v_cur = v_img[kb & 1];
     if (kb + 1 < kb_end) {
-      stage_tr_image<D, kBlockN>(k_img[(kb + 1) & 1], k + k_base, rk,
-                                 (kb + 1) * kBlockN, Sk);
-      stage_tr_image<D, kBlockN>(v_img[(kb + 1) & 1], v + k_base, rk,
-                                 (kb + 1) * kBlockN, Sk);
+      stage_tr_image<D, kBlockN, NT>(k_img[(kb + 1) & 1], k + k_base, rk,
+                                     (kb + 1) * kBlockN, Sk);
+      stage_tr_image<D, kBlockN, NT>(v_img[(kb + 1) & 1], v + k_base, rk,
+                                     (kb + 1) * kBlockN, Sk);
     }
 
     // S = Q K^T ; dP = dO V^T : B-fragments are K/V d-runs from LDS
@@ -787,8 +791,8 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
                        stream, (const __hip_bfloat16*)dout.data_ptr(),        \
                        (const __hip_bfloat16*)out.data_ptr(),                 \
                        delta.data_ptr<float>(), B, Sq, Hq);                   \
-    dim3 gridk((Sk + kBlockN - 1) / kBlockN, Hkv, B);                         \
-    hipLaunchKernelGGL((fa_bwd_dkdv_kernel<DD>), gridk, dim3(kThreads), 0,    \
+    dim3 gridk((Sk + 8 * 16 - 1) / (8 * 16), Hkv, B);                         \
+    hipLaunchKernelGGL((fa_bwd_dkdv_kernel<DD, 8>), gridk, dim3(8 * 64), 0,   \
                        stream, (const __hip_bfloat16*)q.data_ptr(),           \
                        (const __hip_bfloat16*)k.data_ptr(),                   \
                        (const __hip_bfloat16*)v.data_ptr(),                   \
@@ -797,8 +801,8 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
                        (__hip_bfloat16*)dk.data_ptr(),                        \
                        (__hip_bfloat16*)dv.data_ptr(), B, Sq, Sk, Hq, Hkv,    \
                        (float)softmax_scale, causal ? 1 : 0, win);            \
-    dim3 gridq((Sq + kBlockM - 1) / kBlockM, Hq, B);                          \
-    hipLaunchKernelGGL((fa_bwd_dq_kernel<DD>), gridq, dim3(kThreads), 0,      \
+    dim3 gridq((Sq + 12 * 16 - 1) / (12 * 16), Hq, B);                        \
+    hipLaunchKernelGGL((fa_bwd_dq_kernel<DD, 12>), gridq, dim3(12 * 64), 0,   \
                        stream, (const __hip_bfloat16*)q.data_ptr(),           \
                        (const __hip_bfloat16*)k.data_ptr(),                   \
                        (const __hip_bfloat16*)v.data_ptr(),                   \
