@@ -31,6 +31,14 @@ def build_tokenizer(cfg):
     elif cfg.tokenizer_type == "GPT2BPETokenizer":
         assert cfg.vocab_file is not None and cfg.merge_file is not None
         tokenizer = GPT2BPETokenizer(cfg.vocab_file, cfg.merge_file)
+    elif cfg.tokenizer_type in ("BertWordPieceLowerCase",
+                                "BertWordPieceCase"):
+        assert cfg.vocab_file is not None
+        tokenizer = BertWordPieceTokenizer(
+            cfg.vocab_file,
+            lower_case=(cfg.tokenizer_type == "BertWordPieceLowerCase"),
+            vocab_extra_ids=cfg.vocab_extra_ids,
+        )
     elif cfg.tokenizer_type == "FakeTokenizer":
         tokenizer = FakeTokenizer(cfg.padded_vocab_size or 32000)
     else:
@@ -399,3 +407,112 @@ class GPT2BPETokenizer(AbstractTokenizer):
     @property
     def eod(self):
         return self.eod_id
+
+
+class BertWordPieceTokenizer(AbstractTokenizer):
+    """Greedy longest-match-first WordPiece over a local vocab.txt
+    (reference tokenizer.py:123-252 wrapping the original BERT tokenizer)."""
+
+    def __init__(self, vocab_file, lower_case=True, vocab_extra_ids=0):
+        name = "BERT Lower Case" if lower_case else "BERT Upper Case"
+        super().__init__(name)
+        self.lower_case = lower_case
+        self._vocab = {}
+        with open(vocab_file, encoding="utf-8") as f:
+            for i, line in enumerate(f):
+                self._vocab[line.rstrip("\n")] = i
+        self._inv_vocab = {v: k for k, v in self._vocab.items()}
+        for tok in ("[PAD]", "[UNK]", "[CLS]", "[SEP]", "[MASK]"):
+            if tok not in self._vocab:
+                self._vocab[tok] = len(self._vocab)
+                self._inv_vocab[self._vocab[tok]] = tok
+        self._cls_id = self._vocab["[CLS]"]
+        self._sep_id = self._vocab["[SEP]"]
+        self._pad_id = self._vocab["[PAD]"]
+        self._mask_id = self._vocab["[MASK]"]
+        self._unk_id = self._vocab["[UNK]"]
+        self._additional = []
+        for i in range(vocab_extra_ids):
+            tok = f"<extra_id_{i}>"
+            self._vocab[tok] = len(self._vocab)
+            self._inv_vocab[self._vocab[tok]] = tok
+            self._additional.append(self._vocab[tok])
+
+    @property
+    def vocab_size(self):
+        return len(self._vocab)
+
+    @property
+    def vocab(self):
+        return self._vocab
+
+    @property
+    def inv_vocab(self):
+        return self._inv_vocab
+
+    def _wordpiece(self, word):
+        if word in self._vocab:
+            return [self._vocab[word]]
+        tokens = []
+        start = 0
+        while start < len(word):
+            end = len(word)
+            piece = None
+            while start < end:
+                sub = word[start:end]
+                if start > 0:
+                    sub = "##" + sub
+                if sub in self._vocab:
+                    piece = self._vocab[sub]
+                    break
+                end -= 1
+            if piece is None:
+                return [self._unk_id]
+            tokens.append(piece)
+            start = end
+        return tokens
+
+    def tokenize(self, text):
+        import re
+
+        if self.lower_case:
+            text = text.lower()
+        words = re.findall(r"\w+|[^\w\s]", text)
+        ids = []
+        for w in words:
+            ids.extend(self._wordpiece(w))
+        return ids
+
+    def detokenize(self, token_ids):
+        out = []
+        for t in token_ids:
+            piece = self._inv_vocab.get(int(t), "[UNK]")
+            if piece.startswith("##") and out:
+                out[-1] += piece[2:]
+            else:
+                out.append(piece)
+        return " ".join(out)
+
+    @property
+    def cls(self):
+        return self._cls_id
+
+    @property
+    def sep(self):
+        return self._sep_id
+
+    @property
+    def pad(self):
+        return self._pad_id
+
+    @property
+    def mask(self):
+        return self._mask_id
+
+    @property
+    def eod(self):
+        return self._sep_id
+
+    @property
+    def additional_special_tokens_ids(self):
+        return list(self._additional)
