@@ -17,6 +17,17 @@ import json
 import os
 import time
 
+# hipBLASLt TunableOp: use the pre-tuned GEMM algo selection committed under
+# profiles/ when present (tools/tune_gemms.py produces it); read-only mode.
+_TUNED = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                      "profiles", "tunableop_results00.csv")
+if os.path.exists(_TUNED) and "PYTORCH_TUNABLEOP_ENABLED" not in os.environ:
+    os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
+    os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"
+    os.environ["PYTORCH_TUNABLEOP_FILENAME"] = os.path.join(
+        os.path.dirname(_TUNED), "tunableop_results.csv"
+    )
+
 import torch
 
 
@@ -83,9 +94,9 @@ def main():
     pp = args.pp
     assert world_size % (tp * pp) == 0
     dp = world_size // (tp * pp)
-    # mbs sweep on MI355X (profiles/r01): 9.0k/10.4k/11.4k/12.0k tok/s at
-    # mbs 1/2/4/8 — default 4 balances throughput vs step latency
-    mbs = args.micro_batch_size or (4 if have_gpu else 1)
+    # mbs sweep on MI355X (profiles/r01): 15.5k @ mbs4, 16.7k @ mbs8 with
+    # the wide-workgroup FA kernels
+    mbs = args.micro_batch_size or (8 if have_gpu else 1)
     gbs = args.global_batch or (mbs * dp)
 
     dtype_flags = {}

@@ -47,12 +47,23 @@ class BlendableDataset(torch.utils.data.Dataset):
         assert num_datasets < 255
         self.dataset_index = np.zeros(self.size, dtype=np.uint8)
         self.dataset_sample_index = np.zeros(self.size, dtype=np.int64)
-        _build_blending_indices(
-            self.dataset_index, self.dataset_sample_index, weights,
-            num_datasets, self.size,
-            torch.distributed.get_rank() == 0
-            if torch.distributed.is_initialized() else True,
-        )
+        verbose = (torch.distributed.get_rank() == 0
+                   if torch.distributed.is_initialized() else True)
+        from ..ops import ext as _ext
+
+        mod = _ext.load(required=False)
+        if mod is not None and hasattr(mod, "build_blending_indices"):
+            di = torch.from_numpy(self.dataset_index)
+            dsi = torch.from_numpy(self.dataset_sample_index)
+            mod.build_blending_indices(
+                di, dsi, torch.from_numpy(weights), num_datasets, self.size,
+                verbose,
+            )
+        else:
+            _build_blending_indices(
+                self.dataset_index, self.dataset_sample_index, weights,
+                num_datasets, self.size, verbose,
+            )
         print_rank_0(
             f"> elapsed time for building blendable dataset indices: "
             f"{time.time() - start_time:.2f} (sec)"

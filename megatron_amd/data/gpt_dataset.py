@@ -1,10 +1,10 @@
 """GPT pretraining dataset: epoch-spanning sample extraction over concatenated
 documents, with cached doc/sample/shuffle index mappings.
 
-Reference: megatron/data/gpt_dataset.py:20-513. The reference builds
-sample_idx in C++ (helpers.cpp build_sample_idx) because its Python loop is
-slow; here the mapping is built with vectorized numpy (cumsum + searchsorted)
-— same output, no native code needed on this path.
+Reference: megatron/data/gpt_dataset.py:20-513. sample_idx is built by the
+native C++ helper (ops/csrc/data_helpers.cpp, the analog of the reference's
+helpers.cpp build_sample_idx) when the extension is built, with a vectorized
+numpy path (cumsum + searchsorted, same output) as the no-extension fallback.
 """
 
 from __future__ import annotations
@@ -328,7 +328,19 @@ def _build_index_mappings(name, data_prefix, documents, sizes, num_samples,
         np.save(doc_idx_filename, doc_idx, allow_pickle=True)
 
         num_samples_ = (num_epochs * tokens_per_epoch - 1) // seq_length
-        sample_idx = _build_sample_idx(sizes, doc_idx, seq_length, num_samples_)
+        from ..ops import ext as _ext
+
+        _mod = _ext.load(required=False)
+        if _mod is not None and hasattr(_mod, "build_sample_idx"):
+            sample_idx = _mod.build_sample_idx(
+                torch.from_numpy(np.ascontiguousarray(sizes, dtype=np.int32)),
+                torch.from_numpy(np.ascontiguousarray(doc_idx,
+                                                      dtype=np.int32)),
+                seq_length, num_samples_,
+            ).numpy()
+        else:
+            sample_idx = _build_sample_idx(sizes, doc_idx, seq_length,
+                                           num_samples_)
         np.save(sample_idx_filename, sample_idx, allow_pickle=True)
 
         if separate_last_epoch:

@@ -56,6 +56,14 @@ void fused_adam_with_model_copy(std::vector<torch::Tensor> params,
 void wgrad_gemm_accum_fp32(torch::Tensor input, torch::Tensor grad_output,
                            torch::Tensor main_grad);
 
+// data_helpers.cpp (CPU)
+torch::Tensor build_sample_idx(torch::Tensor sizes, torch::Tensor doc_idx,
+                               int64_t seq_length, int64_t num_samples);
+void build_blending_indices(torch::Tensor dataset_index,
+                            torch::Tensor dataset_sample_index,
+                            torch::Tensor weights, int64_t num_datasets,
+                            int64_t size, bool verbose);
+
 // flash_attn.hip
 std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
                                           torch::Tensor v, bool causal,
@@ -83,6 +91,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adam", &fused_adam);
   m.def("fused_adam_with_model_copy", &fused_adam_with_model_copy);
   m.def("wgrad_gemm_accum_fp32", &wgrad_gemm_accum_fp32);
+  m.def("build_sample_idx", &build_sample_idx);
+  m.def("build_blending_indices", &build_blending_indices);
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_bwd", &flash_attn_bwd);
 }
