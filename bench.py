@@ -97,7 +97,8 @@ def main():
     # mbs sweep on MI355X (profiles/r01): 15.5k @ mbs4, 16.7k @ mbs8 with
     # the wide-workgroup FA kernels
     mbs = args.micro_batch_size or (8 if have_gpu else 1)
-    gbs = args.global_batch or (mbs * dp)
+    # pp > 1 needs several in-flight microbatches to fill the 1F1B pipeline
+    gbs = args.global_batch or (mbs * dp * (2 * pp if pp > 1 else 1))
 
     dtype_flags = {}
     if args.dtype == "bf16" and have_gpu:

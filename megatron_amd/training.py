@@ -270,7 +270,7 @@ def training_log(loss_dict, total_loss_dict, learning_rate, iteration,
     """Console/TB logging (reference training.py:462-641, abbreviated)."""
     timers = global_state.get_timers()
     writer = global_state.get_tensorboard_writer()
-    wandb = global_state.get_wandb_writer()
+    wandb_writer = global_state.get_wandb_writer()
 
     advanced_iters_key = "advanced iterations"
     skipped_iters_key = "skipped iterations"
@@ -334,6 +334,11 @@ def training_log(loss_dict, total_loss_dict, learning_rate, iteration,
                 log_string += f" {key}: {avg:.6E} |"
                 if writer:
                     writer.add_scalar(key, avg, iteration)
+                    writer.add_scalar(
+                        key + " vs samples", avg, cfg.consumed_train_samples
+                    )
+                if wandb_writer:
+                    wandb_writer.add_scalar(key, avg, iteration)
                 total_loss_dict[key] = torch.tensor(
                     [0.0], device=total_loss_dict[key].device
                 )
@@ -350,6 +355,19 @@ def training_log(loss_dict, total_loss_dict, learning_rate, iteration,
         log_string += (
             f" number of nan iterations: {total_loss_dict[nan_iters_key]:3d} |"
         )
+        if writer:
+            writer.add_scalar("learning-rate", learning_rate, iteration)
+            writer.add_scalar("tokens-per-sec", tokens_per_sec, iteration)
+            if cfg.log_timers_to_tensorboard:
+                timers.write(
+                    ["forward-compute", "backward-compute", "optimizer"],
+                    writer, iteration, normalizer=cfg.log_interval,
+                )
+        if wandb_writer:
+            wandb_writer.add_scalar("learning-rate", learning_rate, iteration)
+            wandb_writer.add_scalar("tokens-per-sec", tokens_per_sec,
+                                    iteration)
+            wandb_writer.flush_all()
         total_loss_dict[advanced_iters_key] = 0
         total_loss_dict[skipped_iters_key] = 0
         total_loss_dict[nan_iters_key] = 0
