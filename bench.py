@@ -20,7 +20,7 @@ import time
 # hipBLASLt TunableOp: use the pre-tuned GEMM algo selection committed under
 # profiles/ when present (tools/tune_gemms.py produces it); read-only mode.
 _TUNED = os.path.join(os.path.dirname(os.path.abspath(__file__)),
-                      "profiles", "tunableop_results00.csv")
+                      "profiles", "tunableop_results0.csv")
 if os.path.exists(_TUNED) and "PYTORCH_TUNABLEOP_ENABLED" not in os.environ:
     os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
     os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"
