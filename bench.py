@@ -44,6 +44,9 @@ def parse_args():
     p.add_argument("--global-batch", type=int, default=None)
     p.add_argument("--tp", type=int, default=None)
     p.add_argument("--pp", type=int, default=1)
+    p.add_argument("--layers", type=int, default=None,
+                   help="override layer count (shape validation only; "
+                   "results with this flag are not official numbers)")
     p.add_argument("--recompute", action="store_true")
     p.add_argument("--dtype", type=str, default="bf16")
     return p.parse_args()
@@ -89,6 +92,8 @@ def main():
         spec = dict(MODEL_SPECS["llama2-tiny"])
         spec["model_name"] = MODEL_SPECS[args.model]["model_name"]
 
+    if args.layers:
+        spec["num_layers"] = args.layers
     seq = args.seq_len or spec["seq"]
     tp = args.tp if args.tp is not None else 1
     pp = args.pp
@@ -241,7 +246,8 @@ def main():
             "dtype": args.dtype if have_gpu else "fp32",
             "data": "synthetic",
             "config": {
-                "model": args.model if have_gpu else f"{args.model}(cpu-tiny)",
+                "model": (args.model if have_gpu else f"{args.model}(cpu-tiny)")
+                + (f"-L{args.layers}" if args.layers else ""),
                 "global_batch": gbs,
                 "seq_len": seq,
                 "parallelism": f"tp{tp}_pp{pp}_dp{dp}",
