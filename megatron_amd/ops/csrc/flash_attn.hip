@@ -485,30 +485,24 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dkdv_kernel(
       stage_iter(it + 1, (it + 1) & 1);
     }
 
-    // S^T = K Q^T ; dP^T = V dO^T : B-fragments are Q/dO d-runs from LDS
-    frag_f32 stt[4], dpt[4];
+    // S^T = K Q^T ; dP^T = V dO^T : B-fragments are Q/dO d-runs from LDS.
+    // One tile's fragment pair live at a time (VGPR pressure), softmax +
+    // strip write immediately after each tile's K-reduction.
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
-      stt[t] = frag_f32{0.f, 0.f, 0.f, 0.f};
-      dpt[t] = frag_f32{0.f, 0.f, 0.f, 0.f};
-    }
+      frag_f32 stt = frag_f32{0.f, 0.f, 0.f, 0.f};
+      frag_f32 dpt = frag_f32{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-    for (int kk = 0; kk < KFRAGS; ++kk) {
-#pragma unroll
-      for (int t = 0; t < 4; ++t) {
+      for (int kk = 0; kk < KFRAGS; ++kk) {
         frag_b16 qbf = img_dfrag<D>(q_cur, t * 16 + row_in_tile,
                                     kk * 32 + kgroup * 8);
-        stt[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ka[kk], qbf, stt[t],
-                                                         0, 0, 0);
+        stt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ka[kk], qbf, stt, 0, 0,
+                                                      0);
         frag_b16 dbf = img_dfrag<D>(do_cur, t * 16 + row_in_tile,
                                     kk * 32 + kgroup * 8);
-        dpt[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(va[kk], dbf, dpt[t],
-                                                         0, 0, 0);
+        dpt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(va[kk], dbf, dpt, 0, 0,
+                                                      0);
       }
-    }
-
-#pragma unroll
-    for (int t = 0; t < 4; ++t) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int key = kb * BN + wave * 16 + kgroup * 4 + r;
@@ -519,8 +513,8 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dkdv_kernel(
         float pt = 0.f, dst = 0.f;
         if (!masked) {
           float l = lse_h[qrow];
-          pt = __expf(stt[t][r] * scale - l);
-          dst = pt * (dpt[t][r] - delta_h[qrow]) * scale;
+          pt = __expf(stt[r] * scale - l);
+          dst = pt * (dpt[r] - delta_h[qrow]) * scale;
         }
         int lrow = kgroup * 4 + r;
         int lcol = t * 16 + row_in_tile;
@@ -791,8 +785,8 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
                        stream, (const __hip_bfloat16*)dout.data_ptr(),        \
                        (const __hip_bfloat16*)out.data_ptr(),                 \
                        delta.data_ptr<float>(), B, Sq, Hq);                   \
-    dim3 gridk((Sk + 8 * 16 - 1) / (8 * 16), Hkv, B);                         \
-    hipLaunchKernelGGL((fa_bwd_dkdv_kernel<DD, 8>), gridk, dim3(8 * 64), 0,   \
+    dim3 gridk((Sk + 12 * 16 - 1) / (12 * 16), Hkv, B);                       \
+    hipLaunchKernelGGL((fa_bwd_dkdv_kernel<DD, 12>), gridk, dim3(12 * 64), 0, \
                        stream, (const __hip_bfloat16*)q.data_ptr(),           \
                        (const __hip_bfloat16*)k.data_ptr(),                   \
                        (const __hip_bfloat16*)v.data_ptr(),                   \
