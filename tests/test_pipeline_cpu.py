@@ -138,3 +138,108 @@ def test_microbatch_calculator():
     assert r.get_current_global_batch_size() == 16
     r.update(2000, True)
     assert r.get() == 8
+
+
+WORLD4 = 4
+
+
+def _worker4(rank, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(WORLD4)
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo", rank=rank, world_size=WORLD4)
+    try:
+        _body_interleaved(rank)
+    finally:
+        dist.barrier()
+        from megatron_amd import parallel as mpu
+
+        mpu.destroy_model_parallel()
+        dist.destroy_process_group()
+
+
+def _body_interleaved(rank):
+    """pp=4 with 2 virtual chunks per stage: interleaved 1F1B trains."""
+    import functools
+
+    from megatron_amd import global_state
+    from megatron_amd import microbatches as mb
+    from megatron_amd import parallel as mpu
+    from megatron_amd.config import TrainingConfig, set_config
+    from megatron_amd.microbatches import setup_microbatch_calculator
+    from megatron_amd.models import LlamaModel, ModelType
+    from megatron_amd.optim import (
+        get_megatron_optimizer, get_optimizer_param_scheduler,
+    )
+    from megatron_amd.parallel.ddp import DistributedDataParallel as LocalDDP
+    from megatron_amd.training import train_step
+    from megatron_amd.utils import get_ltor_masks_and_position_ids
+
+    global_state.init_timers()
+    mpu.initialize_model_parallel(1, 4, virtual_pipeline_model_parallel_size=2)
+    mpu.model_parallel_cuda_manual_seed(1234)
+
+    cfg = TrainingConfig(
+        num_layers=8, hidden_size=64, num_attention_heads=4,
+        num_attention_heads_kv=2, seq_length=32, max_position_embeddings=64,
+        micro_batch_size=1, global_batch_size=4,
+        pipeline_model_parallel_size=4, world_size=4,
+        num_layers_per_virtual_pipeline_stage=1,
+        hidden_dropout=0.0, attention_dropout=0.0,
+        use_cpu_initialization=True, lr=1e-3, train_iters=4, clip_grad=1.0,
+    )
+    cfg.finalize()
+    assert cfg.virtual_pipeline_model_parallel_size == 2
+    cfg.pad_vocab_size(128)
+    set_config(cfg)
+    setup_microbatch_calculator(cfg)
+
+    models = []
+    for i in range(2):
+        mpu.set_virtual_pipeline_model_parallel_rank(i)
+        m = LlamaModel(
+            cfg,
+            pre_process=mpu.is_pipeline_first_stage(),
+            post_process=mpu.is_pipeline_last_stage(),
+        )
+        m.model_type = ModelType.encoder_or_decoder
+        models.append(LocalDDP(m, True, True))
+    optimizer = get_megatron_optimizer(models, cfg)
+    sched = get_optimizer_param_scheduler(optimizer, cfg)
+
+    torch.manual_seed(55)
+    batches = [torch.randint(0, 128, (1, 33)) for _ in range(32)]
+    its = [iter(batches[:16]), iter(batches[16:])]
+
+    def forward_step_func(data_iterator, model):
+        data = next(data_iterator)
+        tokens = data[:, :-1].contiguous()
+        labels = data[:, 1:].contiguous()
+        am, loss_mask, pids = get_ltor_masks_and_position_ids(
+            tokens, 0, False, False, False
+        )
+        output = model(tokens, pids, am, labels=labels)
+
+        def loss_func(loss_mask, output_tensor):
+            losses = output_tensor.float()
+            lm = loss_mask.view(-1).float()
+            loss = torch.sum(losses.view(-1) * lm) / lm.sum()
+            return loss, {"lm loss": loss.detach()}
+
+        return output, functools.partial(loss_func, loss_mask)
+
+    for step in range(2):
+        loss_dict, skipped, grad_norm, _ = train_step(
+            forward_step_func, its, models, optimizer, sched, cfg
+        )
+        assert skipped == 0
+    if mpu.is_pipeline_last_stage(ignore_virtual=True):
+        assert loss_dict["lm loss"].item() > 0
+        print("interleaved PP loss:", loss_dict["lm loss"].item(), flush=True)
+
+
+def test_pp4_interleaved_trains():
+    mp.spawn(_worker4, args=(29631,), nprocs=WORLD4, join=True)
