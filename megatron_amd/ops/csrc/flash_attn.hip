@@ -654,30 +654,24 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dq_kernel(
                                      (kb + 1) * kBlockN, Sk);
     }
 
-    // S = Q K^T ; dP = dO V^T : B-fragments are K/V d-runs from LDS
-    frag_f32 st[4], dp[4];
+    // S = Q K^T ; dP = dO V^T : B-fragments are K/V d-runs from LDS.
+    // Tile-at-a-time (one st/dp fragment pair live) for VGPR pressure —
+    // dS needs only the stored lse/delta, no cross-tile softmax state.
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
-      st[t] = frag_f32{0.f, 0.f, 0.f, 0.f};
-      dp[t] = frag_f32{0.f, 0.f, 0.f, 0.f};
-    }
+      frag_f32 st = frag_f32{0.f, 0.f, 0.f, 0.f};
+      frag_f32 dp = frag_f32{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-    for (int kk = 0; kk < KFRAGS; ++kk) {
-#pragma unroll
-      for (int t = 0; t < 4; ++t) {
+      for (int kk = 0; kk < KFRAGS; ++kk) {
         frag_b16 kbf = img_dfrag<D>(k_cur, t * 16 + row_in_tile,
                                     kk * 32 + kgroup * 8);
-        st[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[kk], kbf, st[t],
-                                                        0, 0, 0);
+        st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[kk], kbf, st, 0, 0,
+                                                     0);
         frag_b16 vbf = img_dfrag<D>(v_cur, t * 16 + row_in_tile,
                                     kk * 32 + kgroup * 8);
-        dp[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dof[kk], vbf, dp[t],
-                                                        0, 0, 0);
+        dp = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dof[kk], vbf, dp, 0, 0,
+                                                     0);
       }
-    }
-
-#pragma unroll
-    for (int t = 0; t < 4; ++t) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int col = kstart + t * 16 + row_in_tile;
@@ -687,8 +681,8 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dq_kernel(
         if (window > 0 && col < row + skq - window + 1) masked = true;
         float ds = 0.f;
         if (!masked) {
-          float p = __expf(st[t][r] * scale - lse_r[r]);
-          ds = p * (dp[t][r] - delta_r[r]) * scale;
+          float p = __expf(st[r] * scale - lse_r[r]);
+          ds = p * (dp[r] - delta_r[r]) * scale;
         }
         int lrow = kgroup * 4 + r;
         int lcol = t * 16 + row_in_tile;
@@ -795,8 +789,8 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
                        (__hip_bfloat16*)dk.data_ptr(),                        \
                        (__hip_bfloat16*)dv.data_ptr(), B, Sq, Sk, Hq, Hkv,    \
                        (float)softmax_scale, causal ? 1 : 0, win);            \
-    dim3 gridq((Sq + 12 * 16 - 1) / (12 * 16), Hq, B);                        \
-    hipLaunchKernelGGL((fa_bwd_dq_kernel<DD, 12>), gridq, dim3(12 * 64), 0,   \
+    dim3 gridq((Sq + 16 * 16 - 1) / (16 * 16), Hq, B);                        \
+    hipLaunchKernelGGL((fa_bwd_dq_kernel<DD, 16>), gridq, dim3(16 * 64), 0,   \
                        stream, (const __hip_bfloat16*)q.data_ptr(),           \
                        (const __hip_bfloat16*)k.data_ptr(),                   \
                        (const __hip_bfloat16*)v.data_ptr(),                   \
