@@ -130,3 +130,41 @@ def test_generation_gpu(gpu_cfg):
     )
     assert out.shape[1] == 32
     assert (out[:, 8:] < 1000).all()
+
+
+def test_bf16_logit_parity_vs_hf(dist_single):
+    """Whole-model bf16 forward (all HIP kernels on the hot path) vs
+    transformers' fp32 LlamaForCausalLM — documented bf16 tolerance 1e-1
+    (reference docs getting_started.md:154)."""
+    transformers = pytest.importorskip("transformers")
+    import sys
+
+    sys.path.append(os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    from tests.test_conversion_cpu import _convert_to_ours, _our_model
+    from megatron_amd.utils import get_ltor_masks_and_position_ids
+    from transformers import LlamaConfig, LlamaForCausalLM
+
+    big_cfg = LlamaConfig(
+        vocab_size=256, hidden_size=512, intermediate_size=1024,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, rms_norm_eps=1e-5,
+        tie_word_embeddings=False, attention_bias=False,
+    )
+    torch.manual_seed(7)
+    hf_model = LlamaForCausalLM(big_cfg).eval()
+    sd = _convert_to_ours(hf_model, big_cfg)
+    ours, cfg = _our_model(big_cfg, sd)
+    cfg.use_flash_attn = True
+
+    ours = ours.cuda().bfloat16()
+    # RMSNorm stats and RoPE tables are fp32 in-kernel
+    hf_gpu = hf_model.cuda().float()
+
+    tokens = torch.randint(0, 256, (2, 64), device="cuda")
+    am, _, pids = get_ltor_masks_and_position_ids(tokens, 0, False, False,
+                                                  False)
+    with torch.no_grad():
+        ours_logits = ours(tokens, pids, None).float()[:, :, :256]
+        hf_logits = hf_gpu(tokens).logits.float()
+    err = (ours_logits - hf_logits).abs().max().item()
+    assert err < 1e-1, f"bf16 logit error {err}"
