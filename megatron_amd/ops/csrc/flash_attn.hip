@@ -789,8 +789,8 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
                        (__hip_bfloat16*)dk.data_ptr(),                        \
                        (__hip_bfloat16*)dv.data_ptr(), B, Sq, Sk, Hq, Hkv,    \
                        (float)softmax_scale, causal ? 1 : 0, win);            \
-    dim3 gridq((Sq + 16 * 16 - 1) / (16 * 16), Hq, B);                        \
-    hipLaunchKernelGGL((fa_bwd_dq_kernel<DD, 16>), gridq, dim3(16 * 64), 0,   \
+    dim3 gridq((Sq + 12 * 16 - 1) / (12 * 16), Hq, B);                        \
+    hipLaunchKernelGGL((fa_bwd_dq_kernel<DD, 12>), gridq, dim3(12 * 64), 0,   \
                        stream, (const __hip_bfloat16*)q.data_ptr(),           \
                        (const __hip_bfloat16*)k.data_ptr(),                   \
                        (const __hip_bfloat16*)v.data_ptr(),                   \
