@@ -8,8 +8,6 @@ implemented on FastAPI/uvicorn which are available offline. The wire format
 is unchanged.
 """
 
-from __future__ import annotations
-
 import threading
 
 import torch

@@ -118,3 +118,28 @@ def test_beam_search_runs(small_model):
     )
     assert out_tokens.shape[0] == 2
     assert scores.shape[0] == 2
+
+
+def test_rest_server_contract(small_model):
+    """PUT /api with prompts returns text/segments (reference
+    text_generation_server.py wire format) — exercised in-process via the
+    FastAPI test client on the single-rank path."""
+    httpx = pytest.importorskip("httpx")
+    from fastapi.testclient import TestClient
+
+    from megatron_amd.inference.server import MegatronServer
+
+    m, cfg = small_model
+    server = MegatronServer(m)
+    app = server._build_app()
+    client = TestClient(app)
+
+    r = client.put("/api", json={"prompts": ["hello world"],
+                                 "tokens_to_generate": 4, "top_k": 1})
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert "text" in body and len(body["text"]) == 1
+    assert "segments" in body
+
+    r = client.put("/api", json={})
+    assert r.status_code == 400
