@@ -38,7 +38,7 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--model", type=str, default="llama2-7b",
                    choices=["llama2-7b", "llama2-70b", "mistral-7b",
-                            "gpt-125m", "llama2-tiny"])
+                            "falcon-7b", "gpt-125m", "llama2-tiny"])
     p.add_argument("--seq-len", type=int, default=None)
     p.add_argument("--micro-batch-size", type=int, default=None)
     p.add_argument("--global-batch", type=int, default=None)
@@ -66,6 +66,10 @@ MODEL_SPECS = {
                        num_attention_heads=32, num_attention_heads_kv=8,
                        vocab=32000, seq=32768, model_name="mistral",
                        sliding_window_size=4096, rope_scaling_factor=4.0),
+    # Falcon-7B: 32 layers, h=4544, 71 heads, MQA (1 kv head), parallel attn
+    "falcon-7b": dict(num_layers=32, hidden_size=4544, ffn_hidden_size=18176,
+                      num_attention_heads=71, num_attention_heads_kv=1,
+                      vocab=65024, seq=2048, model_name="falcon"),
     "gpt-125m": dict(num_layers=12, hidden_size=768, ffn_hidden_size=3072,
                      num_attention_heads=12, num_attention_heads_kv=12,
                      vocab=50304, seq=1024, model_name="gpt"),
@@ -87,7 +91,8 @@ def main():
 
     have_gpu = torch.cuda.is_available()
     spec = dict(MODEL_SPECS[args.model])
-    if not have_gpu and args.model in ("llama2-7b", "llama2-70b", "mistral-7b"):
+    if not have_gpu and args.model in ("llama2-7b", "llama2-70b",
+                                       "mistral-7b", "falcon-7b"):
         # CPU plumbing check shrinks the model but keeps the code path
         spec = dict(MODEL_SPECS["llama2-tiny"])
         spec["model_name"] = MODEL_SPECS[args.model]["model_name"]

@@ -168,3 +168,40 @@ def test_bf16_logit_parity_vs_hf(dist_single):
         hf_logits = hf_gpu(tokens).logits.float()
     err = (ours_logits - hf_logits).abs().max().item()
     assert err < 1e-1, f"bf16 logit error {err}"
+
+
+def test_determinism_bitwise(gpu_cfg):
+    """Two identical fwd+bwd passes produce bitwise-identical gradients —
+    guards against kernel races / nondeterministic accumulation (the
+    framework's kernels avoid atomics on the training path)."""
+    import functools
+
+    from megatron_amd.utils import get_ltor_masks_and_position_ids
+
+    cfg = gpu_cfg
+    torch.manual_seed(123)
+    torch.cuda.manual_seed(123)
+    model = _build(cfg)[0]
+    tokens = torch.randint(0, 1000, (2, 512), device="cuda")
+    am, loss_mask, pids = get_ltor_masks_and_position_ids(
+        tokens, 0, False, False, False
+    )
+
+    def run():
+        model.zero_grad_buffer()
+        out = model(tokens, pids, None, labels=tokens)
+        loss = out.float().mean()
+        loss.backward()
+        torch.cuda.synchronize()
+        grads = [
+            p.main_grad.clone()
+            for p in model.module.parameters()
+            if hasattr(p, "main_grad")
+        ]
+        return loss.item(), grads
+
+    l1, g1 = run()
+    l2, g2 = run()
+    assert l1 == l2
+    for a, b in zip(g1, g2):
+        assert torch.equal(a, b)
