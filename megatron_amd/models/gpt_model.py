@@ -118,7 +118,31 @@ class GPTModel(MegatronModule):
             # plain nn.Module state dict (flat keys)
             torch.nn.Module.load_state_dict(self, state_dict, strict=strict)
             return
+        state_dict = _flatten_nested_language_model(state_dict)
         self.language_model.load_state_dict(state_dict, strict=strict)
+
+
+def _flatten_nested_language_model(sd):
+    """Accept reference-style nested checkpoints (megatron/language_model.py
+    :556-580 stores {'embedding': {'word_embeddings': {'weight': ...}},
+    'encoder': {...layer-local keys...}, 'lm_head'/'output_layer': ...}) in
+    addition to this repo's flat key scheme."""
+    import torch as _torch
+
+    if not any(isinstance(v, dict) for v in sd.values()):
+        return sd
+    flat = {}
+
+    def rec(prefix, d):
+        for k, v in d.items():
+            key = f"{prefix}.{k}" if prefix else k
+            if isinstance(v, dict):
+                rec(key, v)
+            elif _torch.is_tensor(v):
+                flat[key] = v
+
+    rec("", sd)
+    return flat
 
 
 def _force(cfg, **kwargs):

@@ -365,7 +365,11 @@ void norm_bwd_launch(const torch::Tensor& dy, const torch::Tensor& x,
   auto stream = c10::hip::getCurrentHIPStream();
   if constexpr (std::is_same_v<T, __hip_bfloat16>) {
     if (H % 8 == 0 && H <= kBlock * 8 * 4) {
-      grid = (int)std::min<long>(rows, 608);  // ~2-3 blocks per CU worth
+      static const int grid_env = []() {
+        const char* e = getenv("MEGATRON_AMD_NORM_BWD_GRID");
+        return e ? atoi(e) : 1280;
+      }();
+      grid = (int)std::min<long>(rows, grid_env);
       auto opts = dy.options().dtype(torch::kFloat32);
       auto dw_part = torch::empty({grid, H}, opts);
       torch::Tensor db_part;
