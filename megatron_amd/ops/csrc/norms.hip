@@ -367,7 +367,7 @@ void norm_bwd_launch(const torch::Tensor& dy, const torch::Tensor& x,
     if (H % 8 == 0 && H <= kBlock * 8 * 4) {
       static const int grid_env = []() {
         const char* e = getenv("MEGATRON_AMD_NORM_BWD_GRID");
-        return e ? atoi(e) : 1280;
+        return e ? atoi(e) : 608;
       }();
       grid = (int)std::min<long>(rows, grid_env);
       auto opts = dy.options().dtype(torch::kFloat32);
