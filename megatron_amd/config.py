@@ -245,6 +245,12 @@ class TrainingConfig:
         )
         self.data_parallel_size = self.world_size // mp
 
+        # overlapped bucket all-reduce and the distributed optimizer's
+        # whole-buffer reduce-scatter are alternative DP reduction paths
+        assert not (self.overlap_grad_reduce and self.use_distributed_optimizer), (
+            "overlap_grad_reduce requires the non-distributed optimizer"
+        )
+
         if self.global_batch_size is None:
             self.global_batch_size = self.micro_batch_size * self.data_parallel_size
         assert (

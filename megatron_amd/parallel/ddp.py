@@ -67,7 +67,16 @@ class DistributedDataParallel(MegatronModule):
         self._grad_buffers: Optional[Dict[torch.dtype, MemoryBuffer]] = None
         self._grad_buffer_param_index_map = None
         self.grad_accs = []
-        self._async_handles = []
+        # overlap state: buckets are contiguous grad-buffer ranges; a bucket
+        # all-reduces asynchronously as soon as every one of its params has
+        # produced its grad in the FINAL microbatch backward (enabled by the
+        # schedule via enable_grad_sync()), hiding DP comm behind the rest of
+        # backward. RCCL runs the collective on its own stream, so the
+        # all-reduce genuinely overlaps compute.
+        self._buckets: List[dict] = []
+        self._param_to_bucket: Dict[torch.nn.Parameter, dict] = {}
+        self._sync_enabled = False
+        self._overlap_launched = False
 
         if not self.use_contiguous_buffers:
             return
@@ -119,13 +128,78 @@ class DistributedDataParallel(MegatronModule):
                 grad_acc.register_hook(self._make_param_hook(param))
                 self.grad_accs.append(grad_acc)
 
+        if self.overlap_grad_reduce:
+            self._build_buckets()
+
+    def _build_buckets(self):
+        """Split each dtype's grad buffer into contiguous ~bucket_numel
+        ranges. Views were assigned back-to-front, so walking params in
+        reverse module order walks the buffer from its END downward — the
+        same order backward produces grads — and consecutive params are
+        adjacent, so each bucket is one contiguous slice."""
+        per_dtype: Dict[torch.dtype, List[torch.nn.Parameter]] = {}
+        for param in reversed(list(self.module.parameters())):
+            if param.requires_grad:
+                dtype = (
+                    torch.float
+                    if self.accumulate_allreduce_grads_in_fp32
+                    else param.dtype
+                )
+                per_dtype.setdefault(dtype, []).append(param)
+        for dtype, params in per_dtype.items():
+            cur, cur_numel = [], 0
+            for p in params:
+                cur.append(p)
+                cur_numel += p.data.nelement()
+                if cur_numel >= self.bucket_numel:
+                    self._add_bucket(dtype, cur)
+                    cur, cur_numel = [], 0
+            if cur:
+                self._add_bucket(dtype, cur)
+
+    def _add_bucket(self, dtype, params):
+        index_map = self._grad_buffer_param_index_map[dtype]
+        start = min(index_map[p][0] for p in params)
+        end = max(index_map[p][1] for p in params)
+        bucket = {
+            "data": self._grad_buffers[dtype].data[start:end],
+            "params": set(params),
+            "done": set(),
+            "handle": None,
+        }
+        self._buckets.append(bucket)
+        for p in params:
+            self._param_to_bucket[p] = bucket
+
     def _make_param_hook(self, param):
         def param_hook(*unused):
             if param.grad is not None:
                 param.main_grad.add_(param.grad.data)
                 param.grad = None
+            if self._sync_enabled and param in self._param_to_bucket:
+                bucket = self._param_to_bucket[param]
+                bucket["done"].add(param)
+                if len(bucket["done"]) == len(bucket["params"]):
+                    self._launch_bucket(bucket)
 
         return param_hook
+
+    def _launch_bucket(self, bucket):
+        dp = ps.get_data_parallel_world_size()
+        if dp == 1 or bucket["handle"] is not None:
+            return
+        bucket["data"].div_(dp)
+        bucket["handle"] = torch.distributed.all_reduce(
+            bucket["data"], group=ps.get_data_parallel_group(), async_op=True
+        )
+        self._overlap_launched = True
+
+    def enable_grad_sync(self):
+        """Called by the schedule before the LAST microbatch's backward:
+        arms the per-bucket async all-reduce (grad-accumulation microbatches
+        before the last must not reduce)."""
+        if self.overlap_grad_reduce:
+            self._sync_enabled = True
 
     def forward(self, *inputs, **kwargs):
         return self.module(*inputs, **kwargs)
@@ -147,7 +221,25 @@ class DistributedDataParallel(MegatronModule):
             )
 
     def allreduce_gradients(self):
-        """Whole-buffer DP all-reduce (reference distributed.py:202-232)."""
+        """Finish the DP grad reduction. With overlap armed, buckets were
+        all-reduced asynchronously during backward — wait on the handles and
+        sweep any bucket whose params produced no grad this step; otherwise
+        whole-buffer all-reduce (reference distributed.py:202-232)."""
+        if self._sync_enabled:
+            dp = ps.get_data_parallel_world_size()
+            if dp > 1:
+                for bucket in self._buckets:
+                    if bucket["handle"] is None:
+                        self._launch_bucket(bucket)
+                for bucket in self._buckets:
+                    if bucket["handle"] is not None:
+                        bucket["handle"].wait()
+            for bucket in self._buckets:
+                bucket["done"].clear()
+                bucket["handle"] = None
+            self._sync_enabled = False
+            self._overlap_launched = False
+            return
         if self._grad_buffers is not None:
             for _, buffer_ in self._grad_buffers.items():
                 buffer_.data /= ps.get_data_parallel_world_size()

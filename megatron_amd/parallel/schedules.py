@@ -158,6 +158,10 @@ def forward_backward_no_pipelining(
         forward_data_store, cfg, timers, collect_non_loss_data,
     )
     if not forward_only:
+        # arm the bucketed async DP all-reduce for the final backward so
+        # grad communication overlaps the remaining backward compute
+        if hasattr(model, "enable_grad_sync"):
+            model.enable_grad_sync()
         backward_step(optimizer, input_tensor, output_tensor,
                       output_tensor_grad, cfg, timers)
     return forward_data_store

@@ -346,3 +346,58 @@ def _body_sequence_parallel(rank):
 
 def test_sequence_parallel_matches_dense():
     _spawn("_body_sequence_parallel", 29606)
+
+
+def _body_overlap_grad_reduce(rank):
+    """Bucketed backward-overlapped DP all-reduce produces bitwise-identical
+    main_grads to the whole-buffer reduction, including with 2-microbatch
+    grad accumulation (hooks must only reduce on the final backward)."""
+    from megatron_amd import parallel as mpu
+    from megatron_amd.parallel.ddp import DistributedDataParallel as LocalDDP
+
+    mpu.initialize_model_parallel(1, 1)  # dp = 2
+    torch.manual_seed(77)
+
+    def make():
+        torch.manual_seed(42)
+        return torch.nn.Sequential(
+            torch.nn.Linear(16, 64), torch.nn.Tanh(),
+            torch.nn.Linear(64, 64), torch.nn.Tanh(),
+            torch.nn.Linear(64, 16),
+        )
+
+    class _Wrap(torch.nn.Module):
+        def __init__(self, m):
+            super().__init__()
+            self.inner = m
+
+        def forward(self, x):
+            return self.inner(x)
+
+    # tiny bucket_numel forces multiple buckets
+    ddp_ov = LocalDDP(_Wrap(make()), True, True,
+                      overlap_grad_reduce=True, bucket_numel=1000)
+    ddp_ref = LocalDDP(_Wrap(make()), True, True)
+    assert len(ddp_ov._buckets) >= 3
+
+    torch.manual_seed(500 + rank)
+    micro1 = torch.randn(4, 16)
+    micro2 = torch.randn(4, 16)
+
+    for ddp, overlapped in ((ddp_ov, True), (ddp_ref, False)):
+        ddp.zero_grad_buffer()
+        ddp(micro1).pow(2).mean().backward()          # accumulation microbatch
+        assert not ddp._overlap_launched
+        ddp.enable_grad_sync()                        # what the schedule does
+        ddp(micro2).pow(2).mean().backward()          # final microbatch
+        if overlapped:
+            assert ddp._overlap_launched
+        ddp.allreduce_gradients()
+
+    for p_ov, p_ref in zip(ddp_ov.module.parameters(),
+                           ddp_ref.module.parameters()):
+        assert torch.equal(p_ov.main_grad, p_ref.main_grad)
+
+
+def test_overlap_grad_reduce():
+    _spawn("_body_overlap_grad_reduce", 29608)
