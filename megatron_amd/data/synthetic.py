@@ -30,8 +30,13 @@ def build_synthetic_datasets(cfg, train_val_test_num_samples):
     seq = cfg.seq_length + 1
     train = SyntheticGPTDataset(vocab, seq, train_val_test_num_samples[0],
                                 cfg.seed)
-    valid = SyntheticGPTDataset(vocab, seq, max(1, train_val_test_num_samples[1]),
-                                cfg.seed + 1)
-    test = SyntheticGPTDataset(vocab, seq, max(1, train_val_test_num_samples[2]),
-                               cfg.seed + 2)
+    # 2x margin + floor: a resumed run inherits consumed_valid_samples from
+    # the previous run's end-of-training eval, so the exact budget
+    # (eval_iters * (train_iters//eval_interval + 1)) can under-count
+    valid = SyntheticGPTDataset(
+        vocab, seq, max(64, 2 * train_val_test_num_samples[1]), cfg.seed + 1
+    )
+    test = SyntheticGPTDataset(
+        vocab, seq, max(64, 2 * train_val_test_num_samples[2]), cfg.seed + 2
+    )
     return train, valid, test
