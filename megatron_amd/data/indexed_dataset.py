@@ -196,7 +196,7 @@ class MMapIndexedDataset(torch.utils.data.Dataset):
         return len(self._index)
 
     def __getitem__(self, idx):
-        if isinstance(idx, int):
+        if isinstance(idx, (int, np.integer)):
             ptr, size = self._index[idx]
             np_array = np.frombuffer(
                 self._bin_buffer, dtype=self._index.dtype, count=size,
@@ -216,6 +216,7 @@ class MMapIndexedDataset(torch.utils.data.Dataset):
                 offset=ptr,
             )
             return np.split(np_array, offsets[:-1])
+        raise TypeError(f"unsupported index type {type(idx)}")
 
     def get(self, idx, offset=0, length=None):
         ptr, size = self._index[idx]

@@ -120,11 +120,13 @@ class AbstractTokenizer(ABC):
 
 class FakeTokenizer(AbstractTokenizer):
     """Synthetic-data tokenizer for benchmarks/tests (no vocab files in the
-    offline environment)."""
+    offline environment). vocab_extra_ids reserves T5 sentinel ids at the
+    top of the vocab, mirroring SentencePieceTokenizer."""
 
-    def __init__(self, vocab_size):
+    def __init__(self, vocab_size, vocab_extra_ids=0):
         super().__init__("FakeTokenizer")
         self._vocab_size = vocab_size
+        self._extra_ids = vocab_extra_ids
 
     @property
     def vocab_size(self):
@@ -159,6 +161,31 @@ class FakeTokenizer(AbstractTokenizer):
     @property
     def eos(self):
         return 2
+
+    @property
+    def bos_token_id(self):
+        return self.bos
+
+    @property
+    def eos_token_id(self):
+        return self.eos
+
+    @property
+    def cls(self):
+        return 3
+
+    @property
+    def sep(self):
+        return 4
+
+    @property
+    def mask(self):
+        return 5
+
+    @property
+    def additional_special_tokens_ids(self):
+        return list(range(self._vocab_size - self._extra_ids,
+                          self._vocab_size))
 
 
 class SentencePieceTokenizer(AbstractTokenizer):
