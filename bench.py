@@ -15,6 +15,7 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import sys
 import time
 
 # hipBLASLt TunableOp: use the pre-tuned GEMM algo selection committed under
@@ -216,10 +217,24 @@ def main():
 
     barrier_sync()
     t0 = time.time()
-    for _ in range(args.steps):
-        train_step(forward_step_func, data_iter, model, optimizer, opt_sched,
-                   cfg)
-    barrier_sync()
+    if os.environ.get("MEGATRON_AMD_TORCH_PROF"):
+        # diagnostic mode: attribute kernels to aten ops (not for timing)
+        from torch.profiler import ProfilerActivity, profile
+
+        with profile(activities=[ProfilerActivity.CUDA,
+                                 ProfilerActivity.CPU]) as prof:
+            for _ in range(args.steps):
+                train_step(forward_step_func, data_iter, model, optimizer,
+                           opt_sched, cfg)
+            barrier_sync()
+        print(prof.key_averages().table(
+            sort_by="self_cuda_time_total", row_limit=30,
+        ), file=sys.stderr)
+    else:
+        for _ in range(args.steps):
+            train_step(forward_step_func, data_iter, model, optimizer,
+                       opt_sched, cfg)
+        barrier_sync()
     elapsed = time.time() - t0
 
     # MAX over ranks
