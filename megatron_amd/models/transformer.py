@@ -173,8 +173,11 @@ class FlashSelfAttention(torch.nn.Module):
         self.window_size = window_size
 
     def forward(self, q, k, v):
-        # [s,b,n,h] -> [b,s,n,h]
-        q, k, v = (x.transpose(0, 1).contiguous() for x in (q, k, v))
+        # [s,b,n,h] -> [b,s,n,h] as VIEWS: the FA kernels are stride-aware
+        # over batch/seq, so no transpose copies are materialized and the
+        # kernel writes its output straight into an sbhd buffer (the final
+        # [s,b,n*h] reshape below is then also a view)
+        q, k, v = (x.transpose(0, 1) for x in (q, k, v))
         out = ops_f.flash_attention(
             q, k, v, causal=self.causal, softmax_scale=self.softmax_scale,
             window_size=self.window_size, dropout_p=self.dropout_p,

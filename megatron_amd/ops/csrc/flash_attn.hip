@@ -172,7 +172,11 @@ __global__ __launch_bounds__(NW * 64) void fa_fwd_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ out,
     float* __restrict__ lse, int B, int Sq, int Sk, int Hq, int Hkv,
-    float scale, int causal, int window) {
+    float scale, int causal, int window,
+    // batch/seq strides (elements; head stride is always D, last dim dense)
+    // so both bshd-contiguous and sbhd-transposed-view tensors work with no
+    // host-side transpose copies
+    long q_bs, long q_ss, long kv_bs, long kv_ss, long o_bs, long o_ss) {
   constexpr int KFRAGS = D / 32;
   constexpr int DTILES = D / 16;
   constexpr int IMG = tr_elems<D>(kBlockN);
@@ -193,12 +197,12 @@ __global__ __launch_bounds__(NW * 64) void fa_fwd_kernel(
   const int b = blockIdx.z;
   const int hkv = h / (Hq / Hkv);
 
-  const long q_base = ((long)b * Sq * Hq + h) * D;
-  const long k_base = ((long)b * Sk * Hkv + hkv) * D;
+  const long q_base = (long)b * q_bs + (long)h * D;
+  const long k_base = (long)b * kv_bs + (long)hkv * D;
   const long v_base = k_base;
-  const long o_base = q_base;
-  const int rq = Hq * D;
-  const int rk = Hkv * D;
+  const long o_base = (long)b * o_bs + (long)h * D;
+  const long rq = q_ss;
+  const long rk = kv_ss;
 
   const int qrow0 = qb * BM + wave * 16;
   const int skq = Sk - Sq;
@@ -348,7 +352,7 @@ __global__ __launch_bounds__(NW * 64) void fa_fwd_kernel(
     if (row < Sq) {
 #pragma unroll
       for (int t = 0; t < DTILES; ++t) {
-        out[o_base + (long)row * rq + t * 16 + row_in_tile] =
+        out[o_base + (long)row * o_ss + t * 16 + row_in_tile] =
             __float2bfloat16(o_acc[t][r] * inv_l);
       }
       if (row_in_tile == 0) {
@@ -366,16 +370,19 @@ template <int D>
 __global__ void fa_bwd_delta_kernel(const __hip_bfloat16* __restrict__ dout,
                                     const __hip_bfloat16* __restrict__ out,
                                     float* __restrict__ delta, int B, int Sq,
-                                    int Hq) {
+                                    int Hq, long do_bs, long do_ss, long o_bs,
+                                    long o_ss) {
   const long row = blockIdx.x;
   const int s = row % Sq;
   const long bh = row / Sq;
   const int b = bh / Hq;
   const int h = bh % Hq;
-  const long base = (((long)b * Sq + s) * Hq + h) * D;
+  const long do_base = (long)b * do_bs + (long)s * do_ss + (long)h * D;
+  const long o_base = (long)b * o_bs + (long)s * o_ss + (long)h * D;
   float acc = 0.f;
   for (int i = threadIdx.x; i < D; i += 64) {
-    acc += __bfloat162float(dout[base + i]) * __bfloat162float(out[base + i]);
+    acc += __bfloat162float(dout[do_base + i]) *
+           __bfloat162float(out[o_base + i]);
   }
   acc = wave_reduce_sum(acc);
   if (threadIdx.x == 0) delta[row] = acc;
@@ -392,7 +399,8 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dkdv_kernel(
     const __hip_bfloat16* __restrict__ dout, const float* __restrict__ lse,
     const float* __restrict__ delta, __hip_bfloat16* __restrict__ dk,
     __hip_bfloat16* __restrict__ dv, int B, int Sq, int Sk, int Hq, int Hkv,
-    float scale, int causal, int window) {
+    float scale, int causal, int window, long q_bs, long q_ss, long kv_bs,
+    long kv_ss, long do_bs, long do_ss, long dkv_bs, long dkv_ss) {
   constexpr int KFRAGS = D / 32;
   constexpr int DTILES = D / 16;
   constexpr int IMG = tr_elems<D>(kBlockM);
@@ -414,9 +422,8 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dkdv_kernel(
   const int b = blockIdx.z;
   const int gqa = Hq / Hkv;
 
-  const long k_base = ((long)b * Sk * Hkv + hkv) * D;
-  const int rk = Hkv * D;
-  const int rq = Hq * D;
+  const long k_base = (long)b * kv_bs + (long)hkv * D;
+  const long rk = kv_ss;
   const int skq = Sk - Sq;
 
   const int key0 = kb * BN + wave * 16;
@@ -460,11 +467,12 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dkdv_kernel(
   auto stage_iter = [&](int it, int buf) {
     int hq = hkv * gqa + it / nqb;
     int qbx = qb_start + it % nqb;
-    const long qb_base = ((long)b * Sq * Hq + hq) * D;
-    stage_tr_image<D, kBlockM, NT>(do_img[buf], dout + qb_base, rq,
-                                   qbx * kBlockM, Sq);
-    stage_tr_image<D, kBlockM, NT>(q_img[buf], q + qb_base, rq,
-                                   qbx * kBlockM, Sq);
+    stage_tr_image<D, kBlockM, NT>(
+        do_img[buf], dout + (long)b * do_bs + (long)hq * D, do_ss,
+        qbx * kBlockM, Sq);
+    stage_tr_image<D, kBlockM, NT>(
+        q_img[buf], q + (long)b * q_bs + (long)hq * D, q_ss, qbx * kBlockM,
+        Sq);
   };
 
   if (iters > 0) {
@@ -550,11 +558,12 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dkdv_kernel(
   for (int r = 0; r < 4; ++r) {
     int key = kb * BN + wave * 16 + kgroup * 4 + r;
     if (key < Sk) {
+      const long dkv_base = (long)b * dkv_bs + (long)hkv * D;
 #pragma unroll
       for (int t = 0; t < DTILES; ++t) {
-        dk[k_base + (long)key * rk + t * 16 + row_in_tile] =
+        dk[dkv_base + (long)key * dkv_ss + t * 16 + row_in_tile] =
             __float2bfloat16(dk_acc[t][r]);
-        dv[k_base + (long)key * rk + t * 16 + row_in_tile] =
+        dv[dkv_base + (long)key * dkv_ss + t * 16 + row_in_tile] =
             __float2bfloat16(dv_acc[t][r]);
       }
     }
@@ -570,7 +579,9 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dq_kernel(
     const __hip_bfloat16* __restrict__ v,
     const __hip_bfloat16* __restrict__ dout, const float* __restrict__ lse,
     const float* __restrict__ delta, __hip_bfloat16* __restrict__ dq, int B,
-    int Sq, int Sk, int Hq, int Hkv, float scale, int causal, int window) {
+    int Sq, int Sk, int Hq, int Hkv, float scale, int causal, int window,
+    long q_bs, long q_ss, long kv_bs, long kv_ss, long do_bs, long do_ss,
+    long dq_bs, long dq_ss) {
   constexpr int KFRAGS = D / 32;
   constexpr int DTILES = D / 16;
   constexpr int IMG = tr_elems<D>(kBlockN);
@@ -591,10 +602,11 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dq_kernel(
   const int b = blockIdx.z;
   const int hkv = h / (Hq / Hkv);
 
-  const long q_base = ((long)b * Sq * Hq + h) * D;
-  const long k_base = ((long)b * Sk * Hkv + hkv) * D;
-  const int rq = Hq * D;
-  const int rk = Hkv * D;
+  const long q_base = (long)b * q_bs + (long)h * D;
+  const long do_base = (long)b * do_bs + (long)h * D;
+  const long k_base = (long)b * kv_bs + (long)hkv * D;
+  const long rq = q_ss;
+  const long rk = kv_ss;
   const int skq = Sk - Sq;
   const int qrow0 = qb * BM + wave * 16;
 
@@ -609,7 +621,7 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dq_kernel(
     for (int kk = 0; kk < KFRAGS; ++kk) {
       qf[kk] = global_read16(q + q_base + (long)qr * rq + kk * 32 +
                              kgroup * 8);
-      dof[kk] = global_read16(dout + q_base + (long)qr * rq + kk * 32 +
+      dof[kk] = global_read16(dout + do_base + (long)qr * do_ss + kk * 32 +
                               kgroup * 8);
     }
   }
@@ -713,8 +725,8 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dq_kernel(
     if (row < Sq) {
 #pragma unroll
       for (int t = 0; t < DTILES; ++t) {
-        dq[q_base + (long)row * rq + t * 16 + row_in_tile] =
-            __float2bfloat16(dq_acc[t][r]);
+        dq[(long)b * dq_bs + (long)h * D + (long)row * dq_ss + t * 16 +
+           row_in_tile] = __float2bfloat16(dq_acc[t][r]);
       }
     }
   }
@@ -724,19 +736,40 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dq_kernel(
 
 // ---------------------------------------------------------------------------
 
+// [b,s,n,h] tensors must have the last two dims dense (n,h contiguous);
+// batch/seq strides are free, so a transposed view of an [s,b,n,h] buffer
+// (the model's native layout) is accepted without a copy.
+static void check_bshd(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.dim() == 4, name, ": need 4-D CUDA tensor");
+  TORCH_CHECK(t.stride(3) == 1 && t.stride(2) == t.size(3), name,
+              ": heads/head_dim must be the dense trailing dims");
+}
+
+// allocate a tensor with the same sizes/strides as ref (keeps sbhd buffers
+// sbhd so downstream [s,b,...] reshapes stay views)
+static torch::Tensor empty_like_strided(const torch::Tensor& ref) {
+  return torch::empty_strided(ref.sizes(), ref.strides(), ref.options());
+}
+
 std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
                                           torch::Tensor v, bool causal,
                                           double softmax_scale,
                                           int64_t window_size) {
-  TORCH_CHECK(q.is_cuda() && q.dim() == 4 && q.is_contiguous());
+  check_bshd(q, "q");
+  check_bshd(k, "k");
+  check_bshd(v, "v");
   TORCH_CHECK(q.scalar_type() == torch::kBFloat16,
               "flash_attn: bf16 only (got ", q.scalar_type(), ")");
+  TORCH_CHECK(k.strides() == v.strides(), "flash_attn: k/v layout mismatch");
   int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
   int Sk = k.size(1), Hkv = k.size(2);
   TORCH_CHECK(D == 64 || D == 128, "flash_attn: head dim must be 64/128");
   TORCH_CHECK(Hq % Hkv == 0);
 
-  auto out = torch::empty_like(q);
+  auto out = empty_like_strided(q);
+  const long q_bs = q.stride(0), q_ss = q.stride(1);
+  const long kv_bs = k.stride(0), kv_ss = k.stride(1);
+  const long o_bs = out.stride(0), o_ss = out.stride(1);
   auto lse = torch::empty({B, Hq, Sq}, q.options().dtype(torch::kFloat32));
   auto stream = c10::hip::getCurrentHIPStream();
   int win = window_size > 0 ? (int)window_size : 0;
@@ -744,7 +777,7 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
     const char* e = getenv("MEGATRON_AMD_FA_FWD_WAVES");
     return e ? atoi(e) : 12;
   }();
-#define LAUNCH_FWD(DD, NW)                                                      do {                                                                            dim3 grid((Sq + NW * 16 - 1) / (NW * 16), Hq, B);                             hipLaunchKernelGGL((fa_fwd_kernel<DD, NW>), grid, dim3(NW * 64), 0,                              stream, (const __hip_bfloat16*)q.data_ptr(),                                  (const __hip_bfloat16*)k.data_ptr(),                                          (const __hip_bfloat16*)v.data_ptr(),                                          (__hip_bfloat16*)out.data_ptr(),                                              lse.data_ptr<float>(), B, Sq, Sk, Hq, Hkv,                                    (float)softmax_scale, causal ? 1 : 0, win);              } while (0)
+#define LAUNCH_FWD(DD, NW)                                                      do {                                                                            dim3 grid((Sq + NW * 16 - 1) / (NW * 16), Hq, B);                             hipLaunchKernelGGL((fa_fwd_kernel<DD, NW>), grid, dim3(NW * 64), 0,                              stream, (const __hip_bfloat16*)q.data_ptr(),                                  (const __hip_bfloat16*)k.data_ptr(),                                          (const __hip_bfloat16*)v.data_ptr(),                                          (__hip_bfloat16*)out.data_ptr(),                                              lse.data_ptr<float>(), B, Sq, Sk, Hq, Hkv,                                    (float)softmax_scale, causal ? 1 : 0, win,                                    q_bs, q_ss, kv_bs, kv_ss, o_bs, o_ss);              } while (0)
   if (D == 128) {
     if (nw_env >= 12) LAUNCH_FWD(128, 12);
     else LAUNCH_FWD(128, 4);
@@ -761,16 +794,27 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
                                           torch::Tensor out, torch::Tensor lse,
                                           bool causal, double softmax_scale,
                                           int64_t window_size) {
-  TORCH_CHECK(dout.is_cuda() && dout.is_contiguous());
+  check_bshd(dout, "dout");
+  check_bshd(q, "q");
+  check_bshd(k, "k");
+  check_bshd(v, "v");
+  check_bshd(out, "out");
+  TORCH_CHECK(k.strides() == v.strides(), "flash_attn: k/v layout mismatch");
   int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
   int Sk = k.size(1), Hkv = k.size(2);
   int win = window_size > 0 ? (int)window_size : 0;
 
-  auto dq = torch::empty_like(q);
-  auto dk = torch::empty_like(k);
-  auto dv = torch::empty_like(v);
+  auto dq = empty_like_strided(q);
+  auto dk = empty_like_strided(k);
+  auto dv = empty_like_strided(v);
   auto delta = torch::empty({B, Hq, Sq}, q.options().dtype(torch::kFloat32));
   auto stream = c10::hip::getCurrentHIPStream();
+  const long q_bs = q.stride(0), q_ss = q.stride(1);
+  const long kv_bs = k.stride(0), kv_ss = k.stride(1);
+  const long do_bs = dout.stride(0), do_ss = dout.stride(1);
+  const long o_bs = out.stride(0), o_ss = out.stride(1);
+  const long dq_bs = dq.stride(0), dq_ss = dq.stride(1);
+  const long dkv_bs = dk.stride(0), dkv_ss = dk.stride(1);
 
   long rows = (long)B * Hq * Sq;
 #define LAUNCH_BWD(DD)                                                        \
@@ -778,7 +822,8 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
     hipLaunchKernelGGL((fa_bwd_delta_kernel<DD>), dim3(rows), dim3(64), 0,    \
                        stream, (const __hip_bfloat16*)dout.data_ptr(),        \
                        (const __hip_bfloat16*)out.data_ptr(),                 \
-                       delta.data_ptr<float>(), B, Sq, Hq);                   \
+                       delta.data_ptr<float>(), B, Sq, Hq, do_bs, do_ss,      \
+                       o_bs, o_ss);                                           \
     dim3 gridk((Sk + 12 * 16 - 1) / (12 * 16), Hkv, B);                       \
     hipLaunchKernelGGL((fa_bwd_dkdv_kernel<DD, 12>), gridk, dim3(12 * 64), 0, \
                        stream, (const __hip_bfloat16*)q.data_ptr(),           \
@@ -788,7 +833,8 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
                        lse.data_ptr<float>(), delta.data_ptr<float>(),        \
                        (__hip_bfloat16*)dk.data_ptr(),                        \
                        (__hip_bfloat16*)dv.data_ptr(), B, Sq, Sk, Hq, Hkv,    \
-                       (float)softmax_scale, causal ? 1 : 0, win);            \
+                       (float)softmax_scale, causal ? 1 : 0, win, q_bs, q_ss, \
+                       kv_bs, kv_ss, do_bs, do_ss, dkv_bs, dkv_ss);           \
     dim3 gridq((Sq + 12 * 16 - 1) / (12 * 16), Hq, B);                        \
     hipLaunchKernelGGL((fa_bwd_dq_kernel<DD, 12>), gridq, dim3(12 * 64), 0,   \
                        stream, (const __hip_bfloat16*)q.data_ptr(),           \
@@ -797,7 +843,8 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
                        (const __hip_bfloat16*)dout.data_ptr(),                \
                        lse.data_ptr<float>(), delta.data_ptr<float>(),        \
                        (__hip_bfloat16*)dq.data_ptr(), B, Sq, Sk, Hq, Hkv,    \
-                       (float)softmax_scale, causal ? 1 : 0, win);            \
+                       (float)softmax_scale, causal ? 1 : 0, win, q_bs, q_ss, \
+                       kv_bs, kv_ss, do_bs, do_ss, dq_bs, dq_ss);             \
   } while (0)
 
   if (D == 128) {

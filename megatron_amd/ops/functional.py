@@ -350,8 +350,18 @@ class FlashAttnFunction(torch.autograd.Function):
         ctx.softmax_scale = softmax_scale
         ctx.window_size = window_size
         if ext is not None:
+            # kernels are stride-aware over batch/seq (heads/head_dim must be
+            # the dense trailing dims): sbhd-transposed views pass with no
+            # copy; anything else (e.g. a sliced kv) still needs contiguous
+            def ok(t):
+                return t.stride(-1) == 1 and t.stride(-2) == t.shape[-1]
+
+            q = q if ok(q) else q.contiguous()
+            k = k if ok(k) else k.contiguous()
+            v = v if ok(v) and k.stride() == v.stride() else v.contiguous()
+            k = k if k.stride() == v.stride() else k.contiguous()
             out, lse = ext.flash_attn_fwd(
-                q.contiguous(), k.contiguous(), v.contiguous(),
+                q, k, v,
                 bool(causal), float(softmax_scale),
                 int(window_size) if window_size is not None else -1,
             )
@@ -366,8 +376,11 @@ class FlashAttnFunction(torch.autograd.Function):
         q, k, v, out, lse = ctx.saved_tensors
         ext = _C(dout)
         if ext is not None:
+            if not (dout.stride(-1) == 1
+                    and dout.stride(-2) == dout.shape[-1]):
+                dout = dout.contiguous()
             dq, dk, dv = ext.flash_attn_bwd(
-                dout.contiguous(), q, k, v, out, lse,
+                dout, q, k, v, out, lse,
                 bool(ctx.causal), float(ctx.softmax_scale),
                 int(ctx.window_size) if ctx.window_size is not None else -1,
             )

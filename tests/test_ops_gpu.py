@@ -283,3 +283,37 @@ class TestFlashAttention:
         assert rel_err(dv, rdv) < 8e-2
         assert rel_err(dk, rdk) < 8e-2
         assert rel_err(dq, rdq) < 8e-2
+
+    def test_sbhd_strided_matches_contiguous(self):
+        """sbhd-transposed views (the model's native layout, no transpose
+        copies) must produce bitwise-identical results to bshd-contiguous
+        inputs — fwd out/lse and all three backward grads."""
+        ext = _ext()
+        b, s, n, d = 2, 320, 4, 128
+        nkv = 2
+        scale = 1.0 / math.sqrt(d)
+        # native [s,b,n,h] buffers -> [b,s,n,h] views
+        q_sb = torch.randn(s, b, n, d, device="cuda", dtype=torch.bfloat16)
+        k_sb = torch.randn(s, b, nkv, d, device="cuda", dtype=torch.bfloat16)
+        v_sb = torch.randn(s, b, nkv, d, device="cuda", dtype=torch.bfloat16)
+        qv, kv_, vv = (x.transpose(0, 1) for x in (q_sb, k_sb, v_sb))
+        assert not qv.is_contiguous()
+        qc, kc, vc = (x.contiguous() for x in (qv, kv_, vv))
+
+        out_v, lse_v = ext.flash_attn_fwd(qv, kv_, vv, True, scale, -1)
+        out_c, lse_c = ext.flash_attn_fwd(qc, kc, vc, True, scale, -1)
+        assert not out_v.is_contiguous()  # preserved sbhd layout
+        assert torch.equal(out_v.contiguous(), out_c)
+        assert torch.equal(lse_v, lse_c)
+
+        dout_sb = torch.randn(s, b, n, d, device="cuda",
+                              dtype=torch.bfloat16)
+        dout_v = dout_sb.transpose(0, 1)
+        dq_v, dk_v, dv_v = ext.flash_attn_bwd(dout_v, qv, kv_, vv, out_v,
+                                              lse_v, True, scale, -1)
+        dq_c, dk_c, dv_c = ext.flash_attn_bwd(dout_v.contiguous(), qc, kc,
+                                              vc, out_c, lse_c, True, scale,
+                                              -1)
+        assert torch.equal(dq_v.contiguous(), dq_c)
+        assert torch.equal(dk_v.contiguous(), dk_c)
+        assert torch.equal(dv_v.contiguous(), dv_c)
