@@ -315,13 +315,18 @@ class ParallelAttention(MegatronModule):
             ngroups = nkv
             nq_per_group = np_ // nkv
             mixed = mixed.view(sq, b, ngroups, (nq_per_group + 2) * hn)
-            query = mixed[..., : nq_per_group * hn].reshape(sq, b, np_, hn)
-            key = mixed[
-                ..., nq_per_group * hn : (nq_per_group + 1) * hn
-            ].reshape(sq, b, nkv, hn)
-            value = mixed[..., (nq_per_group + 1) * hn :].reshape(
-                sq, b, nkv, hn
-            )
+            # k/v (and q when nq_per_group == 1) stay uniformly-strided VIEWS
+            # into the fused projection output: the FA and RoPE kernels are
+            # stride-aware, so no slice-materializing copies are needed. A
+            # GQA query (heads non-uniform across groups) still reshapes.
+            if nq_per_group == 1:
+                query = mixed[..., :hn]
+            else:
+                query = mixed[..., : nq_per_group * hn].reshape(
+                    sq, b, np_, hn
+                )
+            key = mixed[..., nq_per_group * hn : (nq_per_group + 1) * hn]
+            value = mixed[..., (nq_per_group + 1) * hn :]
         else:
             # cross attention: q from decoder states, kv from encoder output
             q_out, _ = self.query(hidden_states)

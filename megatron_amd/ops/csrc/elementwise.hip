@@ -144,21 +144,28 @@ template <typename T, int DIR>
 __global__ void rope_kernel(const T* __restrict__ x,
                             const float* __restrict__ cosT,
                             const float* __restrict__ sinT,
-                            T* __restrict__ y, long sbn, int s_stride_rows,
-                            int h_half) {
-  // rows = s*b*n, each row has h = 2*h_half elements
+                            T* __restrict__ y, long sbn, int b, int n,
+                            int h_half, long x_ss, long x_bs, long x_ns) {
+  // rows = s*b*n, each row has h = 2*h_half elements. x may be a strided
+  // VIEW (e.g. the q/k head slices of a fused QKV projection); y is written
+  // contiguous [s,b,n,h].
   for (long idx = (long)blockIdx.x * kBlock + threadIdx.x;
        idx < sbn * (long)h_half; idx += (long)gridDim.x * kBlock) {
     const long row = idx / h_half;
     const int j = idx % h_half;
-    const int s_idx = row / s_stride_rows;  // rows per s = b*n
-    const long base = row * (long)(2 * h_half) + 2 * j;
+    const int n_idx = row % n;
+    const long sb = row / n;
+    const int b_idx = sb % b;
+    const int s_idx = sb / b;
+    const long x_base =
+        (long)s_idx * x_ss + (long)b_idx * x_bs + (long)n_idx * x_ns + 2 * j;
+    const long y_base = row * (long)(2 * h_half) + 2 * j;
     float c = cosT[(long)s_idx * h_half + j];
     float sn = sinT[(long)s_idx * h_half + j] * DIR;
-    float x1 = DTypeTraits<T>::to_float(x[base]);
-    float x2 = DTypeTraits<T>::to_float(x[base + 1]);
-    y[base] = DTypeTraits<T>::from_float(x1 * c - x2 * sn);
-    y[base + 1] = DTypeTraits<T>::from_float(x2 * c + x1 * sn);
+    float x1 = DTypeTraits<T>::to_float(x[x_base]);
+    float x2 = DTypeTraits<T>::to_float(x[x_base + 1]);
+    y[y_base] = DTypeTraits<T>::from_float(x1 * c - x2 * sn);
+    y[y_base + 1] = DTypeTraits<T>::from_float(x2 * c + x1 * sn);
   }
 }
 
@@ -271,12 +278,13 @@ torch::Tensor glu_bwd(torch::Tensor dy, torch::Tensor x, int64_t mode) {
 
 static torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cosT,
                                 torch::Tensor sinT, int dir) {
-  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.is_contiguous());
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.stride(3) == 1,
+              "rope: need 4-D x with dense head_dim");
   TORCH_CHECK(cosT.scalar_type() == torch::kFloat32);
   int s = x.size(0), b = x.size(1), n = x.size(2), h = x.size(3);
   TORCH_CHECK(cosT.size(0) >= s && cosT.size(1) == h / 2,
               "rope table too small");
-  auto y = torch::empty_like(x);
+  auto y = torch::empty({s, b, n, h}, x.options());
   long sbn = (long)s * b * n;
   long total = sbn * (h / 2);
   auto stream = c10::hip::getCurrentHIPStream();
@@ -286,12 +294,14 @@ static torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cosT,
   hipLaunchKernelGGL((rope_kernel<T, 1>), dim3(grid_for(total)),             \
                      dim3(kBlock), 0, stream, (const T*)x.data_ptr(),        \
                      cosc.data_ptr<float>(), sinc.data_ptr<float>(),         \
-                     (T*)y.data_ptr(), sbn, b * n, h / 2)
+                     (T*)y.data_ptr(), sbn, b, n, h / 2, x.stride(0),        \
+                     x.stride(1), x.stride(2))
 #define LAUNCH_ROPE_B(T)                                                     \
   hipLaunchKernelGGL((rope_kernel<T, -1>), dim3(grid_for(total)),            \
                      dim3(kBlock), 0, stream, (const T*)x.data_ptr(),        \
                      cosc.data_ptr<float>(), sinc.data_ptr<float>(),         \
-                     (T*)y.data_ptr(), sbn, b * n, h / 2)
+                     (T*)y.data_ptr(), sbn, b, n, h / 2, x.stride(0),        \
+                     x.stride(1), x.stride(2))
   if (dir > 0) {
     DISPATCH_DTYPE(x, LAUNCH_ROPE_F);
   } else {

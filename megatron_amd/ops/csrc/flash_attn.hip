@@ -54,6 +54,13 @@ constexpr int kBlockN = 64;
 constexpr int kThreads = 256;
 constexpr int kStrip = 72;  // padded strip stride (bf16), P/dS tiles
 
+// per-tensor strides in elements (batch, seq, head); the head_dim axis must
+// be dense. Lets q/k/v/out/grads be strided VIEWS — sbhd transposes and the
+// per-head slices of a fused QKV projection — with no host-side copies.
+struct TStr {
+  long bs, ss, hs;
+};
+
 __device__ __forceinline__ float group16_max(float v) {
 #pragma unroll
   for (int off = 1; off < 16; off <<= 1) {
@@ -172,11 +179,7 @@ __global__ __launch_bounds__(NW * 64) void fa_fwd_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ out,
     float* __restrict__ lse, int B, int Sq, int Sk, int Hq, int Hkv,
-    float scale, int causal, int window,
-    // batch/seq strides (elements; head stride is always D, last dim dense)
-    // so both bshd-contiguous and sbhd-transposed-view tensors work with no
-    // host-side transpose copies
-    long q_bs, long q_ss, long kv_bs, long kv_ss, long o_bs, long o_ss) {
+    float scale, int causal, int window, TStr qs, TStr ks, TStr vs, TStr os) {
   constexpr int KFRAGS = D / 32;
   constexpr int DTILES = D / 16;
   constexpr int IMG = tr_elems<D>(kBlockN);
@@ -197,12 +200,13 @@ __global__ __launch_bounds__(NW * 64) void fa_fwd_kernel(
   const int b = blockIdx.z;
   const int hkv = h / (Hq / Hkv);
 
-  const long q_base = (long)b * q_bs + (long)h * D;
-  const long k_base = (long)b * kv_bs + (long)hkv * D;
-  const long v_base = k_base;
-  const long o_base = (long)b * o_bs + (long)h * D;
-  const long rq = q_ss;
-  const long rk = kv_ss;
+  const long q_base = (long)b * qs.bs + (long)h * qs.hs;
+  const long k_base = (long)b * ks.bs + (long)hkv * ks.hs;
+  const long v_base = (long)b * vs.bs + (long)hkv * vs.hs;
+  const long o_base = (long)b * os.bs + (long)h * os.hs;
+  const long rq = qs.ss;
+  const long rk = ks.ss;
+  const long rv = vs.ss;
 
   const int qrow0 = qb * BM + wave * 16;
   const int skq = Sk - Sq;
@@ -243,7 +247,7 @@ __global__ __launch_bounds__(NW * 64) void fa_fwd_kernel(
 
   stage_tr_image<D, kBlockN, NT>(k_img[kb_start & 1], k + k_base, rk,
                                  kb_start * kBlockN, Sk);
-  stage_tr_image<D, kBlockN, NT>(v_img[kb_start & 1], v + v_base, rk,
+  stage_tr_image<D, kBlockN, NT>(v_img[kb_start & 1], v + v_base, rv,
                                  kb_start * kBlockN, Sk);
   __syncthreads();
 
@@ -256,7 +260,7 @@ __global__ __launch_bounds__(NW * 64) void fa_fwd_kernel(
     if (kb + 1 < kb_end) {
       stage_tr_image<D, kBlockN, NT>(k_img[(kb + 1) & 1], k + k_base, rk,
                                      (kb + 1) * kBlockN, Sk);
-      stage_tr_image<D, kBlockN, NT>(v_img[(kb + 1) & 1], v + v_base, rk,
+      stage_tr_image<D, kBlockN, NT>(v_img[(kb + 1) & 1], v + v_base, rv,
                                      (kb + 1) * kBlockN, Sk);
     }
 
@@ -352,7 +356,7 @@ __global__ __launch_bounds__(NW * 64) void fa_fwd_kernel(
     if (row < Sq) {
 #pragma unroll
       for (int t = 0; t < DTILES; ++t) {
-        out[o_base + (long)row * o_ss + t * 16 + row_in_tile] =
+        out[o_base + (long)row * os.ss + t * 16 + row_in_tile] =
             __float2bfloat16(o_acc[t][r] * inv_l);
       }
       if (row_in_tile == 0) {
@@ -370,15 +374,14 @@ template <int D>
 __global__ void fa_bwd_delta_kernel(const __hip_bfloat16* __restrict__ dout,
                                     const __hip_bfloat16* __restrict__ out,
                                     float* __restrict__ delta, int B, int Sq,
-                                    int Hq, long do_bs, long do_ss, long o_bs,
-                                    long o_ss) {
+                                    int Hq, TStr ds, TStr os) {
   const long row = blockIdx.x;
   const int s = row % Sq;
   const long bh = row / Sq;
   const int b = bh / Hq;
   const int h = bh % Hq;
-  const long do_base = (long)b * do_bs + (long)s * do_ss + (long)h * D;
-  const long o_base = (long)b * o_bs + (long)s * o_ss + (long)h * D;
+  const long do_base = (long)b * ds.bs + (long)s * ds.ss + (long)h * ds.hs;
+  const long o_base = (long)b * os.bs + (long)s * os.ss + (long)h * os.hs;
   float acc = 0.f;
   for (int i = threadIdx.x; i < D; i += 64) {
     acc += __bfloat162float(dout[do_base + i]) *
@@ -399,8 +402,8 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dkdv_kernel(
     const __hip_bfloat16* __restrict__ dout, const float* __restrict__ lse,
     const float* __restrict__ delta, __hip_bfloat16* __restrict__ dk,
     __hip_bfloat16* __restrict__ dv, int B, int Sq, int Sk, int Hq, int Hkv,
-    float scale, int causal, int window, long q_bs, long q_ss, long kv_bs,
-    long kv_ss, long do_bs, long do_ss, long dkv_bs, long dkv_ss) {
+    float scale, int causal, int window, TStr qs, TStr ks, TStr vs, TStr ds,
+    TStr dks, TStr dvs) {
   constexpr int KFRAGS = D / 32;
   constexpr int DTILES = D / 16;
   constexpr int IMG = tr_elems<D>(kBlockM);
@@ -422,8 +425,10 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dkdv_kernel(
   const int b = blockIdx.z;
   const int gqa = Hq / Hkv;
 
-  const long k_base = (long)b * kv_bs + (long)hkv * D;
-  const long rk = kv_ss;
+  const long k_base = (long)b * ks.bs + (long)hkv * ks.hs;
+  const long v_base = (long)b * vs.bs + (long)hkv * vs.hs;
+  const long rk = ks.ss;
+  const long rv = vs.ss;
   const int skq = Sk - Sq;
 
   const int key0 = kb * BN + wave * 16;
@@ -436,7 +441,7 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dkdv_kernel(
     for (int kk = 0; kk < KFRAGS; ++kk) {
       ka[kk] = global_read16(k + k_base + (long)kr * rk + kk * 32 +
                              kgroup * 8);
-      va[kk] = global_read16(v + k_base + (long)kr * rk + kk * 32 +
+      va[kk] = global_read16(v + v_base + (long)kr * rv + kk * 32 +
                              kgroup * 8);
     }
   }
@@ -468,11 +473,11 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dkdv_kernel(
     int hq = hkv * gqa + it / nqb;
     int qbx = qb_start + it % nqb;
     stage_tr_image<D, kBlockM, NT>(
-        do_img[buf], dout + (long)b * do_bs + (long)hq * D, do_ss,
+        do_img[buf], dout + (long)b * ds.bs + (long)hq * ds.hs, ds.ss,
         qbx * kBlockM, Sq);
     stage_tr_image<D, kBlockM, NT>(
-        q_img[buf], q + (long)b * q_bs + (long)hq * D, q_ss, qbx * kBlockM,
-        Sq);
+        q_img[buf], q + (long)b * qs.bs + (long)hq * qs.hs, qs.ss,
+        qbx * kBlockM, Sq);
   };
 
   if (iters > 0) {
@@ -558,12 +563,13 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dkdv_kernel(
   for (int r = 0; r < 4; ++r) {
     int key = kb * BN + wave * 16 + kgroup * 4 + r;
     if (key < Sk) {
-      const long dkv_base = (long)b * dkv_bs + (long)hkv * D;
+      const long dk_base = (long)b * dks.bs + (long)hkv * dks.hs;
+      const long dv_base = (long)b * dvs.bs + (long)hkv * dvs.hs;
 #pragma unroll
       for (int t = 0; t < DTILES; ++t) {
-        dk[dkv_base + (long)key * dkv_ss + t * 16 + row_in_tile] =
+        dk[dk_base + (long)key * dks.ss + t * 16 + row_in_tile] =
             __float2bfloat16(dk_acc[t][r]);
-        dv[dkv_base + (long)key * dkv_ss + t * 16 + row_in_tile] =
+        dv[dv_base + (long)key * dvs.ss + t * 16 + row_in_tile] =
             __float2bfloat16(dv_acc[t][r]);
       }
     }
@@ -580,8 +586,7 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dq_kernel(
     const __hip_bfloat16* __restrict__ dout, const float* __restrict__ lse,
     const float* __restrict__ delta, __hip_bfloat16* __restrict__ dq, int B,
     int Sq, int Sk, int Hq, int Hkv, float scale, int causal, int window,
-    long q_bs, long q_ss, long kv_bs, long kv_ss, long do_bs, long do_ss,
-    long dq_bs, long dq_ss) {
+    TStr qs, TStr ks, TStr vs, TStr ds, TStr dqs) {
   constexpr int KFRAGS = D / 32;
   constexpr int DTILES = D / 16;
   constexpr int IMG = tr_elems<D>(kBlockN);
@@ -602,11 +607,13 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dq_kernel(
   const int b = blockIdx.z;
   const int hkv = h / (Hq / Hkv);
 
-  const long q_base = (long)b * q_bs + (long)h * D;
-  const long do_base = (long)b * do_bs + (long)h * D;
-  const long k_base = (long)b * kv_bs + (long)hkv * D;
-  const long rq = q_ss;
-  const long rk = kv_ss;
+  const long q_base = (long)b * qs.bs + (long)h * qs.hs;
+  const long do_base = (long)b * ds.bs + (long)h * ds.hs;
+  const long k_base = (long)b * ks.bs + (long)hkv * ks.hs;
+  const long v_base = (long)b * vs.bs + (long)hkv * vs.hs;
+  const long rq = qs.ss;
+  const long rk = ks.ss;
+  const long rv = vs.ss;
   const int skq = Sk - Sq;
   const int qrow0 = qb * BM + wave * 16;
 
@@ -621,7 +628,7 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dq_kernel(
     for (int kk = 0; kk < KFRAGS; ++kk) {
       qf[kk] = global_read16(q + q_base + (long)qr * rq + kk * 32 +
                              kgroup * 8);
-      dof[kk] = global_read16(dout + do_base + (long)qr * do_ss + kk * 32 +
+      dof[kk] = global_read16(dout + do_base + (long)qr * ds.ss + kk * 32 +
                               kgroup * 8);
     }
   }
@@ -651,7 +658,7 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dq_kernel(
 
   stage_tr_image<D, kBlockN, NT>(k_img[kb_start & 1], k + k_base, rk,
                                  kb_start * kBlockN, Sk);
-  stage_tr_image<D, kBlockN, NT>(v_img[kb_start & 1], v + k_base, rk,
+  stage_tr_image<D, kBlockN, NT>(v_img[kb_start & 1], v + v_base, rv,
                                  kb_start * kBlockN, Sk);
   __syncthreads();
 
@@ -662,7 +669,7 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dq_kernel(
     if (kb + 1 < kb_end) {
       stage_tr_image<D, kBlockN, NT>(k_img[(kb + 1) & 1], k + k_base, rk,
                                      (kb + 1) * kBlockN, Sk);
-      stage_tr_image<D, kBlockN, NT>(v_img[(kb + 1) & 1], v + k_base, rk,
+      stage_tr_image<D, kBlockN, NT>(v_img[(kb + 1) & 1], v + v_base, rv,
                                      (kb + 1) * kBlockN, Sk);
     }
 
@@ -725,8 +732,8 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dq_kernel(
     if (row < Sq) {
 #pragma unroll
       for (int t = 0; t < DTILES; ++t) {
-        dq[(long)b * dq_bs + (long)h * D + (long)row * dq_ss + t * 16 +
-           row_in_tile] = __float2bfloat16(dq_acc[t][r]);
+        dq[(long)b * dqs.bs + (long)h * dqs.hs + (long)row * dqs.ss +
+           t * 16 + row_in_tile] = __float2bfloat16(dq_acc[t][r]);
       }
     }
   }
@@ -736,18 +743,34 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dq_kernel(
 
 // ---------------------------------------------------------------------------
 
-// [b,s,n,h] tensors must have the last two dims dense (n,h contiguous);
-// batch/seq strides are free, so a transposed view of an [s,b,n,h] buffer
-// (the model's native layout) is accepted without a copy.
+// [b,s,n,h] tensors need only a dense head_dim axis; batch/seq/head strides
+// are free, so sbhd transposes AND the per-head q/k/v slices of a fused QKV
+// projection pass with no copies. 16-byte-aligned rows required (head_dim is
+// 64/128 and torch allocations are 256B-aligned, so any element-strided view
+// of a bf16 buffer qualifies as long as the strides are multiples of 8).
 static void check_bshd(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda() && t.dim() == 4, name, ": need 4-D CUDA tensor");
-  TORCH_CHECK(t.stride(3) == 1 && t.stride(2) == t.size(3), name,
-              ": heads/head_dim must be the dense trailing dims");
+  TORCH_CHECK(t.stride(3) == 1, name, ": head_dim must be dense");
+  TORCH_CHECK(t.stride(0) % 8 == 0 && t.stride(1) % 8 == 0 &&
+                  t.stride(2) % 8 == 0,
+              name, ": strides must keep rows 16-byte aligned");
 }
 
-// allocate a tensor with the same sizes/strides as ref (keeps sbhd buffers
-// sbhd so downstream [s,b,...] reshapes stay views)
+static TStr tstr(const torch::Tensor& t) {
+  return TStr{t.stride(0), t.stride(1), t.stride(2)};
+}
+
+// allocate like ref, preserving strides only when they tile storage densely
+// (keeps sbhd buffers sbhd so downstream [s,b,...] reshapes stay views); a
+// sparse view (e.g. a QKV slice) gets a fresh contiguous buffer instead
 static torch::Tensor empty_like_strided(const torch::Tensor& ref) {
+  long extent = 1;
+  for (int i = 0; i < ref.dim(); ++i) {
+    extent += (ref.size(i) - 1) * ref.stride(i);
+  }
+  if (extent != ref.numel()) {
+    return torch::empty(ref.sizes(), ref.options());
+  }
   return torch::empty_strided(ref.sizes(), ref.strides(), ref.options());
 }
 
@@ -760,16 +783,13 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
   check_bshd(v, "v");
   TORCH_CHECK(q.scalar_type() == torch::kBFloat16,
               "flash_attn: bf16 only (got ", q.scalar_type(), ")");
-  TORCH_CHECK(k.strides() == v.strides(), "flash_attn: k/v layout mismatch");
   int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
   int Sk = k.size(1), Hkv = k.size(2);
   TORCH_CHECK(D == 64 || D == 128, "flash_attn: head dim must be 64/128");
   TORCH_CHECK(Hq % Hkv == 0);
 
   auto out = empty_like_strided(q);
-  const long q_bs = q.stride(0), q_ss = q.stride(1);
-  const long kv_bs = k.stride(0), kv_ss = k.stride(1);
-  const long o_bs = out.stride(0), o_ss = out.stride(1);
+  const TStr qs = tstr(q), ks = tstr(k), vs = tstr(v), os = tstr(out);
   auto lse = torch::empty({B, Hq, Sq}, q.options().dtype(torch::kFloat32));
   auto stream = c10::hip::getCurrentHIPStream();
   int win = window_size > 0 ? (int)window_size : 0;
@@ -777,7 +797,7 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
     const char* e = getenv("MEGATRON_AMD_FA_FWD_WAVES");
     return e ? atoi(e) : 12;
   }();
-#define LAUNCH_FWD(DD, NW)                                                      do {                                                                            dim3 grid((Sq + NW * 16 - 1) / (NW * 16), Hq, B);                             hipLaunchKernelGGL((fa_fwd_kernel<DD, NW>), grid, dim3(NW * 64), 0,                              stream, (const __hip_bfloat16*)q.data_ptr(),                                  (const __hip_bfloat16*)k.data_ptr(),                                          (const __hip_bfloat16*)v.data_ptr(),                                          (__hip_bfloat16*)out.data_ptr(),                                              lse.data_ptr<float>(), B, Sq, Sk, Hq, Hkv,                                    (float)softmax_scale, causal ? 1 : 0, win,                                    q_bs, q_ss, kv_bs, kv_ss, o_bs, o_ss);              } while (0)
+#define LAUNCH_FWD(DD, NW)                                                      do {                                                                            dim3 grid((Sq + NW * 16 - 1) / (NW * 16), Hq, B);                             hipLaunchKernelGGL((fa_fwd_kernel<DD, NW>), grid, dim3(NW * 64), 0,                              stream, (const __hip_bfloat16*)q.data_ptr(),                                  (const __hip_bfloat16*)k.data_ptr(),                                          (const __hip_bfloat16*)v.data_ptr(),                                          (__hip_bfloat16*)out.data_ptr(),                                              lse.data_ptr<float>(), B, Sq, Sk, Hq, Hkv,                                    (float)softmax_scale, causal ? 1 : 0, win,                                    qs, ks, vs, os);              } while (0)
   if (D == 128) {
     if (nw_env >= 12) LAUNCH_FWD(128, 12);
     else LAUNCH_FWD(128, 4);
@@ -799,7 +819,6 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
   check_bshd(k, "k");
   check_bshd(v, "v");
   check_bshd(out, "out");
-  TORCH_CHECK(k.strides() == v.strides(), "flash_attn: k/v layout mismatch");
   int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
   int Sk = k.size(1), Hkv = k.size(2);
   int win = window_size > 0 ? (int)window_size : 0;
@@ -809,12 +828,8 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
   auto dv = empty_like_strided(v);
   auto delta = torch::empty({B, Hq, Sq}, q.options().dtype(torch::kFloat32));
   auto stream = c10::hip::getCurrentHIPStream();
-  const long q_bs = q.stride(0), q_ss = q.stride(1);
-  const long kv_bs = k.stride(0), kv_ss = k.stride(1);
-  const long do_bs = dout.stride(0), do_ss = dout.stride(1);
-  const long o_bs = out.stride(0), o_ss = out.stride(1);
-  const long dq_bs = dq.stride(0), dq_ss = dq.stride(1);
-  const long dkv_bs = dk.stride(0), dkv_ss = dk.stride(1);
+  const TStr qs = tstr(q), ks = tstr(k), vs = tstr(v), ds = tstr(dout),
+             os = tstr(out), dqs = tstr(dq), dks = tstr(dk), dvs = tstr(dv);
 
   long rows = (long)B * Hq * Sq;
 #define LAUNCH_BWD(DD)                                                        \
@@ -822,8 +837,7 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
     hipLaunchKernelGGL((fa_bwd_delta_kernel<DD>), dim3(rows), dim3(64), 0,    \
                        stream, (const __hip_bfloat16*)dout.data_ptr(),        \
                        (const __hip_bfloat16*)out.data_ptr(),                 \
-                       delta.data_ptr<float>(), B, Sq, Hq, do_bs, do_ss,      \
-                       o_bs, o_ss);                                           \
+                       delta.data_ptr<float>(), B, Sq, Hq, ds, os);           \
     dim3 gridk((Sk + 12 * 16 - 1) / (12 * 16), Hkv, B);                       \
     hipLaunchKernelGGL((fa_bwd_dkdv_kernel<DD, 12>), gridk, dim3(12 * 64), 0, \
                        stream, (const __hip_bfloat16*)q.data_ptr(),           \
@@ -833,8 +847,8 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
                        lse.data_ptr<float>(), delta.data_ptr<float>(),        \
                        (__hip_bfloat16*)dk.data_ptr(),                        \
                        (__hip_bfloat16*)dv.data_ptr(), B, Sq, Sk, Hq, Hkv,    \
-                       (float)softmax_scale, causal ? 1 : 0, win, q_bs, q_ss, \
-                       kv_bs, kv_ss, do_bs, do_ss, dkv_bs, dkv_ss);           \
+                       (float)softmax_scale, causal ? 1 : 0, win, qs, ks, vs, \
+                       ds, dks, dvs);                                         \
     dim3 gridq((Sq + 12 * 16 - 1) / (12 * 16), Hq, B);                        \
     hipLaunchKernelGGL((fa_bwd_dq_kernel<DD, 12>), gridq, dim3(12 * 64), 0,   \
                        stream, (const __hip_bfloat16*)q.data_ptr(),           \
@@ -843,8 +857,8 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
                        (const __hip_bfloat16*)dout.data_ptr(),                \
                        lse.data_ptr<float>(), delta.data_ptr<float>(),        \
                        (__hip_bfloat16*)dq.data_ptr(), B, Sq, Sk, Hq, Hkv,    \
-                       (float)softmax_scale, causal ? 1 : 0, win, q_bs, q_ss, \
-                       kv_bs, kv_ss, do_bs, do_ss, dq_bs, dq_ss);             \
+                       (float)softmax_scale, causal ? 1 : 0, win, qs, ks, vs, \
+                       ds, dqs);                                              \
   } while (0)
 
   if (D == 128) {

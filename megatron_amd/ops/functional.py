@@ -237,7 +237,10 @@ class RoPEFunction(torch.autograd.Function):
         ctx.save_for_backward(cos, sin)
         ext = _C(x)
         if ext is not None:
-            return ext.rope_fwd(x.contiguous(), cos, sin)
+            # kernel reads strided views (dense head_dim) directly
+            if x.stride(-1) != 1:
+                x = x.contiguous()
+            return ext.rope_fwd(x, cos, sin)
         xf = x.float()
         x1 = xf[..., 0::2]
         x2 = xf[..., 1::2]
@@ -253,7 +256,9 @@ class RoPEFunction(torch.autograd.Function):
         cos, sin = ctx.saved_tensors
         ext = _C(dy)
         if ext is not None:
-            return ext.rope_bwd(dy.contiguous(), cos, sin), None, None
+            if dy.stride(-1) != 1:
+                dy = dy.contiguous()
+            return ext.rope_bwd(dy, cos, sin), None, None
         dyf = dy.float()
         g1 = dyf[..., 0::2]
         g2 = dyf[..., 1::2]
@@ -350,16 +355,17 @@ class FlashAttnFunction(torch.autograd.Function):
         ctx.softmax_scale = softmax_scale
         ctx.window_size = window_size
         if ext is not None:
-            # kernels are stride-aware over batch/seq (heads/head_dim must be
-            # the dense trailing dims): sbhd-transposed views pass with no
-            # copy; anything else (e.g. a sliced kv) still needs contiguous
+            # kernels are stride-aware over batch/seq/head (head_dim must be
+            # dense, strides 16B-aligned): sbhd transposes and QKV-projection
+            # head slices pass with no copy
             def ok(t):
-                return t.stride(-1) == 1 and t.stride(-2) == t.shape[-1]
+                return t.stride(-1) == 1 and all(
+                    t.stride(i) % 8 == 0 for i in range(3)
+                )
 
             q = q if ok(q) else q.contiguous()
             k = k if ok(k) else k.contiguous()
-            v = v if ok(v) and k.stride() == v.stride() else v.contiguous()
-            k = k if k.stride() == v.stride() else k.contiguous()
+            v = v if ok(v) else v.contiguous()
             out, lse = ext.flash_attn_fwd(
                 q, k, v,
                 bool(causal), float(softmax_scale),
@@ -377,7 +383,7 @@ class FlashAttnFunction(torch.autograd.Function):
         ext = _C(dout)
         if ext is not None:
             if not (dout.stride(-1) == 1
-                    and dout.stride(-2) == dout.shape[-1]):
+                    and all(dout.stride(i) % 8 == 0 for i in range(3))):
                 dout = dout.contiguous()
             dq, dk, dv = ext.flash_attn_bwd(
                 dout, q, k, v, out, lse,
