@@ -315,6 +315,24 @@ class ParallelAttention(MegatronModule):
             ngroups = nkv
             nq_per_group = np_ // nkv
             mixed = mixed.view(sq, b, ngroups, (nq_per_group + 2) * hn)
+            # training fast path: split + RoPE as ONE fused autograd node —
+            # q/k read as strided views, rotated by the stride-aware RoPE
+            # kernel, and the backward assembles d_mixed in a single buffer
+            # (no per-slice zeros+copy+add chains)
+            if (
+                mixed.is_cuda
+                and self.rope_cos is not None
+                and inference_params is None
+            ):
+                cos = self.rope_cos.to(device=mixed.device,
+                                       dtype=torch.float32)
+                sin = self.rope_sin.to(device=mixed.device,
+                                       dtype=torch.float32)
+                query, key, value = ops_f.fused_qkv_split_rope(
+                    mixed, cos[:sq], sin[:sq], np_, nkv, hn
+                )
+                return self._attend(query, key, value, attention_mask,
+                                    inference_params)
             # k/v (and q when nq_per_group == 1) stay uniformly-strided VIEWS
             # into the fused projection output: the FA and RoPE kernels are
             # stride-aware, so no slice-materializing copies are needed. A
@@ -378,14 +396,17 @@ class ParallelAttention(MegatronModule):
             key = k_cache[: start + sq]
             value = v_cache[: start + sq]
 
+        return self._attend(query, key, value, attention_mask,
+                            inference_params)
+
+    def _attend(self, query, key, value, attention_mask, inference_params):
+        sq = query.shape[0]
         use_flash = (
             self.use_flash_attn
             and self.attention_type == AttnType.self_attn
             and (inference_params is None or sq > 1)
         )
         if use_flash:
-            if self.n_rep > 1 and not ops_f._ext.available() and not query.is_cuda:
-                pass  # CPU reference handles GQA natively
             context = self.flash_attention(query, key, value)
         else:
             # expand kv heads for the unfused path

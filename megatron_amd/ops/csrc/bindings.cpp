@@ -31,6 +31,8 @@ torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor cosT,
                        torch::Tensor sinT);
 torch::Tensor rope_bwd(torch::Tensor dy, torch::Tensor cosT,
                        torch::Tensor sinT);
+torch::Tensor rope_bwd_into(torch::Tensor dy, torch::Tensor cosT,
+                            torch::Tensor sinT, torch::Tensor out);
 std::vector<torch::Tensor> bias_dropout_add_fwd(
     torch::Tensor x, c10::optional<torch::Tensor> bias,
     torch::Tensor residual, double p, int64_t seed, int64_t offset);
@@ -86,6 +88,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("glu_bwd", &glu_bwd);
   m.def("rope_fwd", &rope_fwd);
   m.def("rope_bwd", &rope_bwd);
+  m.def("rope_bwd_into", &rope_bwd_into);
   m.def("bias_dropout_add_fwd", &bias_dropout_add_fwd);
   m.def("dropout_bwd", &dropout_bwd);
   m.def("fused_adam", &fused_adam);

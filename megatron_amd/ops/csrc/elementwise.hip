@@ -145,10 +145,11 @@ __global__ void rope_kernel(const T* __restrict__ x,
                             const float* __restrict__ cosT,
                             const float* __restrict__ sinT,
                             T* __restrict__ y, long sbn, int b, int n,
-                            int h_half, long x_ss, long x_bs, long x_ns) {
-  // rows = s*b*n, each row has h = 2*h_half elements. x may be a strided
-  // VIEW (e.g. the q/k head slices of a fused QKV projection); y is written
-  // contiguous [s,b,n,h].
+                            int h_half, long x_ss, long x_bs, long x_ns,
+                            long y_ss, long y_bs, long y_ns) {
+  // rows = s*b*n, each row has h = 2*h_half elements. x and y may be strided
+  // VIEWS (e.g. the q/k head slices of a fused QKV projection or of its
+  // gradient buffer); head_dim is dense.
   for (long idx = (long)blockIdx.x * kBlock + threadIdx.x;
        idx < sbn * (long)h_half; idx += (long)gridDim.x * kBlock) {
     const long row = idx / h_half;
@@ -159,7 +160,8 @@ __global__ void rope_kernel(const T* __restrict__ x,
     const int s_idx = sb / b;
     const long x_base =
         (long)s_idx * x_ss + (long)b_idx * x_bs + (long)n_idx * x_ns + 2 * j;
-    const long y_base = row * (long)(2 * h_half) + 2 * j;
+    const long y_base =
+        (long)s_idx * y_ss + (long)b_idx * y_bs + (long)n_idx * y_ns + 2 * j;
     float c = cosT[(long)s_idx * h_half + j];
     float sn = sinT[(long)s_idx * h_half + j] * DIR;
     float x1 = DTypeTraits<T>::to_float(x[x_base]);
@@ -277,14 +279,23 @@ torch::Tensor glu_bwd(torch::Tensor dy, torch::Tensor x, int64_t mode) {
 }
 
 static torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cosT,
-                                torch::Tensor sinT, int dir) {
+                                torch::Tensor sinT, int dir,
+                                c10::optional<torch::Tensor> out = {}) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.stride(3) == 1,
               "rope: need 4-D x with dense head_dim");
   TORCH_CHECK(cosT.scalar_type() == torch::kFloat32);
   int s = x.size(0), b = x.size(1), n = x.size(2), h = x.size(3);
   TORCH_CHECK(cosT.size(0) >= s && cosT.size(1) == h / 2,
               "rope table too small");
-  auto y = torch::empty({s, b, n, h}, x.options());
+  torch::Tensor y;
+  if (out.has_value()) {
+    y = *out;
+    TORCH_CHECK(y.sizes() == x.sizes() && y.stride(3) == 1 &&
+                    y.scalar_type() == x.scalar_type(),
+                "rope: bad out tensor");
+  } else {
+    y = torch::empty({s, b, n, h}, x.options());
+  }
   long sbn = (long)s * b * n;
   long total = sbn * (h / 2);
   auto stream = c10::hip::getCurrentHIPStream();
@@ -295,13 +306,15 @@ static torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cosT,
                      dim3(kBlock), 0, stream, (const T*)x.data_ptr(),        \
                      cosc.data_ptr<float>(), sinc.data_ptr<float>(),         \
                      (T*)y.data_ptr(), sbn, b, n, h / 2, x.stride(0),        \
-                     x.stride(1), x.stride(2))
+                     x.stride(1), x.stride(2), y.stride(0), y.stride(1),     \
+                     y.stride(2))
 #define LAUNCH_ROPE_B(T)                                                     \
   hipLaunchKernelGGL((rope_kernel<T, -1>), dim3(grid_for(total)),            \
                      dim3(kBlock), 0, stream, (const T*)x.data_ptr(),        \
                      cosc.data_ptr<float>(), sinc.data_ptr<float>(),         \
                      (T*)y.data_ptr(), sbn, b, n, h / 2, x.stride(0),        \
-                     x.stride(1), x.stride(2))
+                     x.stride(1), x.stride(2), y.stride(0), y.stride(1),     \
+                     y.stride(2))
   if (dir > 0) {
     DISPATCH_DTYPE(x, LAUNCH_ROPE_F);
   } else {
@@ -318,6 +331,14 @@ torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor cosT,
 torch::Tensor rope_bwd(torch::Tensor dy, torch::Tensor cosT,
                        torch::Tensor sinT) {
   return rope_apply(dy, cosT, sinT, -1);
+}
+
+// rotate dy and write the result into `out` (typically a strided slice of a
+// fused-QKV gradient buffer) — lets the backward of the QKV split assemble
+// d_mixed in one pass with no zeros+copy+add chain.
+torch::Tensor rope_bwd_into(torch::Tensor dy, torch::Tensor cosT,
+                            torch::Tensor sinT, torch::Tensor out) {
+  return rope_apply(dy, cosT, sinT, -1, out);
 }
 
 std::vector<torch::Tensor> bias_dropout_add_fwd(

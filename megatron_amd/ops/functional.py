@@ -454,3 +454,66 @@ def flash_attention(q, k, v, causal=True, softmax_scale=None, window_size=None,
     return FlashAttnFunction.apply(
         q, k, v, causal, softmax_scale, window_size, dropout_p, training
     )
+
+
+class FusedQKVSplitRope(torch.autograd.Function):
+    """Split the fused QKV projection output [s,b,g,(nq+2)h] into q/k/v and
+    rotate q/k, assembling d_mixed in ONE buffer in the backward.
+
+    Plain autograd slicing costs 3 zeros-fills + 3 copies + 2 full-size adds
+    per layer in the backward (narrow-backward per slice, then fan-in sums);
+    here the stride-aware RoPE kernel writes d(q)/d(k) straight into the
+    corresponding regions of d_mixed and d(v) is one strided copy. q/k are
+    read as strided views (no forward copies); v is cloned so no
+    view-of-input is returned from the Function.
+    """
+
+    @staticmethod
+    def forward(ctx, mixed, cos, sin, np_, nkv, hn):
+        ext = _C(mixed)
+        sq, b, g, _ = mixed.shape
+        nq = np_ // nkv
+        q = mixed[..., : nq * hn]
+        if nq > 1:
+            q = q.reshape(sq, b, np_, hn)
+        k = mixed[..., nq * hn : (nq + 1) * hn]
+        v = mixed[..., (nq + 1) * hn :]
+        if cos is not None:
+            q_rot = ext.rope_fwd(q, cos, sin)
+            k_rot = ext.rope_fwd(k, cos, sin)
+        else:
+            q_rot = q.contiguous()
+            k_rot = k.contiguous()
+        ctx.save_for_backward(cos, sin) if cos is not None else ctx.save_for_backward()
+        ctx.dims = (sq, b, g, nq, hn, np_, nkv, cos is not None)
+        return q_rot, k_rot, v.clone()
+
+    @staticmethod
+    def backward(ctx, dq, dk, dv):
+        ext = _C(dq)
+        sq, b, g, nq, hn, np_, nkv, have_rope = ctx.dims
+        d_mixed = torch.empty(sq, b, g, (nq + 2) * hn, dtype=dq.dtype,
+                              device=dq.device)
+        dq_slice = d_mixed[..., : nq * hn]
+        dk_slice = d_mixed[..., nq * hn : (nq + 1) * hn]
+        dv_slice = d_mixed[..., (nq + 1) * hn :]
+        if have_rope:
+            cos, sin = ctx.saved_tensors
+            if nq == 1:
+                ext.rope_bwd_into(dq.contiguous(), cos, sin, dq_slice)
+            else:
+                # GQA q region is not uniformly strided over heads: rotate
+                # densely, then one strided copy
+                dq_slice.view(sq, b, np_, hn).copy_(
+                    ext.rope_bwd(dq.contiguous(), cos, sin)
+                )
+            ext.rope_bwd_into(dk.contiguous(), cos, sin, dk_slice)
+        else:
+            dq_slice.view(sq, b, np_, hn).copy_(dq)
+            dk_slice.copy_(dk)
+        dv_slice.copy_(dv)
+        return d_mixed, None, None, None, None, None
+
+
+def fused_qkv_split_rope(mixed, cos, sin, np_, nkv, hn):
+    return FusedQKVSplitRope.apply(mixed, cos, sin, np_, nkv, hn)

@@ -317,3 +317,46 @@ class TestFlashAttention:
         assert torch.equal(dq_v.contiguous(), dq_c)
         assert torch.equal(dk_v.contiguous(), dk_c)
         assert torch.equal(dv_v.contiguous(), dv_c)
+
+
+class TestFusedQKVSplitRope:
+    @pytest.mark.parametrize("nq_per_group", [1, 4])
+    def test_matches_unfused(self, nq_per_group):
+        """Fused split+rope (single-buffer d_mixed backward) vs plain
+        autograd slicing + rope: same outputs, same d_mixed."""
+        from megatron_amd.models.rope import precompute_freqs
+        from megatron_amd.ops import functional as F
+
+        s, b, g, hn = 128, 2, 2, 64
+        np_ = g * nq_per_group
+        nkv = g
+        torch.manual_seed(0)
+        mixed = torch.randn(s, b, g, (nq_per_group + 2) * hn, device="cuda",
+                            dtype=torch.bfloat16, requires_grad=True)
+        mixed2 = mixed.detach().clone().requires_grad_(True)
+        cos, sin = precompute_freqs(hn, s)
+        cos = cos.cuda().float()
+        sin = sin.cuda().float()
+
+        q1, k1, v1 = F.fused_qkv_split_rope(mixed, cos, sin, np_, nkv, hn)
+
+        q2 = mixed2[..., : nq_per_group * hn].reshape(s, b, np_, hn)
+        k2 = mixed2[..., nq_per_group * hn : (nq_per_group + 1) * hn]
+        v2 = mixed2[..., (nq_per_group + 1) * hn :]
+        q2r = F.apply_rope(q2, cos, sin)
+        k2r = F.apply_rope(k2, cos, sin)
+
+        assert torch.equal(q1.reshape(s, b, np_, hn), q2r)
+        assert torch.equal(k1, k2r.contiguous())
+        assert torch.equal(v1, v2.contiguous())
+
+        gq = torch.randn_like(q2r)
+        gk = torch.randn_like(k2r)
+        gv = torch.randn_like(v2)
+        (q1.reshape(s, b, np_, hn) * gq).sum().backward(retain_graph=True)
+        (k1 * gk).sum().backward(retain_graph=True)
+        (v1 * gv).sum().backward()
+        (q2r * gq).sum().backward(retain_graph=True)
+        (k2r * gk).sum().backward(retain_graph=True)
+        (v2 * gv).sum().backward()
+        assert torch.equal(mixed.grad, mixed2.grad)
