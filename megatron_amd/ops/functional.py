@@ -503,13 +503,18 @@ class FusedQKVSplitRope(torch.autograd.Function):
                 ext.rope_bwd_into(dq.contiguous(), cos, sin, dq_slice)
             else:
                 # GQA q region is not uniformly strided over heads: rotate
-                # densely, then one strided copy
-                dq_slice.view(sq, b, np_, hn).copy_(
-                    ext.rope_bwd(dq.contiguous(), cos, sin)
+                # densely, then one strided copy (5-D view splits the dense
+                # nq*hn run of each group)
+                dq_slice.view(sq, b, g, nq, hn).copy_(
+                    ext.rope_bwd(dq.contiguous(), cos, sin).view(
+                        sq, b, g, nq, hn
+                    )
                 )
             ext.rope_bwd_into(dk.contiguous(), cos, sin, dk_slice)
         else:
-            dq_slice.view(sq, b, np_, hn).copy_(dq)
+            dq_slice.view(sq, b, g, nq, hn).copy_(
+                dq.reshape(sq, b, g, nq, hn)
+            )
             dk_slice.copy_(dk)
         dv_slice.copy_(dv)
         return d_mixed, None, None, None, None, None
