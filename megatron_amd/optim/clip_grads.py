@@ -16,11 +16,29 @@ from ..utils import param_is_not_tensor_parallel_duplicate
 
 
 def clip_grad_norm_fp32(parameters, grads_for_norm, max_norm, norm_type=2,
-                        model_parallel_group=None):
+                        model_parallel_group=None, flat_buffers=None):
+    """flat_buffers: optional list of contiguous fp32 grad buffers that tile
+    exactly the same elements as grads_for_norm (valid when no param is
+    excluded as a TP duplicate or PP-shared copy, i.e. TP=PP=1, and padding
+    regions are zero). The L2 norm and the clip scale then run as ONE flat
+    kernel per buffer instead of a multi-tensor chain over every param."""
     if isinstance(parameters, torch.Tensor):
         parameters = [parameters]
     if isinstance(grads_for_norm, torch.Tensor):
         grads_for_norm = [grads_for_norm]
+
+    if flat_buffers and norm_type == 2:
+        total_norm = sum(b.pow(2).sum() for b in flat_buffers)
+        torch.distributed.all_reduce(
+            total_norm, op=torch.distributed.ReduceOp.SUM,
+            group=model_parallel_group,
+        )
+        total_norm = total_norm.item() ** 0.5
+        clip_coeff = float(max_norm) / (total_norm + 1.0e-6)
+        if clip_coeff < 1.0:
+            for b in flat_buffers:
+                b.mul_(clip_coeff)
+        return total_norm
 
     grads = []
     for param in parameters:

@@ -42,6 +42,12 @@ class Range:
 
 
 class DistributedOptimizer(MixedPrecisionOptimizer):
+    def _flat_grad_buffers(self):
+        # ZeRO-1 owns only a shard: after the in-place reduce-scatter the
+        # non-owned buffer regions hold stale local grads, so the whole-buffer
+        # norm shortcut is invalid — clip over the shard views instead
+        return None
+
     @classmethod
     def build_model_gbuf_param_range_map(cls, model, dtype, gbuf_world_range):
         param_world_index_map = model._grad_buffer_param_index_map[dtype]

@@ -80,12 +80,34 @@ class MegatronOptimizer(ABC):
     def get_model_parallel_group(self):
         return mpu.get_model_parallel_group()
 
+    def _flat_grad_buffers(self):
+        """The whole-model fp32 grad buffers, when they tile exactly the
+        grads the norm covers: TP=PP=1 (no duplicate exclusions), contiguous
+        local-DDP buffers, fp32 accumulation. Dead pad regions stay zero."""
+        if (
+            mpu.get_tensor_model_parallel_world_size() != 1
+            or mpu.get_pipeline_model_parallel_world_size() != 1
+            or not self.use_contiguous_buffers_in_local_ddp
+        ):
+            return None
+        bufs = []
+        for model in self.models:
+            gb = getattr(model, "_grad_buffers", None)
+            if not gb:
+                return None
+            for dtype, buf in gb.items():
+                if dtype != torch.float:
+                    return None
+                bufs.append(buf.data)
+        return bufs or None
+
     def clip_grad_norm(self, clip_grad):
         params = self.get_parameters()
         grads_for_norm = self.get_main_grads_for_grad_norm()
         return clip_grad_norm_fp32(
             params, grads_for_norm, clip_grad,
             model_parallel_group=self.get_model_parallel_group(),
+            flat_buffers=self._flat_grad_buffers(),
         )
 
     def count_zeros(self):
