@@ -276,6 +276,15 @@ def main():
                 "parallelism": f"tp{tp}_pp{pp}_dp{dp}",
             },
         }
+    else:
+        result = None
+
+    # tear down BEFORE printing: RCCL writes a version banner to stdout at
+    # communicator destruction, which must not land after the JSON line
+    if torch.distributed.is_initialized():
+        torch.distributed.destroy_process_group()
+    if result is not None:
+        sys.stdout.flush()
         print(json.dumps(result), flush=True)
 
 
