@@ -18,6 +18,10 @@ import os
 import sys
 import time
 
+# avoid caching-allocator fragmentation at the 288 GB capacity edge (the
+# 32k-seq configs sit within ~1% of it)
+os.environ.setdefault("PYTORCH_ALLOC_CONF", "expandable_segments:True")
+
 # hipBLASLt TunableOp: use the pre-tuned GEMM algo selection committed under
 # profiles/ when present (tools/tune_gemms.py produces it); read-only mode.
 _TUNED = os.path.join(os.path.dirname(os.path.abspath(__file__)),
@@ -286,6 +290,12 @@ def main():
     if result is not None:
         sys.stdout.flush()
         print(json.dumps(result), flush=True)
+    # exit without running C-level exit handlers: RCCL/HIP flush version
+    # banners to stdout at library unload, which would land after the JSON
+    # contract line
+    sys.stdout.flush()
+    sys.stderr.flush()
+    os._exit(0)
 
 
 if __name__ == "__main__":
