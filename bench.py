@@ -66,11 +66,14 @@ MODEL_SPECS = {
     "llama2-70b": dict(num_layers=80, hidden_size=8192, ffn_hidden_size=28672,
                        num_attention_heads=64, num_attention_heads_kv=8,
                        vocab=32000, seq=4096, model_name="llama2"),
-    # Mistral-7B: 32 layers, h=4096, ffn=14336, 32 heads, 8 kv, SWA 4096
+    # Mistral-7B: 32 layers, h=4096, ffn=14336, 32 heads, 8 kv, SWA 4096.
+    # seq-32k activations exceed 288 GB with selective recompute (~290 GB
+    # measured), so this config checkpoints layers (full recompute)
     "mistral-7b": dict(num_layers=32, hidden_size=4096, ffn_hidden_size=14336,
                        num_attention_heads=32, num_attention_heads_kv=8,
                        vocab=32000, seq=32768, model_name="mistral",
-                       sliding_window_size=4096, rope_scaling_factor=4.0),
+                       sliding_window_size=4096, rope_scaling_factor=4.0,
+                       recompute=True),
     # Falcon-7B: 32 layers, h=4544, 71 heads, MQA (1 kv head), parallel attn
     "falcon-7b": dict(num_layers=32, hidden_size=4544, ffn_hidden_size=18176,
                       num_attention_heads=71, num_attention_heads_kv=1,
@@ -137,8 +140,13 @@ def main():
         lr_warmup_iters=0, clip_grad=1.0,
         hidden_dropout=0.0, attention_dropout=0.0,
         use_flash_attn=True,
-        recompute_granularity="selective" if not args.recompute else "full",
-        recompute_method="uniform" if args.recompute else None,
+        recompute_granularity=(
+            "full" if (args.recompute or spec.get("recompute"))
+            else "selective"
+        ),
+        recompute_method=(
+            "uniform" if (args.recompute or spec.get("recompute")) else None
+        ),
         sequence_parallel=(tp > 1),
         # bucketed async all-reduce hidden behind backward beats ZeRO-1's
         # exposed whole-buffer reduce-scatter + all-gather here: 288 GB HBM3E

@@ -486,7 +486,9 @@ class FusedQKVSplitRope(torch.autograd.Function):
             k_rot = k.contiguous()
         ctx.save_for_backward(cos, sin) if cos is not None else ctx.save_for_backward()
         ctx.dims = (sq, b, g, nq, hn, np_, nkv, cos is not None)
-        return q_rot, k_rot, v.clone()
+        # v is returned as a strided VIEW of the input (no copy, no extra
+        # memory — ~2 GB at seq 32k); autograd tracks it as a view output
+        return q_rot, k_rot, v
 
     @staticmethod
     def backward(ctx, dq, dk, dv):
