@@ -67,13 +67,12 @@ MODEL_SPECS = {
                        num_attention_heads=64, num_attention_heads_kv=8,
                        vocab=32000, seq=4096, model_name="llama2"),
     # Mistral-7B: 32 layers, h=4096, ffn=14336, 32 heads, 8 kv, SWA 4096.
-    # seq-32k activations exceed 288 GB with selective recompute (~290 GB
-    # measured), so this config checkpoints layers (full recompute)
+    # seq-32k activations cap the micro-batch at 1 on 288 GB
     "mistral-7b": dict(num_layers=32, hidden_size=4096, ffn_hidden_size=14336,
                        num_attention_heads=32, num_attention_heads_kv=8,
                        vocab=32000, seq=32768, model_name="mistral",
                        sliding_window_size=4096, rope_scaling_factor=4.0,
-                       recompute=True),
+                       mbs=1),
     # Falcon-7B: 32 layers, h=4544, 71 heads, MQA (1 kv head), parallel attn
     "falcon-7b": dict(num_layers=32, hidden_size=4544, ffn_hidden_size=18176,
                       num_attention_heads=71, num_attention_heads_kv=1,
@@ -114,7 +113,7 @@ def main():
     dp = world_size // (tp * pp)
     # mbs sweep on MI355X (profiles/r01): 15.5k @ mbs4, 16.7k @ mbs8 with
     # the wide-workgroup FA kernels
-    mbs = args.micro_batch_size or (8 if have_gpu else 1)
+    mbs = args.micro_batch_size or spec.get("mbs") or (8 if have_gpu else 1)
     # pp > 1 needs several in-flight microbatches to fill the 1F1B pipeline
     gbs = args.global_batch or (mbs * dp * (2 * pp if pp > 1 else 1))
 
