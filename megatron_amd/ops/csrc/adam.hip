@@ -22,10 +22,13 @@ __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
                             float* __restrict__ m, float* __restrict__ v,
                             OutT* __restrict__ model_out, long n, float lr,
                             float beta1, float beta2, float eps, float wd,
-                            float bc1, float bc2_sqrt, int adam_w) {
+                            float bc1, float bc2_sqrt, int adam_w,
+                            float grad_scale) {
+  // grad_scale folds the grad-clip coefficient (and any unscale factor)
+  // into this pass, avoiding a separate whole-buffer multiply
   for (long i = (long)blockIdx.x * kBlock + threadIdx.x; i < n;
        i += (long)gridDim.x * kBlock) {
-    float grad = g[i];
+    float grad = g[i] * grad_scale;
     float param = p[i];
     if (!adam_w && wd != 0.f) grad += wd * param;
     float m_new = beta1 * m[i] + (1.f - beta1) * grad;
@@ -54,7 +57,7 @@ void fused_adam(std::vector<torch::Tensor> params,
                 std::vector<torch::Tensor> exp_avgs,
                 std::vector<torch::Tensor> exp_avg_sqs, double lr,
                 double beta1, double beta2, double eps, double wd,
-                int64_t step, int64_t adam_w_mode) {
+                int64_t step, int64_t adam_w_mode, double grad_scale) {
   auto stream = c10::hip::getCurrentHIPStream();
   float bc1 = 1.f - powf((float)beta1, (float)step);
   float bc2_sqrt = sqrtf(1.f - powf((float)beta2, (float)step));
@@ -68,7 +71,8 @@ void fused_adam(std::vector<torch::Tensor> params,
                        exp_avgs[i].data_ptr<float>(),
                        exp_avg_sqs[i].data_ptr<float>(), nullptr, n,
                        (float)lr, (float)beta1, (float)beta2, (float)eps,
-                       (float)wd, bc1, bc2_sqrt, (int)adam_w_mode);
+                       (float)wd, bc1, bc2_sqrt, (int)adam_w_mode,
+                       (float)grad_scale);
   }
 }
 
@@ -82,7 +86,7 @@ void fused_adam_with_model_copy(std::vector<torch::Tensor> params,
                                 std::vector<torch::Tensor> model_params,
                                 double lr, double beta1, double beta2,
                                 double eps, double wd, int64_t step,
-                                int64_t adam_w_mode) {
+                                int64_t adam_w_mode, double grad_scale) {
   auto stream = c10::hip::getCurrentHIPStream();
   float bc1 = 1.f - powf((float)beta1, (float)step);
   float bc2_sqrt = sqrtf(1.f - powf((float)beta2, (float)step));
@@ -99,7 +103,8 @@ void fused_adam_with_model_copy(std::vector<torch::Tensor> params,
                          exp_avg_sqs[i].data_ptr<float>(),
                          (__hip_bfloat16*)model_params[i].data_ptr(), n,
                          (float)lr, (float)beta1, (float)beta2, (float)eps,
-                         (float)wd, bc1, bc2_sqrt, (int)adam_w_mode);
+                         (float)wd, bc1, bc2_sqrt, (int)adam_w_mode,
+                         (float)grad_scale);
     } else if (model_params[i].scalar_type() == torch::kFloat16) {
       hipLaunchKernelGGL(adam_kernel<__half>, dim3(grid_for(n)),
                          dim3(kBlock), 0, stream,
@@ -109,7 +114,8 @@ void fused_adam_with_model_copy(std::vector<torch::Tensor> params,
                          exp_avg_sqs[i].data_ptr<float>(),
                          (__half*)model_params[i].data_ptr(), n, (float)lr,
                          (float)beta1, (float)beta2, (float)eps, (float)wd,
-                         bc1, bc2_sqrt, (int)adam_w_mode);
+                         bc1, bc2_sqrt, (int)adam_w_mode,
+                         (float)grad_scale);
     } else {
       TORCH_CHECK(false, "model params must be bf16/fp16");
     }

@@ -44,7 +44,7 @@ void fused_adam(std::vector<torch::Tensor> params,
                 std::vector<torch::Tensor> exp_avgs,
                 std::vector<torch::Tensor> exp_avg_sqs, double lr,
                 double beta1, double beta2, double eps, double wd,
-                int64_t step, int64_t adam_w_mode);
+                int64_t step, int64_t adam_w_mode, double grad_scale);
 void fused_adam_with_model_copy(std::vector<torch::Tensor> params,
                                 std::vector<torch::Tensor> grads,
                                 std::vector<torch::Tensor> exp_avgs,
@@ -52,7 +52,7 @@ void fused_adam_with_model_copy(std::vector<torch::Tensor> params,
                                 std::vector<torch::Tensor> model_params,
                                 double lr, double beta1, double beta2,
                                 double eps, double wd, int64_t step,
-                                int64_t adam_w_mode);
+                                int64_t adam_w_mode, double grad_scale);
 
 // wgrad.hip
 void wgrad_gemm_accum_fp32(torch::Tensor input, torch::Tensor grad_output,

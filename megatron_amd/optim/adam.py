@@ -22,8 +22,13 @@ class FusedAdam(torch.optim.Optimizer):
         super().__init__(params, defaults)
         self.adam_w_mode = adam_w_mode
 
+    supports_grad_scale = True
+
     @torch.no_grad()
-    def step(self, closure=None):
+    def step(self, closure=None, grad_scale=1.0):
+        """grad_scale multiplies grads inside the fused kernel — the
+        mixed-precision optimizer passes the deferred clip coefficient here
+        so clipping costs no extra memory pass."""
         self.wrote_model_params = False
         loss = None
         if closure is not None:
@@ -73,17 +78,19 @@ class FusedAdam(torch.optim.Optimizer):
                         [v.view(-1) for v in exp_avg_sqs],
                         [m.view(-1) for m in model_outs],
                         lr, beta1, beta2, eps, wd, step,
-                        1 if self.adam_w_mode else 0,
+                        1 if self.adam_w_mode else 0, float(grad_scale),
                     )
                     self.wrote_model_params = True
                 else:
                     ext.fused_adam(
                         params, grads, exp_avgs, exp_avg_sqs,
                         lr, beta1, beta2, eps, wd, step,
-                        1 if self.adam_w_mode else 0,
+                        1 if self.adam_w_mode else 0, float(grad_scale),
                     )
                 continue
 
+            if grad_scale != 1.0:
+                grads = list(torch._foreach_mul(grads, grad_scale))
             if self.adam_w_mode and wd != 0.0:
                 torch._foreach_mul_(params, 1 - lr * wd)
             elif wd != 0.0:

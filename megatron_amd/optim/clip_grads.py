@@ -16,7 +16,8 @@ from ..utils import param_is_not_tensor_parallel_duplicate
 
 
 def clip_grad_norm_fp32(parameters, grads_for_norm, max_norm, norm_type=2,
-                        model_parallel_group=None, flat_buffers=None):
+                        model_parallel_group=None, flat_buffers=None,
+                        defer_scale=False):
     """flat_buffers: optional list of contiguous fp32 grad buffers that tile
     exactly the same elements as grads_for_norm (valid when no param is
     excluded as a TP duplicate or PP-shared copy, i.e. TP=PP=1, and padding
@@ -34,7 +35,10 @@ def clip_grad_norm_fp32(parameters, grads_for_norm, max_norm, norm_type=2,
             group=model_parallel_group,
         )
         total_norm = total_norm.item() ** 0.5
-        clip_coeff = float(max_norm) / (total_norm + 1.0e-6)
+        clip_coeff = min(1.0, float(max_norm) / (total_norm + 1.0e-6))
+        if defer_scale:
+            # caller folds the coefficient into the optimizer kernel
+            return total_norm, clip_coeff
         if clip_coeff < 1.0:
             for b in flat_buffers:
                 b.mul_(clip_coeff)

@@ -101,13 +101,14 @@ class MegatronOptimizer(ABC):
                 bufs.append(buf.data)
         return bufs or None
 
-    def clip_grad_norm(self, clip_grad):
+    def clip_grad_norm(self, clip_grad, defer_scale=False):
         params = self.get_parameters()
         grads_for_norm = self.get_main_grads_for_grad_norm()
         return clip_grad_norm_fp32(
             params, grads_for_norm, clip_grad,
             model_parallel_group=self.get_model_parallel_group(),
             flat_buffers=self._flat_grad_buffers(),
+            defer_scale=defer_scale,
         )
 
     def count_zeros(self):
@@ -325,8 +326,22 @@ class MixedPrecisionOptimizer(MegatronOptimizer):
 
         _t("optimizer-clip-main-grad", True)
         grad_norm = None
+        grad_scale = 1.0
+        # with a flat grad buffer and a kernel that accepts a grad multiplier,
+        # the clip coefficient folds into the Adam pass (no extra 2x-buffer
+        # memory sweep); the grad scaler path keeps the eager order
+        can_defer = (
+            self.grad_scaler is None
+            and getattr(self.optimizer, "supports_grad_scale", False)
+            and self._flat_grad_buffers() is not None
+        )
         if self.clip_grad > 0.0:
-            grad_norm = self.clip_grad_norm(self.clip_grad)
+            if can_defer:
+                grad_norm, grad_scale = self.clip_grad_norm(
+                    self.clip_grad, defer_scale=True
+                )
+            else:
+                grad_norm = self.clip_grad_norm(self.clip_grad)
         _t("optimizer-clip-main-grad", False)
 
         _t("optimizer-count-zeros", True)
@@ -334,7 +349,10 @@ class MixedPrecisionOptimizer(MegatronOptimizer):
         _t("optimizer-count-zeros", False)
 
         _t("optimizer-inner-step", True)
-        self.optimizer.step()
+        if grad_scale != 1.0:
+            self.optimizer.step(grad_scale=grad_scale)
+        else:
+            self.optimizer.step()
         _t("optimizer-inner-step", False)
 
         _t("optimizer-copy-main-to-model-params", True)
