@@ -208,7 +208,7 @@ class TestAdam:
                                 eps=1e-8, weight_decay=0.01)
         for step in range(1, 4):
             ext.fused_adam([p], [g], [m], [v], 1e-3, 0.9, 0.999, 1e-8, 0.01,
-                           step, 1)
+                           step, 1, 1.0)
             p_ref.grad = g.clone()
             opt.step()
         assert rel_err(p, p_ref.detach()) < 1e-4
@@ -360,3 +360,25 @@ class TestFusedQKVSplitRope:
         (k2r * gk).sum().backward(retain_graph=True)
         (v2 * gv).sum().backward()
         assert torch.equal(mixed.grad, mixed2.grad)
+
+
+class TestAdamGradScale:
+    def test_grad_scale_matches_prescaled(self):
+        """fused_adam(grad_scale=c) must equal fused_adam on grads*c (the
+        deferred clip-coefficient fold)."""
+        ext = _ext()
+        n = 4099
+        torch.manual_seed(3)
+        p1 = torch.randn(n, device="cuda")
+        p2 = p1.clone()
+        g = torch.randn(n, device="cuda")
+        m1 = torch.zeros(n, device="cuda"); v1 = torch.zeros(n, device="cuda")
+        m2 = torch.zeros(n, device="cuda"); v2 = torch.zeros(n, device="cuda")
+        c = 0.37
+        ext.fused_adam([p1], [g], [m1], [v1], 1e-3, 0.9, 0.999, 1e-8, 0.01,
+                       1, 1, c)
+        ext.fused_adam([p2], [g * c], [m2], [v2], 1e-3, 0.9, 0.999, 1e-8,
+                       0.01, 1, 1, 1.0)
+        assert torch.equal(p1, p2)
+        assert torch.equal(m1, m2)
+        assert torch.equal(v1, v2)
