@@ -299,10 +299,12 @@ def main():
         print(json.dumps(result), flush=True)
     # exit without running C-level exit handlers: RCCL/HIP flush version
     # banners to stdout at library unload, which would land after the JSON
-    # contract line
+    # contract line. rocprofv3 needs its exit handlers to write the trace —
+    # set MEGATRON_AMD_CLEAN_EXIT=0 when profiling.
     sys.stdout.flush()
     sys.stderr.flush()
-    os._exit(0)
+    if os.environ.get("MEGATRON_AMD_CLEAN_EXIT", "1") != "0":
+        os._exit(0)
 
 
 if __name__ == "__main__":
