@@ -223,15 +223,38 @@ __global__ void norm_bwd_kernel_bf16v(const __hip_bfloat16* __restrict__ dy,
   }
 }
 
-// column-sum of the [grid][H] partial buffers
+// column-sum of the [grid][H] partial buffers. blockIdx.y slices the row
+// range so the read is parallel across CUs; a second launch with nblocks =
+// n_slices folds the slice partials — a fixed-order (deterministic) tree.
 __global__ void norm_bwd_reduce_kernel(const float* __restrict__ partial,
                                        float* __restrict__ out, int nblocks,
-                                       int H) {
+                                       int H, int rows_per_slice) {
   int i = blockIdx.x * 256 + threadIdx.x;
   if (i >= H) return;
+  int b0 = blockIdx.y * rows_per_slice;
+  int b1 = min(nblocks, b0 + rows_per_slice);
   float acc = 0.f;
-  for (int b = 0; b < nblocks; ++b) acc += partial[(long)b * H + i];
-  out[i] = acc;
+  for (int b = b0; b < b1; ++b) acc += partial[(long)b * H + i];
+  out[(long)blockIdx.y * H + i] = acc;
+}
+
+// helper: deterministic two-level column reduce of [nblocks][H] -> out[H]
+static void reduce_partials(const float* part, float* out, int nblocks, int H,
+                            torch::Tensor& tmp, hipStream_t stream) {
+  const int kSlices = 32;
+  if (nblocks <= 64) {
+    hipLaunchKernelGGL(norm_bwd_reduce_kernel, dim3((H + 255) / 256, 1),
+                       dim3(256), 0, stream, part, out, nblocks, H, nblocks);
+    return;
+  }
+  int rps = (nblocks + kSlices - 1) / kSlices;
+  int slices = (nblocks + rps - 1) / rps;
+  hipLaunchKernelGGL(norm_bwd_reduce_kernel, dim3((H + 255) / 256, slices),
+                     dim3(256), 0, stream, part, tmp.data_ptr<float>(),
+                     nblocks, H, rps);
+  hipLaunchKernelGGL(norm_bwd_reduce_kernel, dim3((H + 255) / 256, 1),
+                     dim3(256), 0, stream, tmp.data_ptr<float>(), out, slices,
+                     H, slices);
 }
 
 // Backward. MAX_ACC register accumulators per thread for dweight/dbias.
@@ -372,6 +395,7 @@ void norm_bwd_launch(const torch::Tensor& dy, const torch::Tensor& x,
       grid = (int)std::min<long>(rows, grid_env);
       auto opts = dy.options().dtype(torch::kFloat32);
       auto dw_part = torch::empty({grid, H}, opts);
+      auto red_tmp = torch::empty({32, H}, opts);
       torch::Tensor db_part;
       if (!RMS) db_part = torch::empty({grid, H}, opts);
       hipLaunchKernelGGL((norm_bwd_kernel_bf16v<RMS, kBlock, 4>), dim3(grid),
@@ -384,13 +408,11 @@ void norm_bwd_launch(const torch::Tensor& dy, const torch::Tensor& x,
                          (__hip_bfloat16*)dx.data_ptr(),
                          dw_part.data_ptr<float>(),
                          RMS ? nullptr : db_part.data_ptr<float>(), rows, H);
-      hipLaunchKernelGGL(norm_bwd_reduce_kernel, dim3((H + 255) / 256),
-                         dim3(256), 0, stream, dw_part.data_ptr<float>(),
-                         dw.data_ptr<float>(), grid, H);
+      reduce_partials(dw_part.data_ptr<float>(), dw.data_ptr<float>(), grid,
+                      H, red_tmp, stream);
       if (!RMS) {
-        hipLaunchKernelGGL(norm_bwd_reduce_kernel, dim3((H + 255) / 256),
-                           dim3(256), 0, stream, db_part.data_ptr<float>(),
-                           db->data_ptr<float>(), grid, H);
+        reduce_partials(db_part.data_ptr<float>(), db->data_ptr<float>(),
+                        grid, H, red_tmp, stream);
       }
       return;
     }
