@@ -100,6 +100,18 @@ class TrainingConfig:
     exit_duration_in_mins: Optional[int] = None
     exit_signal_handler: bool = False
     skip_iters: List[int] = field(default_factory=list)
+    adlr_autoresume: bool = False
+    adlr_autoresume_interval: int = 1000
+
+    # -- downstream tasks (tasks/main.py; reference tasks/main.py:24-48) --
+    task: Optional[str] = None
+    epochs: int = 0
+    pretrained_checkpoint: Optional[str] = None
+    train_data: Optional[List[str]] = None
+    valid_data: Optional[List[str]] = None
+    overlapping_eval: int = 32
+    keep_last: bool = False
+    strict_lambada: bool = False
 
     # -- parallelism --
     tensor_model_parallel_size: int = 1

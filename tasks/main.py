@@ -1,9 +1,9 @@
 """Downstream-task entry point (reference tasks/main.py): GLUE / RACE
 finetuning and zero-shot GPT evaluation.
 
-  python tasks/main.py --task MNLI --model_name bert ... (glue)
-  python tasks/main.py --task RACE ...                    (race)
-  python tasks/main.py --task LAMBADA|WIKITEXT103 ...     (zeroshot gpt)
+  python tasks/main.py --task MNLI|QQP ...        (glue classification)
+  python tasks/main.py --task RACE ...            (multi-choice)
+  python tasks/main.py --task LAMBADA|WIKITEXT103 (zero-shot gpt eval)
 """
 
 import os
@@ -23,31 +23,29 @@ def get_tasks_args(parser):
     group.add_argument("--train_data", nargs="*", default=None)
     group.add_argument("--valid_data", nargs="*", default=None)
     group.add_argument("--overlapping_eval", type=int, default=32)
+    group.add_argument("--keep_last", action="store_true")
+    group.add_argument("--strict_lambada", action="store_true")
     return parser
+
+
+def dispatch(task):
+    if task in ("MNLI", "QQP"):
+        from tasks.glue.finetune import main as glue_main
+
+        glue_main(task)
+    elif task == "RACE":
+        from tasks.race.finetune import main as race_main
+
+        race_main()
+    elif task in ("LAMBADA", "WIKITEXT103"):
+        from tasks.zeroshot_gpt.evaluate import main as zeroshot_main
+
+        zeroshot_main(task)
+    else:
+        raise NotImplementedError(f"task {task} is not implemented")
 
 
 if __name__ == "__main__":
     initialize_megatron(extra_args_provider=get_tasks_args)
     cfg = get_config()
-    task = cfg.task.upper() if hasattr(cfg, "task") else None
-    import argparse
-    # task arg lives on the parsed namespace; re-parse quickly
-    import sys as _sys
-    task = None
-    for i, a in enumerate(_sys.argv):
-        if a == "--task" and i + 1 < len(_sys.argv):
-            task = _sys.argv[i + 1].upper()
-    if task in ("MNLI", "QQP", "COLA", "SST2"):
-        from tasks.glue import main as glue_main
-
-        glue_main(task)
-    elif task == "RACE":
-        from tasks.race import main as race_main
-
-        race_main()
-    elif task in ("LAMBADA", "WIKITEXT103"):
-        from tasks.zeroshot_gpt import main as zeroshot_main
-
-        zeroshot_main(task)
-    else:
-        raise NotImplementedError(f"task {task} is not implemented")
+    dispatch(cfg.task.upper())
