@@ -112,6 +112,24 @@ class TrainingConfig:
     overlapping_eval: int = 32
     keep_last: bool = False
     strict_lambada: bool = False
+    # msdp (reference tasks/msdp/main.py:18-43)
+    sample_input_file: Optional[str] = None
+    sample_output_file: Optional[str] = None
+    prompt_file: Optional[str] = None
+    prompt_type: Optional[str] = None
+    num_prompt_examples: int = 10
+    guess_file: Optional[str] = None
+    answer_file: Optional[str] = None
+    out_seq_length: int = 100
+    api_prompt: bool = False
+    megatron_api_url: Optional[str] = None
+    # orqa (reference tasks/orqa/, megatron/arguments.py biencoder group)
+    qa_data_dev: Optional[str] = None
+    qa_data_test: Optional[str] = None
+    evidence_data_path: Optional[str] = None
+    biencoder_shared_query_context_model: bool = False
+    report_topk_accuracies: List[int] = field(default_factory=lambda: [1, 5, 20])
+    match: str = "string"
 
     # -- parallelism --
     tensor_model_parallel_size: int = 1

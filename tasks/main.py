@@ -25,6 +25,18 @@ def get_tasks_args(parser):
     group.add_argument("--overlapping_eval", type=int, default=32)
     group.add_argument("--keep_last", action="store_true")
     group.add_argument("--strict_lambada", action="store_true")
+    # msdp
+    group.add_argument("--sample_input_file", type=str, default=None)
+    group.add_argument("--sample_output_file", type=str, default=None)
+    group.add_argument("--prompt_file", type=str, default=None)
+    group.add_argument("--prompt_type", type=str, default=None,
+                       choices=["knowledge", "response"])
+    group.add_argument("--num_prompt_examples", type=int, default=10)
+    group.add_argument("--guess_file", type=str, default=None)
+    group.add_argument("--answer_file", type=str, default=None)
+    group.add_argument("--out_seq_length", type=int, default=100)
+    group.add_argument("--api_prompt", action="store_true")
+    group.add_argument("--megatron_api_url", type=str, default=None)
     return parser
 
 
@@ -41,6 +53,22 @@ def dispatch(task):
         from tasks.zeroshot_gpt.evaluate import main as zeroshot_main
 
         zeroshot_main(task)
+    elif task == "MSDP-PROMPT":
+        from tasks.msdp.prompt import main as msdp_prompt_main
+
+        msdp_prompt_main()
+    elif task == "MSDP-EVAL-F1":
+        from tasks.msdp.evaluate import main as msdp_eval_main
+
+        msdp_eval_main()
+    elif task in ("ICT-ZEROSHOT-NQ", "RETRIEVER-EVAL"):
+        from tasks.orqa.evaluate_orqa import main as orqa_main
+
+        orqa_main()
+    elif task == "RET-FINETUNE-NQ":
+        from tasks.orqa.supervised.finetune import main as ret_main
+
+        ret_main()
     else:
         raise NotImplementedError(f"task {task} is not implemented")
 

@@ -192,3 +192,160 @@ def test_detokenizers():
     assert wikitext_detokenizer("= = h = =") == "== h =="
     assert get_detokenizer("/data/wiki.test.tokens")("a @-@ b") == "a-b"
     assert get_detokenizer("/data/lambada.jsonl")("x  y") == "x  y"
+
+
+def test_msdp_f1_metric_and_eval(tmp_path):
+    from tasks.msdp.evaluate import evaluate_f1
+    from tasks.msdp.metrics import F1Metric, normalize_answer
+
+    assert normalize_answer("The Cat, sat!") == "cat sat"
+    p, r, f1 = F1Metric.compute_each_pair("the cat sat", "a cat sat down")
+    assert 0 < f1 < 1
+    assert F1Metric.compute_each_pair("", "gold")[2] == 0
+    assert F1Metric.compute_each_pair("x", "")[2] is None
+
+    g = tmp_path / "guess.txt"
+    a = tmp_path / "answer.txt"
+    g.write_text("the cat sat\nhello world<|endoftext|>\n")
+    a.write_text("a cat sat down\nhello world\n")
+    precision, recall, f1 = evaluate_f1(str(g), str(a))
+    assert f1 > 0.5
+
+
+def test_msdp_prompts_and_preprocessing(tmp_path):
+    from tasks.msdp.preprocessing import (
+        build_knowledge_prompts, process_wow_dataset,
+    )
+    from tasks.msdp.prompt import build_input, read_prompts
+
+    raw = tmp_path / "wow.json"
+    raw.write_text(json.dumps([{
+        "chosen_topic": "Cats",
+        "dialog": [
+            {"speaker": "0_apprentice", "text": "tell me about cats"},
+            {"speaker": "1_wizard", "text": "cats are felines",
+             "checked_sentence": {"k": "Cats are small felines."},
+             "checked_passage": {"p": "Cats"}},
+        ],
+    }]))
+    tsv = tmp_path / "wow.tsv"
+    kref = tmp_path / "k.txt"
+    rref = tmp_path / "r.txt"
+    process_wow_dataset(str(raw), str(tsv), str(kref), str(rref))
+    line = tsv.read_text().strip()
+    topic, context, knowledge, response = line.split("\t")
+    assert topic == "Cats"
+    assert knowledge == "Cats are small felines."
+
+    pfile = tmp_path / "prompts.jsonl"
+    build_knowledge_prompts(str(tsv), str(pfile), n_examples=5)
+    prompts = read_prompts(str(pfile), "knowledge", 5)
+    key = f"{topic} {context.split(' [SEP] ')[-1]}"
+    assert key in prompts
+    inp = build_input(line, "knowledge", prompts, None)
+    assert inp.endswith("=>")
+
+    rprompt = "Topic: X. User says: hi We know that: y System replies: z \n"
+    (tmp_path / "rp.txt").write_text(rprompt)
+    rp = read_prompts(str(tmp_path / "rp.txt"), "response", 10)
+    inp2 = build_input(line, "response", None, rp)
+    assert inp2.endswith("System replies:")
+
+
+def test_orqa_qa_utils():
+    from tasks.orqa.unsupervised.qa_utils import (
+        calculate_matches, exact_match_score, has_answer,
+    )
+
+    assert has_answer(["Barack Obama"], "mr barack obama was president")
+    assert not has_answer(["Obama"], "nothing here")
+    assert has_answer([r"ob\w+"], "barack obama", match_type="regex")
+    assert exact_match_score("The Answer!", "answer")
+
+    all_docs = {1: ("the sky is blue", "t1"), 2: ("grass is green", "t2")}
+    stats = calculate_matches(
+        all_docs, [["blue"], ["purple"]],
+        [([2, 1], [0.9, 0.8]), ([1, 2], [0.9, 0.8])],
+    )
+    assert stats.top_k_hits == [0, 1]  # first q hits at rank 2, second never
+    assert stats.questions_doc_hits[0] == [False, True]
+
+
+def test_orqa_retrieval_end_to_end(tmp_path, dist_single):
+    """Evidence embedding -> MIPS retrieval -> top-k answer matching with a
+    tiny random biencoder on CPU."""
+    from megatron_amd import global_state
+    from megatron_amd.config import TrainingConfig, set_config
+
+    evidence = tmp_path / "evidence.tsv"
+    evidence.write_text(
+        "id\ttext\ttitle\n"
+        "1\tthe sky is blue today\tweather\n"
+        "2\tcats are small felines\tcats\n"
+        "3\tparis is the capital of france\tfrance\n"
+    )
+    nq = tmp_path / "nq_dev.jsonl"
+    nq.write_text("\n".join(json.dumps(d) for d in [
+        {"question": "what color is the sky?", "answers": ["blue"]},
+        {"question": "what is the capital of france?", "answers": ["paris"]},
+    ]) + "\n")
+
+    cfg = TrainingConfig(
+        model_name="bert", num_layers=2, hidden_size=64,
+        num_attention_heads=4, num_attention_heads_kv=4, seq_length=32,
+        max_position_embeddings=64, micro_batch_size=2, global_batch_size=2,
+        hidden_dropout=0.0, attention_dropout=0.0,
+        use_cpu_initialization=True, position_embedding_type="absolute",
+        use_rms_norm=False, glu_activation=None, use_bias=True,
+        use_flash_attn=False, bert_binary_head=False,
+        evidence_data_path=str(evidence), qa_data_dev=str(nq),
+        num_workers=0, report_topk_accuracies=[1, 2],
+    )
+    cfg.finalize()
+    cfg.pad_vocab_size(VOCAB)
+    set_config(cfg)
+    global_state.init_timers()
+    global_state.set_tokenizer(WordTokenizer())
+
+    from tasks.orqa.evaluate_utils import ORQAEvaluator
+
+    evaluator = ORQAEvaluator()
+    stats = evaluator.evaluate(str(nq), "DEV")
+    # random model: hits are whatever they are, but shapes must line up
+    assert len(stats.questions_doc_hits) == 2
+    assert all(len(h) == 2 for h in stats.questions_doc_hits)
+
+
+def test_orqa_supervised_forward(dist_single):
+    from megatron_amd import global_state
+    from megatron_amd.config import TrainingConfig, set_config
+    from tasks.orqa.supervised import finetune as ret
+
+    cfg = TrainingConfig(
+        model_name="bert", num_layers=2, hidden_size=64,
+        num_attention_heads=4, num_attention_heads_kv=4, seq_length=32,
+        max_position_embeddings=64, micro_batch_size=4, global_batch_size=4,
+        hidden_dropout=0.0, attention_dropout=0.0,
+        use_cpu_initialization=True, position_embedding_type="absolute",
+        use_rms_norm=False, glu_activation=None, use_bias=True,
+        use_flash_attn=False, bert_binary_head=False, num_workers=0,
+    )
+    cfg.finalize()
+    cfg.pad_vocab_size(VOCAB)
+    set_config(cfg)
+    global_state.init_timers()
+    global_state.set_tokenizer(WordTokenizer())
+
+    model = ret.model_provider()
+    batch = {
+        "query": torch.randint(6, VOCAB, (4, 32)),
+        "query_pad_mask": torch.ones(4, 32, dtype=torch.long),
+        "context": torch.randint(6, VOCAB, (4, 32)),
+        "context_pad_mask": torch.ones(4, 32, dtype=torch.long),
+    }
+    scores, loss_closure = ret._forward_step(batch, model)
+    assert scores.shape == (4, 4)
+    loss, stats = loss_closure(scores)
+    assert torch.isfinite(loss)
+    loss.backward()
+    assert "in-batch acc" in stats
