@@ -137,11 +137,36 @@ def rearrange_qkv(wq, wk, wv, n_heads, n_kv_heads, head_dim, hidden):
     return permute_qkv(torch.cat(w_qkv, dim=0), hidden, n_heads, n_kv_heads)
 
 
-def falcon_to_megatron(weights: dict, size: int) -> dict:
-    """HF FalconForCausalLM -> megatron_amd (reference hf_to_megatron.py:60-114)."""
+def _permute_rotary_rows(w: torch.Tensor, n_heads_in_w: int) -> torch.Tensor:
+    """Reorder each head's rows from HF's half-split rotary pairing
+    ((j, j+d/2) rotated together by rotate_half) to the interleaved pairing
+    ((2j, 2j+1)) our RoPE kernel applies — the falcon analog of permute_qkv."""
+    total, cols = w.shape
+    d = total // n_heads_in_w
+    w = w.view(n_heads_in_w, d, cols)
+    half = d // 2
+    idx = torch.arange(d)
+    interleaved = torch.empty_like(idx)
+    interleaved[0::2] = idx[:half]
+    interleaved[1::2] = idx[half:]
+    return w[:, interleaved, :].reshape(total, cols)
+
+
+def falcon_to_megatron(weights: dict, size: int, n_layers: int = None,
+                       n_heads: int = None, n_kv: int = None) -> dict:
+    """HF FalconForCausalLM -> megatron_amd (reference hf_to_megatron.py:60-114).
+
+    The fused query_key_value layout ([q heads..., k, v] per kv group)
+    matches ours directly; only the q/k rotary row pairing is converted from
+    HF's rotate_half convention to our interleaved-pair kernel."""
     sd = {}
     prefix1 = "transformer."
-    n_layers = 32 if size == 7 else 60
+    if n_layers is None:
+        n_layers = 32 if size == 7 else 60
+    if n_heads is None:
+        n_heads = 71 if size == 7 else 128
+    if n_kv is None:
+        n_kv = 1 if size == 7 else 8
 
     sd["embedding.word_embeddings.weight"] = weights[
         f"{prefix1}word_embeddings.weight"
@@ -163,9 +188,27 @@ def falcon_to_megatron(weights: dict, size: int) -> dict:
             sd[f"{o}.input_layernorm.bias"] = weights[f"{p}.ln_attn.bias"]
             sd[f"{o}.mlp_layernorm.weight"] = weights[f"{p}.ln_mlp.weight"]
             sd[f"{o}.mlp_layernorm.bias"] = weights[f"{p}.ln_mlp.bias"]
-        sd[f"{o}.self_attention.query_key_value.weight"] = weights[
-            f"{p}.self_attention.query_key_value.weight"
-        ]
+        qkv = weights[f"{p}.self_attention.query_key_value.weight"]
+        # per kv group the fused rows are [nq q-heads, k, v] * head_dim; q
+        # and k rows get the rotary-pairing permutation, v rows stay
+        n_rows, hidden = qkv.shape
+        hd = hidden // n_heads
+        groups = n_rows // ((n_heads // n_kv + 2) * hd)
+        nq = n_heads // n_kv
+        qkv = qkv.view(groups, (nq + 2) * hd, hidden)
+        q = qkv[:, : nq * hd, :].reshape(groups * nq * hd, hidden)
+        k = qkv[:, nq * hd : (nq + 1) * hd, :].reshape(groups * hd, hidden)
+        v = qkv[:, (nq + 1) * hd :, :].reshape(groups * hd, hidden)
+        q = _permute_rotary_rows(q, groups * nq)
+        k = _permute_rotary_rows(k, groups)
+        sd[f"{o}.self_attention.query_key_value.weight"] = torch.cat(
+            [
+                q.view(groups, nq * hd, hidden),
+                k.view(groups, hd, hidden),
+                v.view(groups, hd, hidden),
+            ],
+            dim=1,
+        ).reshape(n_rows, hidden)
         sd[f"{o}.self_attention.dense.weight"] = weights[
             f"{p}.self_attention.dense.weight"
         ]

@@ -100,6 +100,78 @@ def megatron_to_hf_llama(model_sd: dict, num_layers, hidden, n_heads,
     return out
 
 
+def megatron_to_hf_falcon(model_sd: dict, num_layers, hidden, n_heads,
+                          n_heads_kv, vocab_size):
+    """Inverse of hf_to_megatron.falcon_to_megatron (reference
+    megatron_to_hf.py:333-434): undo the interleaved->rotate_half rotary row
+    permutation on q/k and restore the HF key names."""
+    from weights_conversion.hf_to_megatron import _permute_rotary_rows
+
+    def unpermute(w, heads):
+        total, cols = w.shape
+        d = total // heads
+        wv = w.view(heads, d, cols)
+        half = d // 2
+        inv = torch.empty(d, dtype=torch.long)
+        inv[:half] = torch.arange(0, d, 2)
+        inv[half:] = torch.arange(1, d, 2)
+        return wv[:, inv, :].reshape(total, cols)
+
+    out = {}
+    out["transformer.word_embeddings.weight"] = model_sd[
+        "embedding.word_embeddings.weight"
+    ][:vocab_size]
+    out["transformer.ln_f.weight"] = model_sd["encoder.final_layernorm.weight"]
+    out["transformer.ln_f.bias"] = model_sd["encoder.final_layernorm.bias"]
+    out["lm_head.weight"] = out["transformer.word_embeddings.weight"]
+    hd = hidden // n_heads
+    nq = n_heads // n_heads_kv
+    for i in range(num_layers):
+        p = f"encoder.layers.{i}"
+        o = f"transformer.h.{i}"
+        if f"{p}.mlp_layernorm.weight" in model_sd:  # 40B parallel-LN
+            out[f"{o}.ln_attn.weight"] = model_sd[f"{p}.input_layernorm.weight"]
+            out[f"{o}.ln_attn.bias"] = model_sd[f"{p}.input_layernorm.bias"]
+            out[f"{o}.ln_mlp.weight"] = model_sd[f"{p}.mlp_layernorm.weight"]
+            out[f"{o}.ln_mlp.bias"] = model_sd[f"{p}.mlp_layernorm.bias"]
+        else:
+            out[f"{o}.input_layernorm.weight"] = model_sd[
+                f"{p}.input_layernorm.weight"
+            ]
+            out[f"{o}.input_layernorm.bias"] = model_sd[
+                f"{p}.input_layernorm.bias"
+            ]
+        qkv = model_sd[f"{p}.self_attention.query_key_value.weight"]
+        n_rows, cols = qkv.shape
+        groups = n_heads_kv
+        qkv = qkv.view(groups, (nq + 2) * hd, cols)
+        q = unpermute(qkv[:, : nq * hd, :].reshape(groups * nq * hd, cols),
+                      groups * nq)
+        k = unpermute(
+            qkv[:, nq * hd : (nq + 1) * hd, :].reshape(groups * hd, cols),
+            groups,
+        )
+        v = qkv[:, (nq + 1) * hd :, :].reshape(groups * hd, cols)
+        out[f"{o}.self_attention.query_key_value.weight"] = torch.cat(
+            [
+                q.view(groups, nq * hd, cols),
+                k.view(groups, hd, cols),
+                v.view(groups, hd, cols),
+            ],
+            dim=1,
+        ).reshape(n_rows, cols)
+        out[f"{o}.self_attention.dense.weight"] = model_sd[
+            f"{p}.self_attention.dense.weight"
+        ]
+        out[f"{o}.mlp.dense_h_to_4h.weight"] = model_sd[
+            f"{p}.mlp.dense_h_to_4h.weight"
+        ]
+        out[f"{o}.mlp.dense_4h_to_h.weight"] = model_sd[
+            f"{p}.mlp.dense_4h_to_h.weight"
+        ]
+    return out
+
+
 def write_hf_checkpoint(hf_sd: dict, out_dir: str, config: dict,
                         use_safetensors=True):
     out = Path(out_dir)
