@@ -1,0 +1,113 @@
+"""Auxiliary subsystems: timers, metrics, the wandb TB shim, and the
+distributed SIGTERM consensus handler (reference tests/test_wandb.py +
+SURVEY §5 coverage)."""
+
+import os
+import signal
+import sys
+import time
+from unittest import mock
+
+import pytest
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def test_timers_basic(dist_single):
+    from megatron_amd.timers import DummyTimer, Timers
+
+    timers = Timers(log_level=1, log_option="minmax")
+    t = timers("phase-a", log_level=1)
+    t.start()
+    time.sleep(0.01)
+    t.stop()
+    assert timers("phase-a").elapsed(reset=False) >= 0.01
+
+    # above the configured log level -> dummy (zero-cost, no state)
+    d = timers("too-detailed", log_level=2)
+    assert isinstance(d, DummyTimer)
+    d.start()
+    d.stop()
+
+    # same name must keep its level
+    again = timers("phase-a", log_level=1)
+    assert again is not None
+
+    # log() runs the all-rank gather path
+    timers.log(["phase-a"])
+
+    class _Writer:
+        def __init__(self):
+            self.scalars = {}
+
+        def add_scalar(self, name, value, step):
+            self.scalars[name] = (value, step)
+
+    timers("phase-b", log_level=0).start()
+    timers("phase-b").stop()
+    w = _Writer()
+    timers.write(["phase-b"], w, iteration=3)
+    assert any("phase-b" in k for k in w.scalars)
+
+
+def test_timer_double_start_raises(dist_single):
+    from megatron_amd.timers import Timers
+
+    timers = Timers(log_level=2)
+    timers("x").start()
+    with pytest.raises(AssertionError):
+        timers("x").start()
+    timers("x").stop()
+
+
+def test_metrics(dist_single):
+    from megatron_amd import parallel as mpu
+    from megatron_amd.metrics import METRICS, MetricInput
+
+    if not mpu.model_parallel_is_initialized():
+        mpu.initialize_model_parallel(1, 1)
+    s, b, v = 4, 2, 8
+    logits = torch.zeros(s, b, v)
+    labels = torch.zeros(b, s, dtype=torch.long)
+    logits[..., 3] = 5.0  # argmax = 3 everywhere
+    labels[:, :2] = 3  # half the positions correct
+    inp = MetricInput(
+        batch={"labels": labels, "loss_mask": torch.ones(b, s)},
+        logits=logits, loss=torch.tensor(0.5),
+    )
+    assert abs(METRICS["perplexity"](inp)["perplexity"]
+               - torch.exp(torch.tensor(0.5)).item()) < 1e-6
+    assert METRICS["accuracy"](inp)["accuracy"] == 0.5
+    assert METRICS["count_loss_mask"](inp)["count_loss_mask"] == b * s
+
+
+def test_wandb_shim_batches_by_step():
+    from megatron_amd.wandb_logger import WandBConfig, WandbTBShim
+
+    fake_wandb = mock.MagicMock()
+    fake_run = mock.MagicMock()
+    fake_wandb.init.return_value = fake_run
+    with mock.patch.dict(sys.modules, {"wandb": fake_wandb}):
+        shim = WandbTBShim(WandBConfig(project="p", name="n"))
+        shim.add_scalar("loss", 1.0, 1)
+        shim.add_scalar("lr", 0.1, 1)
+        fake_wandb.log.assert_not_called()  # same step: buffered
+        shim.add_scalar("loss", 0.9, 2)  # new step flushes step 1
+        fake_wandb.log.assert_called_once_with(
+            {"loss": 1.0, "lr": 0.1}, step=1
+        )
+        shim.flush_all()
+        assert fake_wandb.log.call_count == 2
+
+
+def test_signal_handler_consensus(dist_single):
+    from megatron_amd.dist_signal_handler import DistributedSignalHandler
+
+    with DistributedSignalHandler(sig=signal.SIGUSR1) as handler:
+        assert handler.signals_received() == [False]
+        os.kill(os.getpid(), signal.SIGUSR1)
+        # the handler records the signal; consensus gathers across ranks
+        assert handler.signals_received() == [True]
+    # original handler restored on exit
+    assert signal.getsignal(signal.SIGUSR1) not in (None,)
