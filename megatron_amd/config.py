@@ -418,6 +418,12 @@ def parse_config(extra_args_provider=None, args_list=None, defaults=None) -> Tra
             if parser.get_default(k) == kwargs.get(k):  # not overridden on CLI
                 kwargs[k] = v
     cfg = TrainingConfig(**kwargs)
+    # extra_args_provider flags that are not dataclass fields (e.g. --port
+    # of the generation server) still ride on the config object
+    field_names = {f.name for f in dataclasses.fields(TrainingConfig)}
+    for k, v in vars(ns).items():
+        if k not in field_names:
+            setattr(cfg, k, v)
     # environment (torchrun)
     cfg.rank = int(os.environ.get("RANK", "0"))
     cfg.world_size = int(os.environ.get("WORLD_SIZE", "1"))

@@ -57,6 +57,7 @@ if __name__ == "__main__":
         and mpu.get_tensor_model_parallel_rank() == 0
     ):
         server = MegatronServer(model)
-        server.run(port=int(os.environ.get("PORT", 5000)))
+        server.run(port=int(os.environ.get("PORT", getattr(cfg, "port",
+                                                           5000))))
     else:
         run_worker_loop(model)
