@@ -31,15 +31,23 @@ class InferenceParams:
                 self.key_value_memory_dict[layer_number]
             )
             assert len(batch_idx) == inference_key_memory.shape[1]
-            new_inference_key_memory = inference_key_memory[:, batch_idx]
-            new_inference_value_memory = inference_value_memory[:, batch_idx]
-            self.key_value_memory_dict[layer_number] = (
-                new_inference_key_memory, new_inference_value_memory,
+            # swap IN-PLACE so cache pointers stay valid (a captured decode
+            # graph holds them)
+            inference_key_memory.copy_(inference_key_memory[:, batch_idx])
+            inference_value_memory.copy_(
+                inference_value_memory[:, batch_idx]
             )
 
 
 class ForwardStep:
-    """Forward step wrapper handling pipelining (reference :44-204)."""
+    """Forward step wrapper handling pipelining (reference :44-204).
+
+    MI355X addition: single-token decode steps are captured as a HIP graph
+    (torch.cuda.CUDAGraph is hipGraph on ROCm) and replayed — the decode
+    inner loop is launch-bound (~1k tiny kernels per token at 7B), so one
+    graph launch replaces the whole per-token launch storm. Enabled when
+    `cfg.use_hip_graph_decode`, TP=PP=1, on GPU; the prompt step and any
+    variable-shape step stay eager."""
 
     def __init__(self, model, max_batch_size, max_sequence_len):
         assert not isinstance(model, list)
@@ -53,6 +61,14 @@ class ForwardStep:
         self.pipelining_batch_x_seqlen = (
             cfg.inference_batch_times_seqlen_threshold
         )
+        self._graph = None
+        self._graph_batch = None
+        self._use_graph_decode = (
+            getattr(cfg, "use_hip_graph_decode", True)
+            and torch.cuda.is_available()
+            and not self.pipeline_size_larger_than_one
+            and mpu.get_tensor_model_parallel_world_size() == 1
+        )
 
     def __call__(self, tokens, position_ids, attention_mask):
         if self.pipeline_size_larger_than_one:
@@ -64,9 +80,57 @@ class ForwardStep:
                 return self._with_pipelining_forward_step(
                     tokens, position_ids, attention_mask, micro_batch_size
                 )
+        if (
+            self._use_graph_decode
+            and tokens.size(1) == 1
+            and self.inference_params.key_value_memory_dict
+        ):
+            return self._graph_decode_step(tokens, position_ids)
         return self._no_pipelining_forward_step(
             tokens, position_ids, attention_mask
         )
+
+    def _init_graph_state(self, batch_size):
+        ip = self.inference_params
+        device = torch.cuda.current_device()
+        ip.use_graph = True
+        ip.graph_pos = torch.zeros(1, dtype=torch.long, device=device)
+        ip.graph_arange = torch.arange(ip.max_sequence_len, device=device)
+        self._s_tokens = torch.zeros(batch_size, 1, dtype=torch.long,
+                                     device=device)
+        self._s_pos = torch.zeros(batch_size, 1, dtype=torch.long,
+                                  device=device)
+        self._graph_batch = batch_size
+
+    def _graph_decode_step(self, tokens, position_ids):
+        ip = self.inference_params
+        batch_size = tokens.size(0)
+        if self._graph_batch is None:
+            self._init_graph_state(batch_size)
+        assert batch_size == self._graph_batch
+
+        # stage this step's inputs into the static buffers the graph reads
+        self._s_tokens.copy_(tokens)
+        self._s_pos.copy_(position_ids)
+        ip.graph_pos.fill_(ip.sequence_len_offset)
+
+        if self._graph is None:
+            # warm-up eager run through the static path (allocator, blas
+            # workspaces), then capture the second run
+            ip.use_graph = True
+            self.model(self._s_tokens, self._s_pos, None,
+                       inference_params=ip)
+            torch.cuda.synchronize()
+            self._graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self._graph):
+                self._s_logits = self.model(self._s_tokens, self._s_pos,
+                                            None, inference_params=ip)
+            # the warm-up run and the capture pass each wrote the KV slot;
+            # the capture's write is identical, so state is consistent
+        else:
+            self._graph.replay()
+        ip.sequence_len_offset += 1
+        return self._s_logits
 
     def _forward(self, tokens, position_ids, attention_mask):
         return self.model(

@@ -374,6 +374,18 @@ class ParallelAttention(MegatronModule):
             else:
                 query, key = apply_rotary_emb(query, key, cos, sin, None)
 
+        # hipGraph-capturable decode: every op below is shape-static (the
+        # current length lives in device tensors), so ForwardStep can capture
+        # one decode step as a HIP graph and replay it per token — the decode
+        # loop is launch-bound, exactly what the graph removes.
+        if (
+            inference_params is not None
+            and getattr(inference_params, "use_graph", False)
+            and sq == 1
+            and self.layer_number in inference_params.key_value_memory_dict
+        ):
+            return self._static_decode(query, key, value, inference_params)
+
         # KV cache (reference transformer.py:412-419, 492-505)
         if inference_params is not None:
             if self.layer_number not in inference_params.key_value_memory_dict:
@@ -398,6 +410,41 @@ class ParallelAttention(MegatronModule):
 
         return self._attend(query, key, value, attention_mask,
                             inference_params)
+
+    def _static_decode(self, query, key, value, inference_params):
+        """One decode step with every shape fixed: KV written by tensor
+        index, attention over the FULL cache with a length mask read from a
+        device tensor — capturable as a HIP graph (see
+        inference/forward_step.py)."""
+        ip = inference_params
+        k_cache, v_cache = ip.key_value_memory_dict[self.layer_number]
+        # [1,b,nkv,h] written at the device-held position
+        k_cache.index_copy_(0, ip.graph_pos, key)
+        v_cache.index_copy_(0, ip.graph_pos, value)
+
+        b = query.shape[1]
+        n, hn = query.shape[2], query.shape[3]
+        nkv = k_cache.shape[2]
+        L = k_cache.shape[0]
+        q = query.permute(1, 2, 0, 3).reshape(b * n, 1, hn)  # [b*n,1,h]
+        k = k_cache.permute(1, 2, 0, 3)  # [b,nkv,L,h]
+        v = v_cache.permute(1, 2, 0, 3)
+        if n != nkv:
+            rep = n // nkv
+            k = k.repeat_interleave(rep, dim=1)
+            v = v.repeat_interleave(rep, dim=1)
+        k = k.reshape(b * n, L, hn)
+        v = v.reshape(b * n, L, hn)
+
+        scale = 1.0 / math.sqrt(hn)
+        scores = torch.bmm(q.float(), k.float().transpose(1, 2)) * scale
+        invalid = ip.graph_arange > ip.graph_pos  # [L] device-side length
+        scores = scores.masked_fill(invalid.view(1, 1, L), float("-inf"))
+        probs = torch.softmax(scores, dim=-1).to(v.dtype)
+        ctx = torch.bmm(probs, v)  # [b*n,1,h]
+        ctx = ctx.view(b, n, hn).view(b, n * hn).unsqueeze(0)  # [1,b,n*h]
+        output, bias = self.dense(ctx)
+        return output, bias
 
     def _attend(self, query, key, value, attention_mask, inference_params):
         sq = query.shape[0]
