@@ -125,8 +125,10 @@ class ForwardStep:
             with torch.cuda.graph(self._graph):
                 self._s_logits = self.model(self._s_tokens, self._s_pos,
                                             None, inference_params=ip)
-            # the warm-up run and the capture pass each wrote the KV slot;
-            # the capture's write is identical, so state is consistent
+            # capture only RECORDS the work — replay to actually execute
+            # this step (the warm-up already wrote the same KV slot, and the
+            # replay rewrites it identically)
+            self._graph.replay()
         else:
             self._graph.replay()
         ip.sequence_len_offset += 1
