@@ -130,8 +130,10 @@ class TrainingConfig:
     biencoder_shared_query_context_model: bool = False
     report_topk_accuracies: List[int] = field(default_factory=lambda: [1, 5, 20])
     match: str = "string"
-    # hipGraph-captured single-token decode (inference/forward_step.py)
-    use_hip_graph_decode: bool = True
+    # hipGraph-captured single-token decode (inference/forward_step.py).
+    # Opt-in: one replay replaces the per-token launch storm; enable for
+    # serving (tools/run_text_generation_server.py turns it on)
+    use_hip_graph_decode: bool = False
 
     # -- parallelism --
     tensor_model_parallel_size: int = 1

@@ -64,7 +64,7 @@ class ForwardStep:
         self._graph = None
         self._graph_batch = None
         self._use_graph_decode = (
-            getattr(cfg, "use_hip_graph_decode", True)
+            getattr(cfg, "use_hip_graph_decode", False)
             and torch.cuda.is_available()
             and not self.pipeline_size_larger_than_one
             and mpu.get_tensor_model_parallel_world_size() == 1

@@ -221,7 +221,10 @@ def test_graph_decode_matches_eager(gpu_cfg):
     from megatron_amd.tokenizer.tokenizers import FakeTokenizer
     from megatron_amd.utils import get_ltor_masks_and_position_ids
 
+    from megatron_amd.config import set_config
+
     cfg = gpu_cfg
+    set_config(cfg)  # earlier tests may have replaced the global config
     global_state.set_tokenizer(FakeTokenizer(1000))
     torch.manual_seed(7)
     m = LlamaModel(cfg, parallel_output=False).cuda().bfloat16()

@@ -41,7 +41,8 @@ def add_text_generate_args(parser):
 if __name__ == "__main__":
     initialize_megatron(
         extra_args_provider=add_text_generate_args,
-        args_defaults={"no_load_rng": True, "no_load_optim": True},
+        args_defaults={"no_load_rng": True, "no_load_optim": True,
+                       "use_hip_graph_decode": True},
     )
     cfg = get_config()
     model = get_model(model_provider, ModelType.encoder_or_decoder,
