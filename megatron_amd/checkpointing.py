@@ -13,7 +13,6 @@ from __future__ import annotations
 
 import os
 import random
-import sys
 
 import numpy as np
 import torch

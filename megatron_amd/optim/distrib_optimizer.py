@@ -19,7 +19,6 @@ MI355X shape of the same idea:
 
 from __future__ import annotations
 
-import math
 from typing import Dict, List
 
 import torch

@@ -7,7 +7,6 @@ import functools
 import torch
 
 from megatron_amd import global_state
-from megatron_amd import parallel as mpu
 from megatron_amd.config import get_config
 from megatron_amd.models import BertModel, ModelType
 from megatron_amd.parallel import broadcast_data

@@ -3,8 +3,6 @@
 
 from __future__ import annotations
 
-import functools
-
 import torch
 import torch.nn.functional as F
 

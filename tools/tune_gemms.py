@@ -10,7 +10,6 @@ then loads in read-only mode.
 
 import os
 import sys
-import time
 
 sys.path.append(os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 

@@ -16,7 +16,6 @@ max abs logit error <= 0.01 avg at fp32, <= 0.1 bf16.
 
 from __future__ import annotations
 
-import sys
 
 import torch
 
