@@ -53,7 +53,8 @@ def parse_args():
                    help="override layer count (shape validation only; "
                    "results with this flag are not official numbers)")
     p.add_argument("--recompute", action="store_true")
-    p.add_argument("--dtype", type=str, default="bf16")
+    p.add_argument("--dtype", type=str, default="bf16",
+                   choices=["bf16", "fp16", "fp8"])
     return p.parse_args()
 
 
@@ -122,6 +123,11 @@ def main():
         dtype_flags["bf16"] = True
     elif args.dtype == "fp16" and have_gpu:
         dtype_flags["fp16"] = True
+    elif args.dtype == "fp8" and have_gpu:
+        # fp8 fwd/dgrad GEMMs over bf16 params (megatron_amd/fp8.py) — a
+        # separate recipe, never the headline bf16 number
+        dtype_flags["bf16"] = True
+        dtype_flags["fp8"] = True
 
     from megatron_amd.config import TrainingConfig, set_config
     from megatron_amd.initialize import initialize_megatron
@@ -277,7 +283,7 @@ def main():
             "scaling": "weak",
             "vs_baseline": round(tokens_per_sec / baseline, 4)
             if have_gpu and args.model == "llama2-7b" else None,
-            "dtype": args.dtype if have_gpu else "fp32",
+            "dtype": (args.dtype if have_gpu else "fp32"),
             "data": "synthetic",
             "config": {
                 "model": (args.model if have_gpu else f"{args.model}(cpu-tiny)")

@@ -54,6 +54,11 @@ class TrainingConfig:
     fp32_residual_connection: bool = False
     fp16: bool = False
     bf16: bool = False
+    # fp8 (e4m3) forward/dgrad GEMMs via hipBLASLt _scaled_mm (megatron_amd/
+    # fp8.py) — gfx950 fp8 MFMA is 2x bf16. wgrad stays bf16/fp32-accum;
+    # params/optimizer/checkpoints unchanged. NOT used for headline bf16
+    # numbers; enable with --fp8 for the fp8 recipe.
+    fp8: bool = False
     loss_scale: Optional[float] = None
     initial_loss_scale: float = 2 ** 32
     min_loss_scale: float = 1.0

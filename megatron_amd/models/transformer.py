@@ -48,6 +48,7 @@ class ParallelMLP(MegatronModule):
             cfg.hidden_size, cfg.ffn_hidden_size * ffn_mult,
             bias=cfg.use_bias, gather_output=False, init_method=init_method,
             skip_bias_add=True, params_dtype=cfg.params_dtype,
+            fp8=getattr(cfg, "fp8", False),
             use_cpu_initialization=cfg.use_cpu_initialization,
             perform_initialization=cfg.perform_initialization,
             gradient_accumulation_fusion=cfg.gradient_accumulation_fusion,
@@ -65,7 +66,7 @@ class ParallelMLP(MegatronModule):
             cfg.ffn_hidden_size, cfg.hidden_size,
             bias=cfg.use_bias, input_is_parallel=True,
             init_method=output_layer_init_method, skip_bias_add=True,
-            params_dtype=cfg.params_dtype,
+            params_dtype=cfg.params_dtype, fp8=getattr(cfg, "fp8", False),
             use_cpu_initialization=cfg.use_cpu_initialization,
             perform_initialization=cfg.perform_initialization,
             gradient_accumulation_fusion=cfg.gradient_accumulation_fusion,
@@ -227,7 +228,7 @@ class ParallelAttention(MegatronModule):
 
         linear_kwargs = dict(
             bias=cfg.use_bias, gather_output=False, init_method=init_method,
-            params_dtype=cfg.params_dtype,
+            params_dtype=cfg.params_dtype, fp8=getattr(cfg, "fp8", False),
             use_cpu_initialization=cfg.use_cpu_initialization,
             perform_initialization=cfg.perform_initialization,
             gradient_accumulation_fusion=cfg.gradient_accumulation_fusion,
@@ -270,7 +271,7 @@ class ParallelAttention(MegatronModule):
             projection_size, cfg.hidden_size,
             bias=cfg.use_bias, input_is_parallel=True,
             init_method=output_layer_init_method, skip_bias_add=True,
-            params_dtype=cfg.params_dtype,
+            params_dtype=cfg.params_dtype, fp8=getattr(cfg, "fp8", False),
             use_cpu_initialization=cfg.use_cpu_initialization,
             perform_initialization=cfg.perform_initialization,
             gradient_accumulation_fusion=cfg.gradient_accumulation_fusion,

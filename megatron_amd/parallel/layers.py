@@ -68,12 +68,13 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, input, weight, bias, gradient_accumulation_fusion,
-                async_grad_allreduce, sequence_parallel):
+                async_grad_allreduce, sequence_parallel, fp8=False):
         ctx.save_for_backward(input, weight)
         ctx.use_bias = bias is not None
         ctx.gradient_accumulation_fusion = gradient_accumulation_fusion
         ctx.async_grad_allreduce = async_grad_allreduce
         ctx.sequence_parallel = sequence_parallel
+        ctx.fp8 = fp8 and input.is_cuda
 
         if sequence_parallel:
             world_size = ps.get_tensor_model_parallel_world_size()
@@ -90,7 +91,15 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
         else:
             total_input = input
 
-        output = torch.matmul(total_input, weight.t())
+        if ctx.fp8:
+            from ..fp8 import fp8_matmul
+
+            ti2d = total_input.contiguous().view(-1, total_input.shape[-1])
+            output = fp8_matmul(ti2d, weight.t()).view(
+                *total_input.shape[:-1], weight.shape[0]
+            )
+        else:
+            output = torch.matmul(total_input, weight.t())
         if bias is not None:
             output = output + bias
         return output
@@ -115,7 +124,15 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
         else:
             total_input = input
 
-        grad_input = grad_output.matmul(weight)
+        if ctx.fp8:
+            from ..fp8 import fp8_matmul
+
+            go2d = grad_output.contiguous().view(-1, grad_output.shape[-1])
+            grad_input = fp8_matmul(go2d, weight).view(
+                *grad_output.shape[:-1], weight.shape[1]
+            )
+        else:
+            grad_input = grad_output.matmul(weight)
 
         if ctx.sequence_parallel:
             gather_handle.wait()
@@ -151,19 +168,20 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
 
         if ctx.sequence_parallel and not ctx.async_grad_allreduce:
             rs_handle.wait()
-            return sub_grad_input, grad_weight, grad_bias, None, None, None
+            return (sub_grad_input, grad_weight, grad_bias, None, None, None,
+                    None)
         if ctx.async_grad_allreduce:
             allreduce_handle.wait()
-        return grad_input, grad_weight, grad_bias, None, None, None
+        return grad_input, grad_weight, grad_bias, None, None, None, None
 
 
 def linear_with_grad_accumulation_and_async_allreduce(
     input, weight, bias, gradient_accumulation_fusion,
-    async_grad_allreduce, sequence_parallel_enabled,
+    async_grad_allreduce, sequence_parallel_enabled, fp8=False,
 ):
     return LinearWithGradAccumulationAndAsyncCommunication.apply(
         input, weight, bias, gradient_accumulation_fusion,
-        async_grad_allreduce, sequence_parallel_enabled,
+        async_grad_allreduce, sequence_parallel_enabled, fp8,
     )
 
 
@@ -279,10 +297,12 @@ class ColumnParallelLinear(torch.nn.Module):
                  async_tensor_model_parallel_allreduce=True,
                  params_dtype=torch.float32, use_cpu_initialization=False,
                  perform_initialization=True, gradient_accumulation_fusion=False,
-                 sequence_parallel_enabled: bool = False, world_size: Optional[int] = None):
+                 sequence_parallel_enabled: bool = False, world_size: Optional[int] = None,
+                 fp8: bool = False):
         super().__init__()
         self.input_size = input_size
         self.output_size = output_size
+        self.fp8 = fp8
         self.gather_output = gather_output
         world_size = world_size if world_size is not None else (
             ps.get_tensor_model_parallel_world_size()
@@ -353,7 +373,7 @@ class ColumnParallelLinear(torch.nn.Module):
             input_parallel, self.weight, bias,
             self.gradient_accumulation_fusion,
             self.async_tensor_model_parallel_allreduce,
-            self.sequence_parallel_enabled,
+            self.sequence_parallel_enabled, self.fp8,
         )
         if self.gather_output:
             assert not self.sequence_parallel_enabled
@@ -373,10 +393,12 @@ class RowParallelLinear(torch.nn.Module):
                  stride=1, keep_master_weight_for_test=False, skip_bias_add=False,
                  params_dtype=torch.float32, use_cpu_initialization=False,
                  perform_initialization=True, gradient_accumulation_fusion=False,
-                 sequence_parallel_enabled: bool = False, world_size: Optional[int] = None):
+                 sequence_parallel_enabled: bool = False, world_size: Optional[int] = None,
+                 fp8: bool = False):
         super().__init__()
         self.input_size = input_size
         self.output_size = output_size
+        self.fp8 = fp8
         self.input_is_parallel = input_is_parallel
         world_size = world_size if world_size is not None else (
             ps.get_tensor_model_parallel_world_size()
@@ -438,7 +460,7 @@ class RowParallelLinear(torch.nn.Module):
             input_parallel, self.weight, None,
             self.gradient_accumulation_fusion,
             False,  # row-parallel dgrad needs no all-reduce
-            False,
+            False, self.fp8,
         )
         if self.sequence_parallel_enabled:
             output_ = reduce_scatter_to_sequence_parallel_region(output_parallel)
