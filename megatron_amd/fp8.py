@@ -50,3 +50,47 @@ def fp8_matmul(a2d: torch.Tensor, b_t: torch.Tensor, out_dtype=None):
         a8, b8.t(), scale_a=sa, scale_b=sb,
         out_dtype=out_dtype or a2d.dtype,
     )
+
+
+# --- per-step weight quantization cache --------------------------------------
+# Weights only change at the optimizer step, so their fp8 copies (and the
+# transposed copies the dgrad GEMM needs as column-major B) are reused across
+# every layer invocation of a step. The optimizer bumps the epoch.
+
+_step_epoch = 0
+_weight_cache: dict = {}
+
+
+def bump_weight_epoch():
+    global _step_epoch
+    _step_epoch += 1
+    _weight_cache.clear()
+
+
+def _cached_weight_fp8(weight: torch.Tensor):
+    key = weight.data_ptr()
+    ent = _weight_cache.get(key)
+    if ent is not None and ent[0] == _step_epoch:
+        return ent[1]
+    q, s = quantize_e4m3(weight)                     # [N,K] row-major
+    qt, st = quantize_e4m3(weight.t().contiguous())  # [K,N] row-major
+    out = (q, s, qt, st)
+    _weight_cache[key] = (_step_epoch, out)
+    return out
+
+
+def fp8_linear_fwd(x2d: torch.Tensor, weight: torch.Tensor):
+    """x2d [M,K] @ weight[N,K].t() -> [M,N] with cached fp8 weight."""
+    a8, sa = quantize_e4m3(x2d)
+    q, s, _, _ = _cached_weight_fp8(weight)
+    return torch._scaled_mm(a8, q.t(), scale_a=sa, scale_b=s,
+                            out_dtype=x2d.dtype)
+
+
+def fp8_linear_dgrad(dy2d: torch.Tensor, weight: torch.Tensor):
+    """dy2d [M,N] @ weight[N,K] -> [M,K]; B column-major = cached
+    transposed fp8 copy."""
+    a8, sa = quantize_e4m3(dy2d)
+    _, _, qt, st = _cached_weight_fp8(weight)
+    return torch._scaled_mm(a8, qt.t(), scale_a=sa, scale_b=st,
+                            out_dtype=dy2d.dtype)

@@ -360,6 +360,11 @@ class MixedPrecisionOptimizer(MegatronOptimizer):
             self._copy_main_params_to_model_params()
         _t("optimizer-copy-main-to-model-params", False)
 
+        if getattr(self.cfg, "fp8", False):
+            from ..fp8 import bump_weight_epoch
+
+            bump_weight_epoch()
+
         return True, grad_norm, num_zeros_in_grad
 
 

@@ -92,10 +92,10 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
             total_input = input
 
         if ctx.fp8:
-            from ..fp8 import fp8_matmul
+            from ..fp8 import fp8_linear_fwd
 
             ti2d = total_input.contiguous().view(-1, total_input.shape[-1])
-            output = fp8_matmul(ti2d, weight.t()).view(
+            output = fp8_linear_fwd(ti2d, weight).view(
                 *total_input.shape[:-1], weight.shape[0]
             )
         else:
@@ -125,10 +125,10 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
             total_input = input
 
         if ctx.fp8:
-            from ..fp8 import fp8_matmul
+            from ..fp8 import fp8_linear_dgrad
 
             go2d = grad_output.contiguous().view(-1, grad_output.shape[-1])
-            grad_input = fp8_matmul(go2d, weight).view(
+            grad_input = fp8_linear_dgrad(go2d, weight).view(
                 *grad_output.shape[:-1], weight.shape[1]
             )
         else:
