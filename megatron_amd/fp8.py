@@ -67,22 +67,24 @@ def bump_weight_epoch():
     _weight_cache.clear()
 
 
-def _cached_weight_fp8(weight: torch.Tensor):
+def _cached_weight_t_fp8(weight: torch.Tensor):
+    """fp8 copy of weight.t() (the column-major B the dgrad GEMM needs) —
+    the only cached copy: caching the forward copy too would cost another
+    ~7 GB at 7B, and the forward cast has no transpose so it is cheap to
+    redo per call."""
     key = weight.data_ptr()
     ent = _weight_cache.get(key)
     if ent is not None and ent[0] == _step_epoch:
         return ent[1]
-    q, s = quantize_e4m3(weight)                     # [N,K] row-major
     qt, st = quantize_e4m3(weight.t().contiguous())  # [K,N] row-major
-    out = (q, s, qt, st)
-    _weight_cache[key] = (_step_epoch, out)
-    return out
+    _weight_cache[key] = (_step_epoch, (qt, st))
+    return qt, st
 
 
 def fp8_linear_fwd(x2d: torch.Tensor, weight: torch.Tensor):
-    """x2d [M,K] @ weight[N,K].t() -> [M,N] with cached fp8 weight."""
+    """x2d [M,K] @ weight[N,K].t() -> [M,N]."""
     a8, sa = quantize_e4m3(x2d)
-    q, s, _, _ = _cached_weight_fp8(weight)
+    q, s = quantize_e4m3(weight)  # no transpose: cheap per-call cast
     return torch._scaled_mm(a8, q.t(), scale_a=sa, scale_b=s,
                             out_dtype=x2d.dtype)
 
@@ -91,6 +93,6 @@ def fp8_linear_dgrad(dy2d: torch.Tensor, weight: torch.Tensor):
     """dy2d [M,N] @ weight[N,K] -> [M,K]; B column-major = cached
     transposed fp8 copy."""
     a8, sa = quantize_e4m3(dy2d)
-    _, _, qt, st = _cached_weight_fp8(weight)
+    qt, st = _cached_weight_t_fp8(weight)
     return torch._scaled_mm(a8, qt.t(), scale_a=sa, scale_b=st,
                             out_dtype=dy2d.dtype)
