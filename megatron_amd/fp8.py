@@ -96,3 +96,58 @@ def fp8_linear_dgrad(dy2d: torch.Tensor, weight: torch.Tensor):
     qt, st = _cached_weight_t_fp8(weight)
     return torch._scaled_mm(a8, qt.t(), scale_a=sa, scale_b=st,
                             out_dtype=dy2d.dtype)
+
+
+# --- delayed scaling (TransformerEngine-style) -------------------------------
+# Each Linear call site keeps per-tensor-role state: the dequant scale used
+# to cast THIS step comes from the max of the last H amaxes, so the cast is
+# a single fused kernel pass (ops ext fp8_quantize: cast + amax in one read)
+# instead of the JIT path's three eager passes.
+
+
+class Fp8TensorMeta:
+    def __init__(self, device, history: int = 16):
+        self.scale = torch.ones(1, device=device)  # dequant multiplier
+        self.amax_history = torch.zeros(history, device=device)
+        self._i = 0
+
+    def update(self, amax):
+        self.amax_history[self._i % self.amax_history.numel()] = amax[0]
+        self._i += 1
+        self.scale.copy_(
+            (self.amax_history.max() / E4M3_MAX).clamp(min=1e-12)
+        )
+
+
+def _ext():
+    from .ops import ext as ops_ext
+
+    mod = ops_ext.load(required=False)
+    return mod if (mod is not None and hasattr(mod, "fp8_quantize")) else None
+
+
+def quantize_delayed(t: torch.Tensor, meta: Fp8TensorMeta):
+    """One-pass cast with last step's scale + amax collection; falls back to
+    JIT scaling when the ops extension is unavailable."""
+    mod = _ext()
+    if mod is None:
+        return quantize_e4m3(t)
+    t = t.contiguous()
+    scale_used = meta.scale.clone()
+    q, amax = mod.fp8_quantize(t.view(-1), meta.scale)
+    meta.update(amax)
+    return q.view(t.shape), scale_used
+
+
+def fp8_linear_fwd_delayed(x2d, weight, meta_x):
+    a8, sa = quantize_delayed(x2d, meta_x)
+    q, s = quantize_e4m3(weight)
+    return torch._scaled_mm(a8, q.t(), scale_a=sa, scale_b=s,
+                            out_dtype=x2d.dtype)
+
+
+def fp8_linear_dgrad_delayed(dy2d, weight, meta_dy):
+    a8, sa = quantize_delayed(dy2d, meta_dy)
+    qt, st = _cached_weight_t_fp8(weight)
+    return torch._scaled_mm(a8, qt.t(), scale_a=sa, scale_b=st,
+                            out_dtype=dy2d.dtype)

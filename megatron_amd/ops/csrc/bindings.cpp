@@ -55,6 +55,8 @@ void fused_adam_with_model_copy(std::vector<torch::Tensor> params,
                                 int64_t adam_w_mode, double grad_scale);
 
 // wgrad.hip
+std::vector<torch::Tensor> fp8_quantize(torch::Tensor x,
+                                        torch::Tensor scale);
 void wgrad_gemm_accum_fp32(torch::Tensor input, torch::Tensor grad_output,
                            torch::Tensor main_grad);
 
@@ -96,6 +98,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("wgrad_gemm_accum_fp32", &wgrad_gemm_accum_fp32);
   m.def("build_sample_idx", &build_sample_idx);
   m.def("build_blending_indices", &build_blending_indices);
+  m.def("fp8_quantize", &fp8_quantize);
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_bwd", &flash_attn_bwd);
 }

@@ -68,13 +68,15 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, input, weight, bias, gradient_accumulation_fusion,
-                async_grad_allreduce, sequence_parallel, fp8=False):
+                async_grad_allreduce, sequence_parallel, fp8=False,
+                fp8_meta=None):
         ctx.save_for_backward(input, weight)
         ctx.use_bias = bias is not None
         ctx.gradient_accumulation_fusion = gradient_accumulation_fusion
         ctx.async_grad_allreduce = async_grad_allreduce
         ctx.sequence_parallel = sequence_parallel
         ctx.fp8 = fp8 and input.is_cuda
+        ctx.fp8_meta = fp8_meta
 
         if sequence_parallel:
             world_size = ps.get_tensor_model_parallel_world_size()
@@ -92,12 +94,15 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
             total_input = input
 
         if ctx.fp8:
-            from ..fp8 import fp8_linear_fwd
+            from ..fp8 import fp8_linear_fwd, fp8_linear_fwd_delayed
 
             ti2d = total_input.contiguous().view(-1, total_input.shape[-1])
-            output = fp8_linear_fwd(ti2d, weight).view(
-                *total_input.shape[:-1], weight.shape[0]
-            )
+            if ctx.fp8_meta is not None:
+                out2d = fp8_linear_fwd_delayed(ti2d, weight,
+                                               ctx.fp8_meta["x"])
+            else:
+                out2d = fp8_linear_fwd(ti2d, weight)
+            output = out2d.view(*total_input.shape[:-1], weight.shape[0])
         else:
             output = torch.matmul(total_input, weight.t())
         if bias is not None:
@@ -125,12 +130,15 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
             total_input = input
 
         if ctx.fp8:
-            from ..fp8 import fp8_linear_dgrad
+            from ..fp8 import fp8_linear_dgrad, fp8_linear_dgrad_delayed
 
             go2d = grad_output.contiguous().view(-1, grad_output.shape[-1])
-            grad_input = fp8_linear_dgrad(go2d, weight).view(
-                *grad_output.shape[:-1], weight.shape[1]
-            )
+            if ctx.fp8_meta is not None:
+                gi2d = fp8_linear_dgrad_delayed(go2d, weight,
+                                                ctx.fp8_meta["dy"])
+            else:
+                gi2d = fp8_linear_dgrad(go2d, weight)
+            grad_input = gi2d.view(*grad_output.shape[:-1], weight.shape[1])
         else:
             grad_input = grad_output.matmul(weight)
 
@@ -169,23 +177,39 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
         if ctx.sequence_parallel and not ctx.async_grad_allreduce:
             rs_handle.wait()
             return (sub_grad_input, grad_weight, grad_bias, None, None, None,
-                    None)
+                    None, None)
         if ctx.async_grad_allreduce:
             allreduce_handle.wait()
-        return grad_input, grad_weight, grad_bias, None, None, None, None
+        return (grad_input, grad_weight, grad_bias, None, None, None, None,
+                None)
 
 
 def linear_with_grad_accumulation_and_async_allreduce(
     input, weight, bias, gradient_accumulation_fusion,
     async_grad_allreduce, sequence_parallel_enabled, fp8=False,
+    fp8_meta=None,
 ):
     return LinearWithGradAccumulationAndAsyncCommunication.apply(
         input, weight, bias, gradient_accumulation_fusion,
-        async_grad_allreduce, sequence_parallel_enabled, fp8,
+        async_grad_allreduce, sequence_parallel_enabled, fp8, fp8_meta,
     )
 
 
 # ---------------------------------------------------------------------------
+
+
+def _linear_fp8_meta(module, ref_tensor):
+    """Lazily create per-call-site delayed-scaling state (device-bound)."""
+    if not module.fp8 or not ref_tensor.is_cuda:
+        return None
+    if getattr(module, "_fp8_meta", None) is None:
+        from ..fp8 import Fp8TensorMeta
+
+        module._fp8_meta = {
+            "x": Fp8TensorMeta(ref_tensor.device),
+            "dy": Fp8TensorMeta(ref_tensor.device),
+        }
+    return module._fp8_meta
 
 
 def _initialize_affine_weight_gpu(weight, init_method, partition_dim, stride=1):
@@ -374,6 +398,7 @@ class ColumnParallelLinear(torch.nn.Module):
             self.gradient_accumulation_fusion,
             self.async_tensor_model_parallel_allreduce,
             self.sequence_parallel_enabled, self.fp8,
+            _linear_fp8_meta(self, input_parallel),
         )
         if self.gather_output:
             assert not self.sequence_parallel_enabled
@@ -460,7 +485,7 @@ class RowParallelLinear(torch.nn.Module):
             input_parallel, self.weight, None,
             self.gradient_accumulation_fusion,
             False,  # row-parallel dgrad needs no all-reduce
-            False, self.fp8,
+            False, self.fp8, _linear_fp8_meta(self, input_parallel),
         )
         if self.sequence_parallel_enabled:
             output_ = reduce_scatter_to_sequence_parallel_region(output_parallel)
