@@ -26,6 +26,9 @@ def main():
         # (fwd: X[M,K] @ W[K,N]) and the matching dgrad/wgrad orientations
         (M, H, QKV), (M, H, H), (M, H, 2 * FFN), (M, FFN, H), (M, H, V),
     ]
+    tune_fp8 = os.environ.get("TUNE_FP8", "1") == "1" and hasattr(
+        torch, "_scaled_mm"
+    )
     for (m, k, n) in shapes:
         x = torch.randn(m, k, device=dev, dtype=torch.bfloat16)
         w = torch.randn(n, k, device=dev, dtype=torch.bfloat16)
@@ -35,7 +38,23 @@ def main():
             dx = g @ w             # dgrad
             dw = g.t() @ x         # wgrad
         torch.cuda.synchronize()
-        print(f"tuned {m}x{k}x{n}", flush=True)
+        print(f"tuned bf16 {m}x{k}x{n}", flush=True)
+
+        if tune_fp8:
+            # fp8 recipe GEMMs (megatron_amd/fp8.py): fwd X8@W8^T and
+            # dgrad dY8@(W^T)8^T — TunableOp also covers ScaledGemm
+            one = torch.ones(1, device=dev)
+            x8 = x.to(torch.float8_e4m3fn)
+            w8 = w.to(torch.float8_e4m3fn)
+            wt8 = w.t().contiguous().to(torch.float8_e4m3fn)
+            g8 = g.to(torch.float8_e4m3fn)
+            for _ in range(3):
+                torch._scaled_mm(x8, w8.t(), scale_a=one, scale_b=one,
+                                 out_dtype=torch.bfloat16)
+                torch._scaled_mm(g8, wt8.t(), scale_a=one, scale_b=one,
+                                 out_dtype=torch.bfloat16)
+            torch.cuda.synchronize()
+            print(f"tuned fp8 {m}x{k}x{n}", flush=True)
     print("done")
 
 
