@@ -1,5 +1,14 @@
 """Micro-batch calculator: constant or ramped global batch size
-(reference megatron/microbatches.py:17-144)."""
+(reference megatron/microbatches.py:17-144).
+
+The number of microbatches per global step is
+global_batch_size / (micro_batch_size * data_parallel_size); with batch-size
+rampup the global batch grows from `start` to the target in `increment`
+steps spread evenly over `ramp_samples` consumed samples, and every
+intermediate size must stay divisible by micro_batch * dp. On MI355X the
+288 GB of HBM3E usually lets the micro-batch carry the whole per-GPU batch
+(num_microbatches == 1) until pipeline parallelism needs in-flight
+microbatches to fill the 1F1B schedule."""
 
 from __future__ import annotations
 

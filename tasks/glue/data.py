@@ -1,5 +1,11 @@
 """GLUE base dataset (reference tasks/glue/data.py): TSV files ->
-{text_a, text_b, label, uid} samples, tokenized lazily per item."""
+{text_a, text_b, label, uid} samples, tokenized lazily per item.
+
+Subclasses implement `process_samples_from_single_path` for their TSV
+schema (MNLI: sentence pair at columns 8/9, gold label last; QQP: questions
+at 3/4, is_duplicate at 5); this base class handles file iteration and the
+[CLS] A [SEP] B [SEP] packing via tasks.data_utils at __getitem__ time, so
+large train sets tokenize on the fly instead of up front."""
 
 from __future__ import annotations
 
