@@ -216,3 +216,16 @@ def test_checkpoint_resharding_roundtrip(tmp_path, dist_single):
     assert set(back_sd.keys()) == set(sd.keys())
     for key in sd:
         assert torch.equal(back_sd[key], sd[key]), key
+
+
+def test_bench_decode_cpu_smoke(dist_single):
+    """tools/bench_decode.py runs end to end on CPU with the tiny model
+    (eager path; the hipGraph path is GPU-only)."""
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "tools", "bench_decode.py"),
+         "--model", "llama2-tiny", "--tokens", "4", "--prompt", "4"],
+        capture_output=True, text=True, timeout=300,
+        env={**os.environ, "MASTER_PORT": "29681"},
+    )
+    assert r.returncode == 0, r.stderr[-800:]
+    assert "speedup" in r.stdout
