@@ -485,3 +485,76 @@ def _body_tp2_merge_matches_tp1_forward(rank):
 
 def test_tp2_merge_matches_tp1_forward():
     _spawn("_body_tp2_merge_matches_tp1_forward", 29609)
+
+
+def _body_deferred_clip_parity(rank):
+    """Deferred clip (flat-buffer norm + grad_scale folded into Adam) must
+    produce the same parameters as the eager clip path."""
+    from megatron_amd import parallel as mpu
+    from megatron_amd.config import TrainingConfig, set_config
+    from megatron_amd.models import LlamaModel
+    from megatron_amd.optim import get_megatron_optimizer
+    from megatron_amd.optim.adam import FusedAdam
+    from megatron_amd.parallel.ddp import DistributedDataParallel as LocalDDP
+
+    mpu.initialize_model_parallel(1, 1)  # dp = 2
+    mpu.model_parallel_cuda_manual_seed(99)
+
+    def build():
+        cfg = TrainingConfig(
+            num_layers=2, hidden_size=32, num_attention_heads=4,
+            num_attention_heads_kv=2, seq_length=16,
+            max_position_embeddings=32, micro_batch_size=1,
+            world_size=2, bf16=True, lr=1e-2, clip_grad=1e-4,  # always clips
+            hidden_dropout=0.0, attention_dropout=0.0,
+            use_cpu_initialization=True,
+        )
+        cfg.finalize()
+        cfg.pad_vocab_size(64)
+        set_config(cfg)
+        torch.manual_seed(7)
+        m = LlamaModel(cfg).bfloat16()
+        ddp = LocalDDP(m, True, True)
+        return cfg, ddp, get_megatron_optimizer([ddp], cfg)
+
+    def step(ddp, opt, cfg):
+        tokens = torch.randint(0, 60, (1, 16))
+        torch.distributed.broadcast(tokens, 0)
+        ddp.zero_grad_buffer()
+        opt.zero_grad()
+        out = ddp(tokens, torch.arange(16).unsqueeze(0), None, labels=tokens)
+        out.float().mean().backward()
+        opt.reduce_model_grads()
+        ok, grad_norm, _ = opt.step()
+        assert ok
+        return grad_norm
+
+    cfg1, ddp1, opt1 = build()
+    _INITIAL_SD = {k: v.clone()
+                   for k, v in ddp1.module.state_dict().items()}
+    n1 = step(ddp1, opt1, cfg1)
+
+    # disable the deferral: eager foreach-norm + mul path
+    orig = FusedAdam.supports_grad_scale
+    FusedAdam.supports_grad_scale = False
+    try:
+        cfg2, ddp2, opt2 = build()
+        # the RNG-tracker stream advanced during the first build: sync
+        # weights (and the optimizer's fp32 masters) to run 1's start point
+        ddp2.module.load_state_dict(
+            {k: v.clone() for k, v in _INITIAL_SD.items()}
+        )
+        opt2.reload_model_params()
+        n2 = step(ddp2, opt2, cfg2)
+    finally:
+        FusedAdam.supports_grad_scale = orig
+
+    assert abs(n1 - n2) / max(n1, 1e-12) < 1e-3, (n1, n2)
+    for p1, p2 in zip(ddp1.module.parameters(), ddp2.module.parameters()):
+        assert torch.allclose(p1.float(), p2.float(), atol=1e-5), (
+            (p1.float() - p2.float()).abs().max()
+        )
+
+
+def test_deferred_clip_parity():
+    _spawn("_body_deferred_clip_parity", 29610)
