@@ -229,3 +229,43 @@ def test_bench_decode_cpu_smoke(dist_single):
     )
     assert r.returncode == 0, r.stderr[-800:]
     assert "speedup" in r.stdout
+
+
+def test_blended_datasets(tmp_path, dist_single):
+    """Weighted blending of two tokenized corpora: sample mix tracks the
+    weights and every sample comes from the right source (native
+    build_blending_indices path)."""
+    import numpy as np
+
+    from megatron_amd.config import TrainingConfig, set_config
+    from megatron_amd.data import indexed_dataset
+    from megatron_amd.data.gpt_dataset import build_train_valid_test_datasets
+
+    cfg = TrainingConfig(seq_length=8)
+    cfg.finalize()
+    set_config(cfg)
+
+    prefixes = []
+    for tag, token in (("a", 7), ("b", 9)):
+        prefix = str(tmp_path / f"ds_{tag}")
+        b = indexed_dataset.make_builder(prefix + ".bin", dtype=np.int32)
+        for _ in range(40):
+            b.add_item(np.full(50, token, dtype=np.int32))
+            b.end_document()
+        b.finalize(prefix + ".idx")
+        prefixes.append(prefix)
+
+    train, valid, test = build_train_valid_test_datasets(
+        data_prefix=["0.75", prefixes[0], "0.25", prefixes[1]],
+        data_impl="mmap", splits_string="100,0,0",
+        train_valid_test_num_samples=[400, 0, 0],
+        seq_length=8, seed=123, skip_warmup=True,
+    )
+    counts = {7: 0, 9: 0}
+    n = 400
+    for i in range(n):
+        tok = int(train[i]["text"][0])
+        counts[tok] += 1
+    frac_a = counts[7] / n
+    assert 0.70 < frac_a < 0.80, counts  # tracks the 0.75 weight
+    assert counts[9] > 0
