@@ -558,3 +558,107 @@ def _body_deferred_clip_parity(rank):
 
 def test_deferred_clip_parity():
     _spawn("_body_deferred_clip_parity", 29610)
+
+
+def _body_distrib_optimizer_checkpoint(rank, tmpdir):
+    """ZeRO-1 checkpoint roundtrip: per-DP-rank optim.pt shards save and
+    restore bitwise (params, fp32 shards, Adam moments)."""
+    import os as _os
+
+    from megatron_amd import parallel as mpu
+    from megatron_amd.checkpointing import load_checkpoint, save_checkpoint
+    from megatron_amd.config import TrainingConfig, set_config
+    from megatron_amd.models import LlamaModel, ModelType
+    from megatron_amd.optim import (
+        get_megatron_optimizer, get_optimizer_param_scheduler,
+    )
+    from megatron_amd.parallel.ddp import DistributedDataParallel as LocalDDP
+
+    mpu.initialize_model_parallel(1, 1)  # dp = 2
+    mpu.model_parallel_cuda_manual_seed(5)
+
+    def build():
+        cfg = TrainingConfig(
+            num_layers=2, hidden_size=32, num_attention_heads=4,
+            num_attention_heads_kv=2, seq_length=16,
+            max_position_embeddings=32, micro_batch_size=1,
+            world_size=2, bf16=True, lr=1e-2, clip_grad=0.0,
+            use_distributed_optimizer=True, train_iters=4,
+            hidden_dropout=0.0, attention_dropout=0.0,
+            use_cpu_initialization=True,
+            save=tmpdir, load=tmpdir,
+        )
+        cfg.finalize()
+        cfg.pad_vocab_size(64)
+        set_config(cfg)
+        m = LlamaModel(cfg).bfloat16()
+        m.model_type = ModelType.encoder_or_decoder
+        ddp = LocalDDP(m, True, True)
+        opt = get_megatron_optimizer([ddp], cfg)
+        sched = get_optimizer_param_scheduler(opt, cfg)
+        return cfg, ddp, opt, sched
+
+    cfg, ddp, opt, sched = build()
+    # take one real step so Adam state exists
+    tokens = torch.randint(0, 60, (1, 16))
+    torch.distributed.broadcast(tokens, 0)
+    ddp.zero_grad_buffer()
+    opt.zero_grad()
+    out = ddp(tokens, torch.arange(16).unsqueeze(0), None, labels=tokens)
+    out.float().mean().backward()
+    opt.reduce_model_grads()
+    ok, _, _ = opt.step()
+    assert ok
+    save_checkpoint(3, [ddp], opt, sched, cfg)
+    torch.distributed.barrier()
+    assert _os.path.exists(_os.path.join(
+        tmpdir, "iter_0000003",
+        f"mp_rank_00_{mpu.get_data_parallel_rank():03d}", "optim.pt",
+    ))
+
+    params_before = [p.detach().clone() for p in ddp.module.parameters()]
+    shards_before = [
+        s.detach().clone()
+        for g in opt.shard_fp32_from_float16_groups for s in g
+    ]
+
+    cfg2, ddp2, opt2, sched2 = build()
+    it = load_checkpoint([ddp2], opt2, sched2, cfg2)
+    assert it == 3
+    for p1, p2 in zip(params_before, ddp2.module.parameters()):
+        assert torch.equal(p1, p2.detach())
+    shards_after = [
+        s.detach().clone()
+        for g in opt2.shard_fp32_from_float16_groups for s in g
+    ]
+    assert len(shards_before) == len(shards_after)
+    for a, b in zip(shards_before, shards_after):
+        assert torch.equal(a, b)
+
+
+def test_distrib_optimizer_checkpoint(tmp_path):
+    import functools
+
+    d = str(tmp_path)
+    mp.spawn(functools.partial(_dist_worker_args, args=(d,)),
+             args=("_body_distrib_optimizer_checkpoint", 29611),
+             nprocs=WORLD, join=True)
+
+
+def _dist_worker_args(rank, fn_name, port, args=()):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(WORLD)
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo", rank=rank, world_size=WORLD)
+    from megatron_amd import parallel as mpu
+
+    fn = globals()[fn_name]
+    try:
+        fn(rank, *args)
+    finally:
+        dist.barrier()
+        mpu.destroy_model_parallel()
+        dist.destroy_process_group()
