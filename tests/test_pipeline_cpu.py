@@ -243,3 +243,93 @@ def _body_interleaved(rank):
 
 def test_pp4_interleaved_trains():
     mp.spawn(_worker4, args=(29631,), nprocs=WORLD4, join=True)
+
+
+def _worker_pp2_checkpoint(rank, port, tmpdir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(WORLD)
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo", rank=rank, world_size=WORLD)
+    try:
+        _body_pp2_checkpoint(rank, tmpdir)
+    finally:
+        dist.barrier()
+        from megatron_amd import parallel as mpu
+
+        mpu.destroy_model_parallel()
+        dist.destroy_process_group()
+
+
+def _body_pp2_checkpoint(rank, tmpdir):
+    """PP=2 checkpoints use the mp_rank_{tp:02d}_{pp:03d} layout and
+    restore per-stage partitions + use_checkpoint_args round-trips the
+    architecture flags."""
+    from megatron_amd import parallel as mpu
+    from megatron_amd.checkpointing import (
+        load_args_from_checkpoint, load_checkpoint, save_checkpoint,
+    )
+    from megatron_amd.config import TrainingConfig, set_config
+    from megatron_amd.models import LlamaModel, ModelType
+    from megatron_amd.optim import (
+        get_megatron_optimizer, get_optimizer_param_scheduler,
+    )
+    from megatron_amd.parallel.ddp import DistributedDataParallel as LocalDDP
+
+    mpu.initialize_model_parallel(1, 2)
+    mpu.model_parallel_cuda_manual_seed(11)
+
+    def build():
+        cfg = TrainingConfig(
+            num_layers=4, hidden_size=64, num_attention_heads=4,
+            num_attention_heads_kv=2, seq_length=32,
+            max_position_embeddings=64, micro_batch_size=2,
+            global_batch_size=4, pipeline_model_parallel_size=2,
+            world_size=2, hidden_dropout=0.0, attention_dropout=0.0,
+            use_cpu_initialization=True, lr=1e-3, train_iters=4,
+            save=tmpdir, load=tmpdir,
+        )
+        cfg.finalize()
+        cfg.pad_vocab_size(128)
+        set_config(cfg)
+        m = LlamaModel(cfg, pre_process=mpu.is_pipeline_first_stage(),
+                       post_process=mpu.is_pipeline_last_stage())
+        m.model_type = ModelType.encoder_or_decoder
+        ddp = LocalDDP(m, True, True)
+        opt = get_megatron_optimizer([ddp], cfg)
+        sched = get_optimizer_param_scheduler(opt, cfg)
+        return cfg, ddp, opt, sched
+
+    cfg, ddp, opt, sched = build()
+    save_checkpoint(2, [ddp], opt, sched, cfg)
+    torch.distributed.barrier()
+    stage_dir = os.path.join(
+        tmpdir, "iter_0000002",
+        f"mp_rank_00_{mpu.get_pipeline_model_parallel_rank():03d}",
+    )
+    assert os.path.isdir(stage_dir), stage_dir
+
+    params_before = [p.detach().clone() for p in ddp.module.parameters()]
+    cfg2, ddp2, opt2, sched2 = build()
+    it = load_checkpoint([ddp2], opt2, sched2, cfg2)
+    assert it == 2
+    for a, b in zip(params_before, ddp2.module.parameters()):
+        assert torch.equal(a, b.detach())
+
+    # args restoration from the checkpoint
+    cfg3 = TrainingConfig(load=tmpdir, use_checkpoint_args=True,
+                          world_size=2,
+                          pipeline_model_parallel_size=2)
+    load_args_from_checkpoint(cfg3)
+    assert cfg3.num_layers == 4
+    assert cfg3.hidden_size == 64
+    assert cfg3.num_attention_heads_kv == 2
+
+
+def test_pp2_checkpoint_roundtrip(tmp_path):
+    import functools
+
+    mp.spawn(functools.partial(_worker_pp2_checkpoint, tmpdir=str(tmp_path)),
+             args=(29651,), nprocs=WORLD, join=True)
