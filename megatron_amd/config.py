@@ -297,6 +297,22 @@ class TrainingConfig:
             == 0
         ), "global batch must be a multiple of micro_batch * dp"
 
+        # sample-based training: derive the iteration budget (reference
+        # validate_args converts train_samples for the constant-batch case)
+        if self.train_iters is None and self.train_samples is not None:
+            assert self.rampup_batch_size is None, (
+                "use --train_iters with batch-size rampup"
+            )
+            self.train_iters = self.train_samples // self.global_batch_size
+            if self.lr_decay_samples is not None and self.lr_decay_iters is None:
+                self.lr_decay_iters = (
+                    self.lr_decay_samples // self.global_batch_size
+                )
+            if self.lr_warmup_samples and not self.lr_warmup_iters:
+                self.lr_warmup_iters = (
+                    self.lr_warmup_samples // self.global_batch_size
+                )
+
         # virtual pipeline
         if self.num_layers_per_virtual_pipeline_stage is not None:
             assert self.pipeline_model_parallel_size > 2, (

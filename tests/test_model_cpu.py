@@ -221,3 +221,18 @@ def test_kv_cache_incremental_decode_matches_full():
     assert torch.allclose(full_logits[:, 15], last[:, 0], atol=1e-4), (
         (full_logits[:, 15] - last[:, 0]).abs().max()
     )
+
+
+def test_train_samples_derives_iters():
+    from megatron_amd.config import TrainingConfig
+
+    cfg = TrainingConfig(
+        num_layers=2, hidden_size=32, num_attention_heads=4,
+        micro_batch_size=2, global_batch_size=8,
+        train_samples=800, lr_decay_samples=400, lr_warmup_samples=80,
+        lr=1e-4,
+    )
+    cfg.finalize()
+    assert cfg.train_iters == 100
+    assert cfg.lr_decay_iters == 50
+    assert cfg.lr_warmup_iters == 10
