@@ -101,11 +101,18 @@ def forward_step(data_iterator, model):
 
 def train_valid_test_datasets_provider(train_val_test_num_samples):
     cfg = get_config()
-    if cfg.data_path is None:
+    if cfg.data_path is None and not (cfg.train_data_path
+                                      or cfg.valid_data_path
+                                      or cfg.test_data_path):
         from megatron_amd.data.synthetic import build_synthetic_datasets
 
         return build_synthetic_datasets(cfg, train_val_test_num_samples)
     builder = instruct_datasets if cfg.model_type == "instruction" else gpt_datasets
+    kwargs = {}
+    if builder is gpt_datasets:
+        kwargs = dict(train_data_prefix=cfg.train_data_path,
+                      valid_data_prefix=cfg.valid_data_path,
+                      test_data_prefix=cfg.test_data_path)
     return builder(
         data_prefix=cfg.data_path,
         data_impl=cfg.data_impl,
@@ -114,6 +121,7 @@ def train_valid_test_datasets_provider(train_val_test_num_samples):
         seq_length=cfg.seq_length + 1,
         seed=cfg.seed,
         skip_warmup=(not cfg.mmap_warmup),
+        **kwargs,
     )
 
 

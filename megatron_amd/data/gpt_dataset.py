@@ -75,8 +75,41 @@ def get_train_valid_test_split_(splits_string, size):
 
 def build_train_valid_test_datasets(data_prefix, data_impl, splits_string,
                                     train_valid_test_num_samples, seq_length,
-                                    seed, skip_warmup):
-    """(reference gpt_dataset.py:20-124)"""
+                                    seed, skip_warmup,
+                                    train_data_prefix=None,
+                                    valid_data_prefix=None,
+                                    test_data_prefix=None):
+    """(reference gpt_dataset.py:20-124). When per-split prefixes are given
+    (--train_data_path / --valid_data_path / --test_data_path) each split is
+    built over its own corpus and the split string is ignored."""
+    if train_data_prefix or valid_data_prefix or test_data_prefix:
+        def one_split(prefix_list, n):
+            if not prefix_list:
+                return None
+            if len(prefix_list) == 1:
+                ds, _, _ = _build_train_valid_test_datasets(
+                    prefix_list[0], data_impl, "1,0,0", [n, 0, 0],
+                    seq_length, seed, skip_warmup,
+                )
+                return ds
+            prefixes, weights, nums = get_datasets_weights_and_num_samples(
+                prefix_list, [n, 0, 0]
+            )
+            parts = [
+                _build_train_valid_test_datasets(
+                    pref, data_impl, "1,0,0", [nm[0], 0, 0], seq_length,
+                    seed, skip_warmup,
+                )[0]
+                for pref, nm in zip(prefixes, nums)
+            ]
+            return BlendableDataset(parts, weights)
+
+        return (
+            one_split(train_data_prefix, train_valid_test_num_samples[0]),
+            one_split(valid_data_prefix, train_valid_test_num_samples[1]),
+            one_split(test_data_prefix, train_valid_test_num_samples[2]),
+        )
+
     if len(data_prefix) == 1:
         return _build_train_valid_test_datasets(
             data_prefix[0], data_impl, splits_string,

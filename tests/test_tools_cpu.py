@@ -269,3 +269,38 @@ def test_blended_datasets(tmp_path, dist_single):
     frac_a = counts[7] / n
     assert 0.70 < frac_a < 0.80, counts  # tracks the 0.75 weight
     assert counts[9] > 0
+
+
+def test_per_split_data_paths(tmp_path, dist_single):
+    """--train_data_path / --valid_data_path build each split from its own
+    corpus (split string ignored), matching the reference's behavior."""
+    import numpy as np
+
+    from megatron_amd.config import TrainingConfig, set_config
+    from megatron_amd.data import indexed_dataset
+    from megatron_amd.data.gpt_dataset import build_train_valid_test_datasets
+
+    cfg = TrainingConfig(seq_length=8)
+    cfg.finalize()
+    set_config(cfg)
+
+    def make(tag, token):
+        prefix = str(tmp_path / f"sp_{tag}")
+        b = indexed_dataset.make_builder(prefix + ".bin", dtype=np.int32)
+        for _ in range(10):
+            b.add_item(np.full(40, token, dtype=np.int32))
+            b.end_document()
+        b.finalize(prefix + ".idx")
+        return prefix
+
+    ptrain, pvalid = make("train", 3), make("valid", 5)
+    train, valid, test = build_train_valid_test_datasets(
+        data_prefix=None, data_impl="mmap", splits_string="969,30,1",
+        train_valid_test_num_samples=[50, 20, 0], seq_length=8, seed=1,
+        skip_warmup=True,
+        train_data_prefix=[ptrain], valid_data_prefix=[pvalid],
+    )
+    assert test is None
+    assert int(train[0]["text"][0]) == 3
+    assert int(valid[0]["text"][0]) == 5
+    assert len(train) >= 50 and len(valid) >= 20
