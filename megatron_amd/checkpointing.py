@@ -293,12 +293,20 @@ def load_checkpoint(model, optimizer, opt_param_scheduler, cfg=None,
 
     tracker_filename = get_checkpoint_tracker_filename(load_dir)
     if not os.path.isfile(tracker_filename):
+        if getattr(cfg, "exit_on_missing_checkpoint", False):
+            raise FileNotFoundError(
+                f"--exit_on_missing_checkpoint: no metadata file "
+                f"{tracker_filename}"
+            )
         print_rank_0(
             f"WARNING: could not find the metadata file {tracker_filename}; "
             "will not load any checkpoints and will start from random"
         )
         return 0
     iteration, release = read_metadata(tracker_filename)
+    if getattr(cfg, "load_iters", None):
+        # load a specific saved iteration instead of the tracker's latest
+        iteration, release = int(cfg.load_iters), False
 
     if cfg.use_distributed_optimizer:
         model_name, optim_name = get_checkpoint_name(

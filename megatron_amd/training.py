@@ -237,7 +237,9 @@ def train_step(forward_step_func, data_iterator, model, optimizer,
 
     optimizer.reduce_model_grads(timers)
 
-    timers("optimizer", log_level=1).start()
+    timers("optimizer", log_level=1).start(
+        barrier=cfg.barrier_with_L1_time
+    )
     update_successful, grad_norm, num_zeros_in_grad = optimizer.step(timers)
     timers("optimizer").stop()
 
@@ -356,13 +358,27 @@ def training_log(loss_dict, total_loss_dict, learning_rate, iteration,
         log_string += (
             f" number of nan iterations: {total_loss_dict[nan_iters_key]:3d} |"
         )
-        if writer:
+        if writer and iteration % cfg.tensorboard_log_interval == 0:
             writer.add_scalar("learning-rate", learning_rate, iteration)
             writer.add_scalar("tokens-per-sec", tokens_per_sec, iteration)
             if cfg.log_timers_to_tensorboard:
                 timers.write(
                     ["forward-compute", "backward-compute", "optimizer"],
                     writer, iteration, normalizer=cfg.log_interval,
+                )
+            if cfg.log_memory_to_tensorboard and torch.cuda.is_available():
+                stats = torch.cuda.memory_stats()
+                writer.add_scalar(
+                    "mem-allocated-bytes",
+                    stats.get("allocated_bytes.all.current", 0), iteration,
+                )
+                writer.add_scalar(
+                    "mem-reserved-bytes",
+                    stats.get("reserved_bytes.all.current", 0), iteration,
+                )
+                writer.add_scalar(
+                    "mem-max-allocated-bytes",
+                    stats.get("allocated_bytes.all.peak", 0), iteration,
                 )
         if wandb_writer:
             wandb_writer.add_scalar("learning-rate", learning_rate, iteration)
