@@ -116,7 +116,13 @@ class BertModel(MegatronModule):
         self._language_model_key = "language_model"
 
         if self.post_process:
-            self.pooler = Pooler(cfg.hidden_size, init_method)
+            # reference language_model.py:448: the pooler optionally uses
+            # xavier-uniform init
+            pooler_init = (
+                torch.nn.init.xavier_uniform_
+                if cfg.init_method_xavier_uniform else init_method
+            )
+            self.pooler = Pooler(cfg.hidden_size, pooler_init)
             self.lm_head = BertLMHead(
                 self.language_model.embedding.word_embeddings.weight.size(0)
                 if pre_process else mpu.divide(
