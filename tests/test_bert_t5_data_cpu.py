@@ -206,3 +206,55 @@ def test_pretrain_ict_forward_step(indexed_docs, fake_tokenizer, dist_single,
     assert torch.isfinite(loss)
     loss.backward()
     assert "retrieval loss" in stats
+
+
+def test_pretrain_bert_full_driver(tmp_path, dist_single, fake_tokenizer):
+    """pretrain() end to end for BERT: the entry's dataset provider over a
+    real indexed corpus, 2 train iters, eval, checkpoint save."""
+    import numpy as np
+
+    import pretrain_bert
+    from megatron_amd import global_state
+    from megatron_amd.config import TrainingConfig, set_config
+    from megatron_amd.models import ModelType
+    from megatron_amd.training import pretrain
+
+    p = str(tmp_path / "bertdocs")
+    builder = __import__(
+        "megatron_amd.data.indexed_dataset", fromlist=["make_builder"]
+    ).make_builder(p + ".bin", dtype=np.int32)
+    rng = np.random.RandomState(2)
+    for _ in range(12):
+        builder.add_item(rng.randint(6, 120, size=40).astype(np.int32))
+        builder.end_document()
+    builder.finalize(p + ".idx")
+
+    cfg = TrainingConfig(
+        model_name="bert", num_layers=2, hidden_size=64,
+        num_attention_heads=4, num_attention_heads_kv=4, seq_length=64,
+        max_position_embeddings=64, micro_batch_size=2, global_batch_size=2,
+        train_iters=2, lr=1e-3, min_lr=1e-4, lr_decay_style="constant",
+        lr_warmup_iters=0, eval_interval=10, eval_iters=1, log_interval=1,
+        hidden_dropout=0.0, attention_dropout=0.0,
+        use_cpu_initialization=True, position_embedding_type="absolute",
+        use_rms_norm=False, glu_activation=None, use_bias=True,
+        use_flash_attn=False, bert_binary_head=True, clip_grad=1.0,
+        data_path=[p], split="10,1,1", save=str(tmp_path / "ckpt"),
+        save_interval=100, world_size=1, rank=0,
+    )
+    cfg.finalize()
+    cfg.pad_vocab_size(128)
+    set_config(cfg)
+
+    pretrain(
+        pretrain_bert.train_valid_test_datasets_provider,
+        pretrain_bert.model_provider,
+        ModelType.encoder_or_decoder,
+        pretrain_bert.forward_step,
+        cfg=cfg,
+    )
+    import os as _os
+
+    assert _os.path.isfile(_os.path.join(
+        str(tmp_path / "ckpt"), "latest_checkpointed_iteration.txt"
+    ))
