@@ -304,3 +304,22 @@ def test_per_split_data_paths(tmp_path, dist_single):
     assert int(train[0]["text"][0]) == 3
     assert int(valid[0]["text"][0]) == 5
     assert len(train) >= 50 and len(valid) >= 20
+
+
+def test_verify_correctness_cli(dist_single):
+    """verify_correctness.py runs offline (--random_init) on a tiny llama
+    and reports per-iteration logit errors."""
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "verify_correctness.py"),
+         "--model_name", "llama2", "--random_init",
+         "--num_layers", "2", "--hidden_size", "64",
+         "--num_attention_heads", "4", "--num_attention_heads_kv", "2",
+         "--seq_length", "32", "--max_position_embeddings", "64",
+         "--micro_batch_size", "1", "--iters", "2",
+         "--use_cpu_initialization", "--make_vocab_size_divisible_by", "16"],
+        capture_output=True, text=True, timeout=600,
+        env={**os.environ, "MASTER_ADDR": "127.0.0.1",
+             "MASTER_PORT": "29691", "RANK": "0", "WORLD_SIZE": "1"},
+    )
+    assert r.returncode == 0, (r.stdout[-500:], r.stderr[-800:])
+    assert "max abs error" in r.stdout or "error" in r.stdout.lower()
