@@ -151,6 +151,10 @@ class TrainingConfig:
     DDP_impl: str = "local"
     use_contiguous_buffers_in_local_ddp: bool = True
     overlap_grad_reduce: bool = False
+    # elements (fp32) per overlapped all-reduce bucket; sized for the xGMI
+    # per-link bandwidth (160 MB buckets keep the ring busy without delaying
+    # the first launch) — the N=2..8 scaling tuning knob
+    overlap_bucket_numel: int = 40_000_000
     scatter_gather_tensors_in_pipeline: bool = True
     variable_seq_lengths: bool = False
     no_async_tensor_model_parallel_allreduce: bool = False

@@ -176,6 +176,7 @@ def get_model(model_provider_func, model_type=ModelType.encoder_or_decoder,
                     cfg.accumulate_allreduce_grads_in_fp32,
                     cfg.use_contiguous_buffers_in_local_ddp,
                     overlap_grad_reduce=cfg.overlap_grad_reduce,
+                    bucket_numel=cfg.overlap_bucket_numel,
                 )
                 for model_module in model
             ]
