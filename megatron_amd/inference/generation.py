@@ -118,6 +118,12 @@ def generate_tokens_probs_and_return_on_first_stage(
                                   attention_mask2use)
 
             if mpu.is_pipeline_last_stage():
+                if prevent_newline_after_colon:
+                    # disable "\n" right after ":" (reference generation.py
+                    # :191; token ids resolved through the live tokenizer)
+                    colon = tokenizer.tokenize(":")[0]
+                    newline = tokenizer.tokenize("\n")[0]
+                    logits[tokens2use[:, -1] == colon, -1, newline] = -1e10
                 last_token_logits = logits[:, -1, :]
                 new_sample = sample(
                     last_token_logits.float(), top_k=top_k, top_p=top_p,
@@ -144,14 +150,24 @@ def generate_tokens_probs_and_return_on_first_stage(
             )
             prev_context_length = context_length
 
-            # termination check
+            # termination check. The eol-based stops use the GPT-2 BPE ids
+            # the reference hard-codes (628 = "\n\n", 198 = "\n";
+            # generation.py:244-251) — tokenizer-dependent by design.
             if mpu.is_pipeline_last_stage():
                 if use_eod_token_for_early_termination:
-                    just_finished = (
-                        (tokens[:, context_length] == termination_id)
-                        & started
-                        & (is_generation_done == 0)
-                    )
+                    new_tok = tokens[:, context_length]
+                    if stop_on_double_eol:
+                        done_token = ((new_tok == 628) | (
+                            (new_tok == 198)
+                            & (tokens[:, context_length - 1] == 198)
+                        )) & started
+                    elif stop_on_eol:
+                        done_token = (
+                            (new_tok == 628) | (new_tok == 198)
+                        ) & started
+                    else:
+                        done_token = (new_tok == termination_id) & started
+                    just_finished = done_token & (is_generation_done == 0)
                     generated_sequence_lengths[just_finished.bool()] = (
                         context_length + 1
                     )

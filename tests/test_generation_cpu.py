@@ -143,3 +143,44 @@ def test_rest_server_contract(small_model):
 
     r = client.put("/api", json={})
     assert r.status_code == 400
+
+
+def test_stop_on_eol_and_colon_guard(small_model):
+    """stop_on_eol terminates at the newline token (GPT-2 id 198, matching
+    the reference's hard-coded convention) and prevent_newline_after_colon
+    bans a newline directly after ':'."""
+    import types
+
+    from megatron_amd import global_state
+    from megatron_amd.inference.generation import (
+        generate_tokens_probs_and_return_on_first_stage,
+    )
+
+    model, cfg = small_model
+
+    class Tok:
+        eod = 0
+        pad = 0
+        vocab_size = 100
+
+        def tokenize(self, text):
+            return {":": [25], "\n": [198 % 100]}.get(text, [1])
+
+        def detokenize(self, ids):
+            return " ".join(map(str, ids))
+
+    global_state.set_tokenizer(Tok())
+
+    tokens = torch.zeros(1, 24, dtype=torch.long)
+    tokens[:, :4] = torch.tensor([5, 6, 7, 8])
+    lengths = torch.tensor([4])
+    out, glen, _ = generate_tokens_probs_and_return_on_first_stage(
+        model, tokens.clone(), lengths, top_k=1, stop_on_eol=True,
+        prevent_newline_after_colon=True,
+    )
+    # newline id never follows a colon id
+    seq = out[0].tolist()
+    for a, b in zip(seq, seq[1:]):
+        assert not (a == 25 and b == 198 % 100)
+    # if a newline/eod was generated, the length reflects early stop
+    assert glen[0] <= 24
