@@ -349,6 +349,14 @@ class FlashAttnFunction(torch.autograd.Function):
     def forward(ctx, q, k, v, causal, softmax_scale, window_size, dropout_p,
                 training):
         ext = _C(q)
+        if training and dropout_p and dropout_p > 0.0:
+            # fail loudly: the CDNA4 FA kernels do not implement attention
+            # dropout (use attention_dropout 0, or the CoreAttention path)
+            raise NotImplementedError(
+                "flash attention with attention_dropout > 0 is not "
+                "supported; set --attention_dropout 0 or disable "
+                "--use_flash_attn"
+            )
         if softmax_scale is None:
             softmax_scale = 1.0 / math.sqrt(q.shape[-1])
         ctx.causal = causal

@@ -12,39 +12,47 @@ from .optimizer import FP32Optimizer, Float16OptimizerWithFloat16Params
 from .scheduler import OptimizerParamScheduler
 
 
-def _get_params_for_weight_decay_optimization(models):
-    """No weight decay on biases and norm weights (reference
-    megatron/optimizer/__init__.py:28-60)."""
-    weight_decay_params = {"params": []}
-    no_weight_decay_params = {"params": [], "weight_decay": 0.0}
+def _get_params_for_weight_decay_optimization(models, no_wd_decay_cond=None,
+                                              scale_lr_cond=None,
+                                              lr_mult=1.0):
+    """Split params into (wd x lr_mult) groups (reference
+    megatron/optimizer/__init__.py:28-60 + the training.py:320-350 cond
+    hooks): default policy = no weight decay on biases and norm weights;
+    `no_wd_decay_cond(name, param)` overrides it, `scale_lr_cond(name,
+    param)` moves a param into a group whose lr is multiplied by lr_mult
+    (used e.g. to train a task head faster than the backbone)."""
+    groups = {
+        (True, False): {"params": []},
+        (False, False): {"params": [], "weight_decay": 0.0},
+        (True, True): {"params": [], "lr_mult": lr_mult},
+        (False, True): {"params": [], "weight_decay": 0.0,
+                        "lr_mult": lr_mult},
+    }
     for module in models:
-        for module_ in module.modules():
-            if isinstance(module_, (LayerNorm, RMSNorm)):
-                no_weight_decay_params["params"].extend(
-                    [p for p in module_._parameters.values() if p is not None]
-                )
-            else:
-                weight_decay_params["params"].extend(
-                    [
-                        p
-                        for n, p in module_._parameters.items()
-                        if p is not None and n != "bias"
-                    ]
-                )
-                no_weight_decay_params["params"].extend(
-                    [
-                        p
-                        for n, p in module_._parameters.items()
-                        if p is not None and n == "bias"
-                    ]
-                )
-    return weight_decay_params, no_weight_decay_params
+        for mod_name, module_ in module.named_modules():
+            is_norm = isinstance(
+                module_, (LayerNorm, RMSNorm, torch.nn.LayerNorm)
+            )
+            for n, p in module_._parameters.items():
+                if p is None:
+                    continue
+                name = f"{mod_name}.{n}" if mod_name else n
+                if no_wd_decay_cond is not None:
+                    no_wd = no_wd_decay_cond(name, p)
+                else:
+                    no_wd = is_norm or n == "bias"
+                scale = bool(scale_lr_cond and scale_lr_cond(name, p))
+                groups[(not no_wd, scale)]["params"].append(p)
+    return tuple(groups.values())
 
 
 def get_megatron_optimizer(models, cfg, no_wd_decay_cond=None,
                            scale_lr_cond=None, lr_mult=1.0):
     param_groups = [
-        g for g in _get_params_for_weight_decay_optimization(models)
+        g
+        for g in _get_params_for_weight_decay_optimization(
+            models, no_wd_decay_cond, scale_lr_cond, lr_mult
+        )
         if g["params"]
     ]
 

@@ -195,7 +195,8 @@ def _setup_model_and_optimizer(model_provider_func, model_type, cfg,
     model = get_model(model_provider_func, model_type, cfg=cfg)
     unwrapped_model = unwrap_model(model, (LocalDDP, Float16Module))
 
-    optimizer = get_megatron_optimizer(model, cfg)
+    optimizer = get_megatron_optimizer(model, cfg, no_wd_decay_cond,
+                                       scale_lr_cond, lr_mult)
     opt_param_scheduler = get_optimizer_param_scheduler(optimizer, cfg)
 
     if cfg.load is not None:

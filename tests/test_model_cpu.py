@@ -236,3 +236,37 @@ def test_train_samples_derives_iters():
     assert cfg.train_iters == 100
     assert cfg.lr_decay_iters == 50
     assert cfg.lr_warmup_iters == 10
+
+
+def test_param_group_conds():
+    """no_wd_decay_cond / scale_lr_cond / lr_mult build the reference's
+    custom param groups (scheduler honors per-group lr_mult)."""
+    import torch as t
+
+    from megatron_amd.optim import _get_params_for_weight_decay_optimization
+
+    m = t.nn.Sequential(t.nn.Linear(4, 4), t.nn.LayerNorm(4))
+    groups = _get_params_for_weight_decay_optimization(
+        [m],
+        scale_lr_cond=lambda name, p: "1." in name,  # the LN params
+        lr_mult=5.0,
+    )
+    wd, no_wd, wd_scaled, no_wd_scaled = groups
+    assert len(wd["params"]) == 1          # linear weight
+    assert len(no_wd["params"]) == 1       # linear bias
+    assert no_wd_scaled["lr_mult"] == 5.0
+    assert len(no_wd_scaled["params"]) == 2  # LN weight+bias (norm => no wd)
+
+    groups = _get_params_for_weight_decay_optimization(
+        [m], no_wd_decay_cond=lambda name, p: False  # decay EVERYTHING
+    )
+    assert len(groups[0]["params"]) == 4
+    assert not groups[1]["params"]
+
+
+def test_flash_dropout_raises():
+    from megatron_amd.ops.functional import flash_attention
+
+    q = torch.randn(1, 8, 2, 16)
+    with pytest.raises(NotImplementedError):
+        flash_attention(q, q, q, dropout_p=0.1, training=True)
