@@ -24,6 +24,7 @@ from .optim import get_megatron_optimizer, get_optimizer_param_scheduler
 from .parallel.ddp import DistributedDataParallel as LocalDDP
 from .parallel.schedules import get_forward_backward_func
 from .utils import (
+    is_last_rank,
     calc_params_l2_norm,
     print_rank_0,
     print_rank_last,
@@ -530,11 +531,21 @@ def evaluate(forward_step_func, data_iterator, model,
                 mpu.get_data_parallel_world_size() * cfg.micro_batch_size
                 * mb.get_num_microbatches()
             )
+        collected_non_loss_data = None
+        if process_non_loss_data_func is not None and is_last_rank():
+            # one extra collection pass whose raw outputs go to the
+            # post-processor (reference training.py:813-818)
+            forward_backward_func = get_forward_backward_func(cfg)
+            collected_non_loss_data = forward_backward_func(
+                forward_step_func, data_iterator, model, None, cfg,
+                global_state.get_timers(), forward_only=True,
+                collect_non_loss_data=True,
+            )
     for model_module in model:
         model_module.train()
     for key in total_loss_dict:
         total_loss_dict[key] /= cfg.eval_iters * mb.get_num_microbatches()
-    return total_loss_dict, None
+    return total_loss_dict, collected_non_loss_data
 
 
 def evaluate_and_print_results(prefix, forward_step_func, data_iterator, model,
@@ -542,10 +553,13 @@ def evaluate_and_print_results(prefix, forward_step_func, data_iterator, model,
                                verbose=False):
     """(reference training.py:829-874)"""
     writer = global_state.get_tensorboard_writer()
-    total_loss_dict, _ = evaluate(
+    total_loss_dict, collected_non_loss_data = evaluate(
         forward_step_func, data_iterator, model, process_non_loss_data_func,
         cfg, verbose,
     )
+    if process_non_loss_data_func is not None and is_last_rank():
+        process_non_loss_data_func(collected_non_loss_data, iteration,
+                                   writer)
     string = f" validation loss at {prefix} | "
     for key in total_loss_dict:
         value = total_loss_dict[key].item() if torch.is_tensor(
