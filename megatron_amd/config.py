@@ -180,6 +180,9 @@ class TrainingConfig:
 
     # -- attention implementation --
     use_flash_attn: bool = True
+    # chunk the LM-head GEMM + cross entropy over the sequence (0 = off):
+    # bounds the fp32 logits transient at long sequence / large vocab
+    loss_chunk_size: int = 0
     masked_softmax_fusion: bool = True
     bias_gelu_fusion: bool = True
     bias_dropout_fusion: bool = True

@@ -22,20 +22,47 @@ from .module import MegatronModule
 def post_language_model_processing(lm_output, labels, logit_weights,
                                    parallel_output, fp16_lm_cross_entropy, cfg):
     """(reference gpt_model.py:18-42)."""
-    output = parallel_lm_logits(lm_output, logit_weights, parallel_output, cfg)
-
     if labels is None:
+        output = parallel_lm_logits(lm_output, logit_weights,
+                                    parallel_output, cfg)
         # [s b v] -> [b s v]
         return output.transpose(0, 1).contiguous()
 
     # [b s] -> [s b]
     labels = labels.transpose(0, 1).contiguous()
+
+    chunk = getattr(cfg, "loss_chunk_size", 0) or 0
+    if chunk > 0 and lm_output.shape[0] > chunk:
+        # chunk the head GEMM + CE over the sequence so the fp32 logits
+        # transient is chunk-sized (a [32k, 32k-vocab] fp32 logits buffer
+        # plus CE internals is tens of GB at long sequence); each chunk is
+        # activation-checkpointed, so only lm_output stays resident
+        import torch.utils.checkpoint as _ckpt
+
+        def _chunk_loss(out_c, lab_c):
+            logits_c = parallel_lm_logits(out_c, logit_weights,
+                                          parallel_output, cfg)
+            if fp16_lm_cross_entropy:
+                return mpu.vocab_parallel_cross_entropy(logits_c, lab_c)
+            return mpu.vocab_parallel_cross_entropy(logits_c.float(), lab_c)
+
+        parts = []
+        for s0 in range(0, lm_output.shape[0], chunk):
+            parts.append(_ckpt.checkpoint(
+                _chunk_loss, lm_output[s0 : s0 + chunk],
+                labels[s0 : s0 + chunk], use_reentrant=False,
+            ))
+        loss = torch.cat(parts, dim=0)
+        return loss.transpose(0, 1).contiguous()
+
+    output = parallel_lm_logits(lm_output, logit_weights, parallel_output,
+                                cfg)
     if fp16_lm_cross_entropy:
         assert output.dtype == torch.half
         loss = mpu.vocab_parallel_cross_entropy(output, labels)
     else:
         loss = mpu.vocab_parallel_cross_entropy(output.float(), labels)
-    # [s b] -> [b s]
+    # [s b] -> [s b] ... [s b] -> [b s]
     return loss.transpose(0, 1).contiguous()
 
 

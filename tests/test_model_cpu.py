@@ -270,3 +270,34 @@ def test_flash_dropout_raises():
     q = torch.randn(1, 8, 2, 16)
     with pytest.raises(NotImplementedError):
         flash_attention(q, q, q, dropout_p=0.1, training=True)
+
+
+def test_chunked_lm_loss_matches_unchunked():
+    """loss_chunk_size chunks the head GEMM + CE with checkpointing; loss
+    and input gradients must match the unchunked path."""
+    from megatron_amd.config import get_config
+
+    cfg = _tiny_cfg()
+    cfg.loss_chunk_size = 0
+    from megatron_amd.models import LlamaModel
+
+    torch.manual_seed(4)
+    m = LlamaModel(cfg)
+    tokens = torch.randint(0, 90, (2, 32))
+    pids = torch.arange(32).unsqueeze(0).expand(2, -1)
+
+    def run():
+        m.zero_grad()
+        loss = m(tokens, pids, None, labels=tokens).float().mean()
+        loss.backward()
+        g = next(
+            p.grad.clone() for p in m.parameters() if p.grad is not None
+        )
+        return loss.detach().clone(), g
+
+    cfg.loss_chunk_size = 0
+    l0, g0 = run()
+    cfg.loss_chunk_size = 8
+    l1, g1 = run()
+    assert torch.allclose(l0, l1, atol=1e-6), (l0, l1)
+    assert torch.allclose(g0, g1, atol=1e-6)
