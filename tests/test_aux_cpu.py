@@ -111,3 +111,72 @@ def test_signal_handler_consensus(dist_single):
         assert handler.signals_received() == [True]
     # original handler restored on exit
     assert signal.getsignal(signal.SIGUSR1) not in (None,)
+
+
+def _make_sched(style, warmup=10, decay=100, max_lr=1.0, min_lr=0.1):
+    import torch as t
+
+    from megatron_amd.optim.scheduler import OptimizerParamScheduler
+
+    p = t.nn.Parameter(t.zeros(1))
+    opt = t.optim.SGD([p], lr=max_lr)
+    return OptimizerParamScheduler(
+        opt, max_lr, min_lr, warmup, decay, style,
+        start_wd=0.1, end_wd=0.1, wd_incr_steps=decay,
+        wd_incr_style="constant",
+    )
+
+
+def test_lr_schedules():
+    import math as _math
+
+    # linear warmup reaches max_lr at warmup end
+    s = _make_sched("linear")
+    s.step(increment=10)
+    assert abs(s.get_lr() - 1.0) < 1e-9
+    # linear decay hits min_lr at decay end
+    s.step(increment=90)
+    assert abs(s.get_lr() - 0.1) < 1e-9
+
+    # cosine: halfway through decay = midpoint of (max, min)
+    s = _make_sched("cosine")
+    s.step(increment=10 + 45)  # halfway through the 90 decay steps
+    expected = 0.1 + (1.0 - 0.1) * 0.5 * (1 + _math.cos(_math.pi * 0.5))
+    assert abs(s.get_lr() - expected) < 1e-6
+
+    # constant stays at max after warmup
+    s = _make_sched("constant", min_lr=0.0)
+    s.step(increment=50)
+    assert abs(s.get_lr() - 1.0) < 1e-9
+
+    # state roundtrip
+    s = _make_sched("cosine")
+    s.step(increment=30)
+    sd = s.state_dict()
+    s2 = _make_sched("cosine")
+    s2.load_state_dict(sd)
+    assert abs(s.get_lr() - s2.get_lr()) < 1e-12
+
+
+def test_dynamic_grad_scaler():
+    from megatron_amd.optim.grad_scaler import DynamicGradScaler
+
+    sc = DynamicGradScaler(
+        initial_scale=2.0 ** 10, min_scale=1.0, growth_factor=2.0,
+        backoff_factor=0.5, growth_interval=2, hysteresis=2,
+    )
+    s0 = sc.scale.item()
+    # hysteresis: first inf does NOT backoff, second does
+    sc.update(found_inf=True)
+    assert sc.scale.item() == s0
+    sc.update(found_inf=True)
+    assert sc.scale.item() == s0 * 0.5
+    # growth after `growth_interval` clean steps
+    sc.update(found_inf=False)
+    sc.update(found_inf=False)
+    assert sc.scale.item() == s0
+    # state roundtrip
+    sd = sc.state_dict()
+    sc2 = DynamicGradScaler(2.0, 1.0, 2.0, 0.5, 2, 2)
+    sc2.load_state_dict(sd)
+    assert sc2.scale.item() == sc.scale.item()
