@@ -301,3 +301,47 @@ def test_chunked_lm_loss_matches_unchunked():
     l1, g1 = run()
     assert torch.allclose(l0, l1, atol=1e-6), (l0, l1)
     assert torch.allclose(g0, g1, atol=1e-6)
+
+
+def test_checkpoint_restores_rng_tracker(tmp_path):
+    """Resume restores torch + model-parallel RNG tracker states, so
+    post-resume dropout draws match an uninterrupted run (reference
+    checkpointing.py:217-240, 655-687)."""
+    from megatron_amd import parallel as mpu
+    from megatron_amd.checkpointing import load_checkpoint, save_checkpoint
+    from megatron_amd.models import LlamaModel, ModelType
+    from megatron_amd.optim import (
+        get_megatron_optimizer, get_optimizer_param_scheduler,
+    )
+    from megatron_amd.parallel.ddp import DistributedDataParallel as LocalDDP
+
+    cfg = _tiny_cfg(lr=1e-3, train_iters=4)
+    cfg.save = str(tmp_path)
+    cfg.load = str(tmp_path)
+    m = LlamaModel(cfg)
+    m.model_type = ModelType.encoder_or_decoder
+    ddp = LocalDDP(m, True, True)
+    opt = get_megatron_optimizer([ddp], cfg)
+    sched = get_optimizer_param_scheduler(opt, cfg)
+
+    torch.manual_seed(777)
+    with mpu.get_cuda_rng_tracker().fork():
+        _ = torch.rand(3)  # advance the tracker stream
+    save_checkpoint(1, [ddp], opt, sched, cfg)
+
+    # the "uninterrupted" continuation draws
+    ref_plain = torch.rand(4)
+    with mpu.get_cuda_rng_tracker().fork():
+        ref_tracked = torch.rand(4)
+
+    # perturb both streams, then resume
+    torch.manual_seed(123456)
+    with mpu.get_cuda_rng_tracker().fork():
+        _ = torch.rand(99)
+    load_checkpoint([ddp], opt, sched, cfg)
+
+    got_plain = torch.rand(4)
+    with mpu.get_cuda_rng_tracker().fork():
+        got_tracked = torch.rand(4)
+    assert torch.equal(ref_plain, got_plain)
+    assert torch.equal(ref_tracked, got_tracked)
