@@ -333,3 +333,54 @@ def test_pp2_checkpoint_roundtrip(tmp_path):
 
     mp.spawn(functools.partial(_worker_pp2_checkpoint, tmpdir=str(tmp_path)),
              args=(29651,), nprocs=WORLD, join=True)
+
+
+def _worker_varseq(rank, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(WORLD)
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo", rank=rank, world_size=WORLD)
+    try:
+        _body_varseq(rank)
+    finally:
+        dist.barrier()
+        from megatron_amd import parallel as mpu
+
+        mpu.destroy_model_parallel()
+        dist.destroy_process_group()
+
+
+def _body_varseq(rank):
+    """variable_seq_lengths p2p: the receiver learns the shape from the
+    pre-exchange, so stages can pass different sequence lengths each
+    microbatch (the instruction-tuning collator emits them)."""
+    from megatron_amd import parallel as mpu
+    from megatron_amd.config import TrainingConfig, set_config
+    from megatron_amd.parallel import p2p
+
+    mpu.initialize_model_parallel(1, 2)
+    cfg = TrainingConfig(
+        num_layers=2, hidden_size=8, num_attention_heads=2,
+        pipeline_model_parallel_size=2, world_size=2, micro_batch_size=1,
+        variable_seq_lengths=True, scatter_gather_tensors_in_pipeline=False,
+        params_dtype=torch.float32,
+    )
+    cfg.finalize()
+    set_config(cfg)
+
+    for seq in (5, 9, 3):  # changing shapes across "microbatches"
+        if mpu.is_pipeline_first_stage():
+            t = torch.full((seq, 1, 8), float(seq))
+            p2p.send_forward(t, cfg)
+        else:
+            # receiver passes the NOMINAL shape; the pre-exchange overrides
+            got = p2p.recv_forward((4, 1, 8), cfg, dtype_=torch.float32)
+            assert got.shape == (seq, 1, 8)
+            assert torch.all(got == seq)
+
+
+def test_variable_seq_p2p():
+    mp.spawn(_worker_varseq, args=(29671,), nprocs=WORLD, join=True)
