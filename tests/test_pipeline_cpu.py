@@ -384,3 +384,114 @@ def _body_varseq(rank):
 
 def test_variable_seq_p2p():
     mp.spawn(_worker_varseq, args=(29671,), nprocs=WORLD, join=True)
+
+
+def _worker_pp2_generation(rank, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(WORLD)
+    import torch.multiprocessing  # noqa
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo", rank=rank, world_size=WORLD)
+    try:
+        _body_pp2_generation(rank)
+    finally:
+        dist.barrier()
+        from megatron_amd import parallel as mpu
+
+        mpu.destroy_model_parallel()
+        dist.destroy_process_group()
+
+
+def _body_pp2_generation(rank):
+    """Greedy generation through the 2-stage pipeline (recv/send forward,
+    last->first token broadcasts) equals the single-stage run of the merged
+    model."""
+    import sys as _sys
+
+    from megatron_amd import global_state
+    from megatron_amd import parallel as mpu
+    from megatron_amd.config import TrainingConfig, set_config
+    from megatron_amd.inference.generation import (
+        generate_tokens_probs_and_return_on_first_stage,
+    )
+    from megatron_amd.models import LlamaModel
+    from megatron_amd.tokenizer.tokenizers import FakeTokenizer
+
+    _sys.path.insert(0, os.path.join(REPO_PIPE, "tools"))
+    from checkpoint_util import merge_full_state
+
+    global_state.set_tokenizer(FakeTokenizer(96))
+
+    def make_cfg(pp, ws):
+        cfg = TrainingConfig(
+            num_layers=4, hidden_size=64, num_attention_heads=4,
+            num_attention_heads_kv=2, seq_length=24,
+            max_position_embeddings=32, micro_batch_size=1,
+            pipeline_model_parallel_size=pp, world_size=ws,
+            hidden_dropout=0.0, attention_dropout=0.0,
+            use_cpu_initialization=True, use_flash_attn=False,
+        )
+        cfg.finalize()
+        cfg.pad_vocab_size(96)
+        set_config(cfg)
+        return cfg
+
+    # phase 1: PP2 model, generate
+    mpu.initialize_model_parallel(1, 2)
+    mpu.model_parallel_cuda_manual_seed(21)
+    cfg = make_cfg(2, 2)
+    m = LlamaModel(cfg, parallel_output=False,
+                   pre_process=mpu.is_pipeline_first_stage(),
+                   post_process=mpu.is_pipeline_last_stage())
+    m.eval()
+
+    tokens = torch.zeros(1, 16, dtype=torch.long)
+    tokens[:, :5] = torch.tensor([7, 11, 13, 17, 19])
+    torch.distributed.broadcast(tokens, 0)
+    lengths = torch.tensor([5])
+    with torch.no_grad():
+        out_pp2, _, _ = generate_tokens_probs_and_return_on_first_stage(
+            m, tokens.clone(), lengths, top_k=1,
+            use_eod_token_for_early_termination=False,
+        )
+
+    local_sd = {k: v.detach().clone()
+                for k, v in m.language_model.state_dict().items()}
+    gathered = [None, None]
+    torch.distributed.all_gather_object(gathered, local_sd)
+    out_pp2 = out_pp2.clone() if out_pp2 is not None else None
+    # ship rank0's generation result to compare on both ranks
+    holder = [out_pp2.tolist() if rank == 0 and out_pp2 is not None else None]
+    torch.distributed.broadcast_object_list(holder, src=0)
+    pp2_tokens = holder[0]
+    mpu.destroy_model_parallel()
+    torch.distributed.barrier()
+
+    # phase 2: merged single-stage model generates the same sequence
+    mpu.initialize_model_parallel(1, 1)
+    mpu.model_parallel_cuda_manual_seed(21)
+    cfg = make_cfg(1, 2)
+    shards = {(0, pp): {"model": gathered[pp]} for pp in range(2)}
+    full = merge_full_state(shards, 1, 2, 4, glu=True)
+    m1 = LlamaModel(cfg, parallel_output=False)
+    missing, unexpected = m1.language_model.load_state_dict(full,
+                                                            strict=False)
+    assert not unexpected, unexpected
+    assert not missing, missing
+    m1.eval()
+    with torch.no_grad():
+        out_pp1, _, _ = generate_tokens_probs_and_return_on_first_stage(
+            m1, tokens.clone(), lengths, top_k=1,
+            use_eod_token_for_early_termination=False,
+        )
+    assert out_pp1.tolist() == pp2_tokens, (out_pp1.tolist(), pp2_tokens)
+
+
+REPO_PIPE = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_pp2_generation_matches_merged():
+    mp.spawn(_worker_pp2_generation, args=(29681,), nprocs=WORLD, join=True)
