@@ -349,3 +349,69 @@ def test_orqa_supervised_forward(dist_single):
     assert torch.isfinite(loss)
     loss.backward()
     assert "in-batch acc" in stats
+
+
+def test_index_builder_and_blockdata_roundtrip(tmp_path, dist_single):
+    """IndexBuilder embeds blocks with the biencoder context tower, shards
+    save/merge/load, and the MIPS index retrieves the nearest block."""
+    import numpy as np
+
+    from megatron_amd import global_state
+    from megatron_amd.config import TrainingConfig, set_config
+    from megatron_amd.inference.indexer import IndexBuilder
+    from megatron_amd.inference.realm_index import (
+        BlockData, FaissMIPSIndex,
+    )
+    from megatron_amd.models.biencoder_model import BiEncoderModel
+
+    cfg = TrainingConfig(
+        model_name="bert", num_layers=2, hidden_size=64,
+        num_attention_heads=4, num_attention_heads_kv=4, seq_length=16,
+        max_position_embeddings=32, micro_batch_size=2,
+        hidden_dropout=0.0, attention_dropout=0.0,
+        use_cpu_initialization=True, position_embedding_type="absolute",
+        use_rms_norm=False, glu_activation=None, use_bias=True,
+        use_flash_attn=False, bert_binary_head=False,
+    )
+    cfg.finalize()
+    cfg.pad_vocab_size(VOCAB)
+    set_config(cfg)
+    global_state.set_tokenizer(WordTokenizer())
+    model = BiEncoderModel(cfg, projection_dim=16)
+
+    class Blocks(torch.utils.data.Dataset):
+        def __len__(self):
+            return 6
+
+        def __getitem__(self, i):
+            t = torch.full((16,), 6 + i, dtype=torch.long)
+            return {"context_tokens": t,
+                    "context_mask": torch.ones(16, dtype=torch.long),
+                    "block_id": i}
+
+    path = str(tmp_path / "embeds.pkl")
+    builder = IndexBuilder(model, Blocks(), batch_size=3,
+                           embedding_path=path)
+    bd = builder.build_and_save_index()
+    assert len(bd.embed_data) == 6
+
+    merged = BlockData(path)
+    merged.merge_shards_and_save([path + ".rank0"])
+    loaded = BlockData(path, load_from_path=True)
+    assert len(loaded.embed_data) == 6
+
+    index = FaissMIPSIndex(embed_size=16, embed_data=loaded, use_gpu=False)
+    q = model.embed_query(
+        torch.full((1, 16), 8, dtype=torch.long),
+        torch.ones(1, 16, dtype=torch.long),
+    ).detach()
+    scores, ids = index.search_mips_index(q, top_k=3)
+    assert scores.shape == (1, 3) and ids.shape == (1, 3)
+    # the index must agree with brute-force inner products
+    import numpy as _np
+
+    mat = _np.stack([loaded.embed_data[i] for i in range(6)]).astype(
+        _np.float32
+    )
+    brute = (q.numpy() @ mat.T)[0]
+    assert list(ids[0]) == list(_np.argsort(-brute)[:3])
