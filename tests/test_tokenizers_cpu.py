@@ -17,9 +17,10 @@ def gpt2_files(tmp_path):
     vocab = {"<|endoftext|>": 0}
     for i, ch in enumerate("helo wrd", start=1):
         vocab[ch] = i
-    vocab["he"] = 9
-    vocab["ll"] = 10
-    vocab["hell"] = 11
+    vocab["Ġ"] = 9  # byte-level BPE maps the space byte to this symbol
+    vocab["he"] = 10
+    vocab["ll"] = 11
+    vocab["hell"] = 12
     vf = tmp_path / "vocab.json"
     vf.write_text(json.dumps(vocab))
     mf = tmp_path / "merges.txt"
@@ -74,3 +75,38 @@ def test_build_tokenizer_pads_vocab(tmp_path):
     tok = build_tokenizer(cfg)
     assert cfg.padded_vocab_size % 16 == 0
     assert cfg.padded_vocab_size >= tok.vocab_size
+
+
+def test_preprocess_data_cli(tmp_path, gpt2_files):
+    """tools/preprocess_data.py: jsonl -> .bin/.idx with the GPT-2 BPE
+    tokenizer, tokens round-trip through the indexed dataset."""
+    import subprocess
+
+    import numpy as np
+
+    from megatron_amd.data import indexed_dataset
+
+    vf, mf = gpt2_files
+    inp = tmp_path / "docs.jsonl"
+    inp.write_text("\n".join(
+        json.dumps({"text": t}) for t in ["hello world", "hell he herd"]
+    ) + "\n")
+    prefix = str(tmp_path / "corpus")
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, os.path.join(repo, "tools", "preprocess_data.py"),
+         "--input", str(inp), "--output_prefix", prefix,
+         "--tokenizer_type", "GPT2BPETokenizer",
+         "--vocab_file", vf, "--merge_file", mf,
+         "--append_eod", "--workers", "1", "--chunk_size", "2"],
+        capture_output=True, text=True, timeout=300,
+    )
+    assert r.returncode == 0, r.stderr[-600:]
+
+    from megatron_amd.tokenizer.tokenizers import GPT2BPETokenizer
+
+    tok = GPT2BPETokenizer(vf, mf)
+    ds = indexed_dataset.make_dataset(prefix + "_text_document", "infer")
+    assert len(ds) == 2
+    expected = tok.tokenize("hello world") + [tok.eod]
+    assert list(np.asarray(ds[0])) == expected
