@@ -184,3 +184,29 @@ def test_stop_on_eol_and_colon_guard(small_model):
         assert not (a == 25 and b == 198 % 100)
     # if a newline/eod was generated, the length reflects early stop
     assert glen[0] <= 24
+
+
+def test_beam_search_scores_sorted_and_deterministic(small_model):
+    from megatron_amd.inference.generation import (
+        beam_search_and_return_on_first_stage,
+    )
+
+    m, cfg = small_model
+    tokens = torch.zeros(1, 12, dtype=torch.long)
+    tokens[:, :4] = torch.tensor([3, 5, 7, 9])
+    lengths = torch.tensor([4])
+
+    def run():
+        return beam_search_and_return_on_first_stage(
+            m, tokens.clone(), lengths, beam_size=3, stop_token=0,
+            num_return_gen=3, length_penalty=1.0,
+        )
+
+    out1, scores1 = run()
+    out2, scores2 = run()
+    # deterministic (no sampling in beam search)
+    assert torch.equal(out1, out2)
+    assert torch.allclose(scores1, scores2)
+    # returned hypotheses are best-first
+    s = scores1.tolist()
+    assert all(a >= b for a, b in zip(s, s[1:])), s
