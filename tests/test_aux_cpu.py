@@ -180,3 +180,36 @@ def test_dynamic_grad_scaler():
     sc2 = DynamicGradScaler(2.0, 1.0, 2.0, 0.5, 2, 2)
     sc2.load_state_dict(sd)
     assert sc2.scale.item() == sc.scale.item()
+
+
+def test_clip_grad_inf_norm_and_count_zeros(dist_single):
+    from megatron_amd import parallel as mpu
+    from megatron_amd.optim.clip_grads import (
+        clip_grad_norm_fp32, count_zeros_fp32,
+    )
+
+    if not mpu.model_parallel_is_initialized():
+        mpu.initialize_model_parallel(1, 1)
+
+    p = torch.nn.Parameter(torch.zeros(4))
+    p.grad = torch.tensor([3.0, -7.0, 0.0, 1.0])
+    norm = clip_grad_norm_fp32(
+        [p], [p.grad], max_norm=100.0, norm_type=torch.inf,
+        model_parallel_group=mpu.get_model_parallel_group(),
+    )
+    assert norm == 7.0
+
+    # L2 with clipping applied
+    p.grad = torch.tensor([3.0, 4.0, 0.0, 0.0])  # norm 5
+    norm = clip_grad_norm_fp32(
+        [p], [p.grad], max_norm=1.0,
+        model_parallel_group=mpu.get_model_parallel_group(),
+    )
+    assert abs(norm - 5.0) < 1e-6
+    assert abs(p.grad.norm().item() - 1.0) < 1e-5  # scaled down to max_norm
+
+    p.grad = torch.tensor([3.0, 0.0, 0.0, 1.0])
+    zeros = count_zeros_fp32(
+        [p], model_parallel_group=mpu.get_model_parallel_group()
+    )
+    assert zeros == 2
