@@ -353,3 +353,29 @@ def test_pretraining_sampler_resume(dist_single):
     )
     flat3 = [i for b in iter(s3) for i in b]
     assert flat3 == [2, 3, 6, 7]
+
+
+def test_finetune_instruction_loss_mask(dist_single):
+    """finetune.get_batch (instruction mode): loss covers assistant tokens
+    at weight 1 and the rest of the conversation at scalar_loss_mask."""
+    import finetune
+    from megatron_amd.config import TrainingConfig, set_config
+
+    cfg = TrainingConfig(model_type="instruction", scalar_loss_mask=0.25,
+                         seq_length=7)
+    cfg.finalize()
+    set_config(cfg)
+
+    batch = {
+        "text": torch.tensor([[5, 6, 7, 8, 9, 10, 11, 12]]),
+        # roles per TOKEN: prompter x4, assistant x3, pad x1
+        "assistant_mask": torch.tensor([[0, 0, 0, 0, 1, 1, 1, 0]]),
+        "pad_mask": torch.tensor([[1, 1, 1, 1, 1, 1, 1, 0]]),
+    }
+    tokens, labels, loss_mask, am, pids = finetune.get_batch(iter([batch]))
+    assert tokens.shape == (1, 7)
+    assert torch.equal(labels, batch["text"][:, 1:])
+    # positions 0..2 predict prompter tokens -> 0.25; 3..5 assistant -> 1;
+    # 6 predicts the pad token -> 0
+    expect = torch.tensor([[0.25, 0.25, 0.25, 1.0, 1.0, 1.0, 0.0]])
+    assert torch.allclose(loss_mask, expect), loss_mask
