@@ -61,9 +61,14 @@ class _VocabParallelCrossEntropy(torch.autograd.Function):
 
         loss = torch.log(sum_exp_logits) - predicted_logits
 
+        # normalize FIRST: the smoothing term needs true log-probabilities
+        # (reference cross_entropy.py:67-86 divides before smoothing)
+        exp_logits.div_(sum_exp_logits.unsqueeze(dim=-1))
+
         vocab_size = exp_logits.size(-1)
         if label_smoothing > 0:
-            # see reference cross_entropy.py:67-84
+            # NeMo-style smoothing (reference cross_entropy.py:70-86):
+            # (1 - alpha*K/(K-1)) * y_gt + alpha*K/(K-1) * mean_i(y_i)
             assert 1.0 > label_smoothing > 0.0
             smoothing = label_smoothing * vocab_size / (vocab_size - 1)
             log_probs = torch.log(exp_logits)
@@ -71,8 +76,6 @@ class _VocabParallelCrossEntropy(torch.autograd.Function):
             loss = (1.0 - smoothing) * loss - smoothing * mean_log_probs
 
         ctx.label_smoothing, ctx.vocab_size = label_smoothing, vocab_size
-
-        exp_logits.div_(sum_exp_logits.unsqueeze(dim=-1))
         ctx.save_for_backward(exp_logits, target_mask, masked_target_1d)
         return loss
 

@@ -213,3 +213,36 @@ def test_clip_grad_inf_norm_and_count_zeros(dist_single):
         [p], model_parallel_group=mpu.get_model_parallel_group()
     )
     assert zeros == 2
+
+
+def test_vocab_parallel_ce_label_smoothing(dist_single):
+    from megatron_amd import parallel as mpu
+
+    if not mpu.model_parallel_is_initialized():
+        mpu.initialize_model_parallel(1, 1)
+    torch.manual_seed(0)
+    s, b, v = 4, 2, 16
+    logits = torch.randn(s, b, v, requires_grad=True)
+    target = torch.randint(0, v, (s, b))
+    eps = 0.1
+    loss = mpu.vocab_parallel_cross_entropy(logits, target,
+                                            label_smoothing=eps)
+    # NeMo-style convention the reference implements:
+    # (1 - a) * CE - a * mean(log_softmax), a = eps * K / (K - 1)
+    lp = torch.log_softmax(logits.detach(), dim=-1)
+    ce = torch.nn.functional.cross_entropy(
+        logits.detach().view(-1, v), target.view(-1), reduction="none"
+    ).view(s, b)
+    a = eps * v / (v - 1)
+    ref = (1 - a) * ce - a * lp.mean(-1)
+    assert torch.allclose(loss, ref, atol=1e-5), (loss - ref).abs().max()
+    loss.sum().backward()
+    lr = logits.detach().clone().requires_grad_(True)
+    lp2 = torch.log_softmax(lr, dim=-1)
+    ce2 = torch.nn.functional.cross_entropy(
+        lr.view(-1, v), target.view(-1), reduction="none"
+    ).view(s, b)
+    ((1 - a) * ce2 - a * lp2.mean(-1)).sum().backward()
+    assert torch.allclose(logits.grad, lr.grad, atol=1e-5), (
+        (logits.grad - lr.grad).abs().max()
+    )
