@@ -210,3 +210,37 @@ def test_beam_search_scores_sorted_and_deterministic(small_model):
     # returned hypotheses are best-first
     s = scores1.tolist()
     assert all(a >= b for a, b in zip(s, s[1:])), s
+
+
+def test_sampling_modes():
+    """sample(): greedy argmax, top-k restriction, top-p nucleus mass,
+    temperature effect, vocab clamp (reference sampling.py:45-98)."""
+    from megatron_amd.inference.sampling import sample
+
+    torch.manual_seed(0)
+    logits = torch.tensor([[1.0, 5.0, 3.0, 0.5, -2.0]])
+
+    # greedy
+    assert sample(logits, top_k=1).item() == 1
+
+    # top-k=2: only the two best ids can ever be drawn
+    draws = {sample(logits.clone(), top_k=2).item() for _ in range(50)}
+    assert draws <= {1, 2}
+
+    # top-p: with p just over the top token's mass, only id 1 survives
+    probs = torch.softmax(logits, -1)
+    draws = {
+        sample(logits.clone(), top_p=probs[0, 1].item() - 0.01).item()
+        for _ in range(20)
+    }
+    assert draws == {1}
+
+    # near-zero temperature sharpens to greedy
+    draws = {
+        sample(logits.clone(), temperature=1e-4).item() for _ in range(20)
+    }
+    assert draws == {1}
+
+    # vocab_size clamps out-of-vocab padding logits
+    padded = torch.cat([logits, torch.full((1, 3), 100.0)], dim=1)
+    assert sample(padded, top_k=1, vocab_size=5).item() < 5
