@@ -345,3 +345,18 @@ def test_checkpoint_restores_rng_tracker(tmp_path):
         got_tracked = torch.rand(4)
     assert torch.equal(ref_plain, got_plain)
     assert torch.equal(ref_tracked, got_tracked)
+
+
+def test_rope_scaling_tables():
+    """Linear position interpolation: scaling_factor divides positions, so
+    the table at position p*f with scaling f equals the unscaled table at
+    p (reference positional_embeddings.py scaling semantics)."""
+    from megatron_amd.models.rope import precompute_freqs
+
+    cos1, sin1 = precompute_freqs(16, 64)
+    cos4, sin4 = precompute_freqs(16, 64, scaling_factor=4.0)
+    assert torch.allclose(cos4[20], cos1[5], atol=1e-6)
+    assert torch.allclose(sin4[20], sin1[5], atol=1e-6)
+    # theta change alters frequencies
+    cos_t, _ = precompute_freqs(16, 64, theta=1e6)
+    assert not torch.allclose(cos_t[10], cos1[10])
