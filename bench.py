@@ -170,13 +170,11 @@ def main():
     initialize_megatron(cfg=cfg)
 
     import megatron_amd.parallel as mpu
-    from megatron_amd import global_state
     from megatron_amd.models import MODEL_CLASSES, ModelType
     from megatron_amd.training import get_model, train_step
     from megatron_amd.optim import (
         get_megatron_optimizer, get_optimizer_param_scheduler,
     )
-    from megatron_amd.parallel import broadcast_data
     from megatron_amd.utils import get_ltor_masks_and_position_ids
     import functools
 

@@ -12,7 +12,6 @@ from __future__ import annotations
 import torch
 
 from megatron_amd import global_state
-from megatron_amd import parallel as mpu
 from megatron_amd.config import get_config
 from megatron_amd.data.gpt_dataset import (
     build_train_valid_test_datasets as gpt_datasets,
@@ -20,7 +19,6 @@ from megatron_amd.data.gpt_dataset import (
 from megatron_amd.data.instruction_dataset import (
     build_train_valid_test_datasets as instruct_datasets,
 )
-from megatron_amd.metrics import METRICS
 from megatron_amd.models import MODEL_CLASSES, ModelType
 from megatron_amd.parallel import broadcast_data
 from megatron_amd.training import pretrain

@@ -9,7 +9,6 @@ numpy path (cumsum + searchsorted, same output) as the no-extension fallback.
 
 from __future__ import annotations
 
-import hashlib
 import os
 import time
 

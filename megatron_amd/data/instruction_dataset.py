@@ -16,7 +16,6 @@ import numpy as np
 import torch
 
 from ..config import get_config
-from ..utils import print_rank_0
 from .blendable_dataset import BlendableDataset
 from .gpt_dataset import (
     get_datasets_weights_and_num_samples,

@@ -5,8 +5,6 @@ from __future__ import annotations
 
 import torch
 
-from .. import parallel as mpu
-from ..config import get_config
 from .realm_index import BlockData
 
 

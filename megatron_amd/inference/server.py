@@ -12,7 +12,6 @@ import threading
 
 import torch
 
-from .. import parallel as mpu
 from .api import beam_search_and_post_process, generate_and_post_process
 from .communication import broadcast_float_list
 

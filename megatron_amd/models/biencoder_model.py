@@ -6,7 +6,6 @@ from __future__ import annotations
 
 import torch
 
-from .. import parallel as mpu
 from .bert_model import bert_extended_attention_mask, bert_position_ids
 from .enums import AttnMaskType
 from .language_model import (

@@ -5,7 +5,7 @@ from __future__ import annotations
 
 import torch
 
-from .bert_model import BertModel, bert_extended_attention_mask, bert_position_ids
+from .bert_model import bert_extended_attention_mask, bert_position_ids
 from .language_model import TransformerLanguageModel, init_method_normal, scaled_init_method_normal
 from .bert_model import Pooler
 from .enums import AttnMaskType

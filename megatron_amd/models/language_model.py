@@ -7,11 +7,10 @@ Embedding :133-326, TransformerLanguageModel :329-638).
 from __future__ import annotations
 
 import torch
-import torch.nn.functional as F
 
 from .. import parallel as mpu
 from ..parallel import mappings
-from .enums import AttnMaskType, LayerType
+from .enums import AttnMaskType
 from .module import MegatronModule
 from .transformer import ParallelTransformer
 

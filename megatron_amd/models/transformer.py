@@ -17,14 +17,13 @@ from __future__ import annotations
 
 import math
 from contextlib import nullcontext
-from typing import Optional
 
 import torch
 import torch.nn.functional as F
 
 from .. import parallel as mpu
 from ..ops import functional as ops_f
-from .enums import AttnMaskType, AttnType, LayerType, PositionEmbeddingType
+from .enums import AttnMaskType, AttnType, LayerType
 from .module import MegatronModule
 from .norms import get_norm
 from .rope import apply_rotary_emb, precompute_freqs

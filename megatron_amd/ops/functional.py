@@ -15,8 +15,6 @@ Reference ops replaced (epfLLM/Megatron-LLM):
 from __future__ import annotations
 
 import math
-from typing import Optional
-
 import torch
 
 from . import ext as _ext

@@ -10,8 +10,6 @@ is the same math through _foreach.
 from __future__ import annotations
 
 import math
-from typing import List
-
 import torch
 
 

@@ -10,7 +10,6 @@ from __future__ import annotations
 
 import torch
 
-from .. import parallel as mpu
 from ..models.module import param_is_not_shared
 from ..utils import param_is_not_tensor_parallel_duplicate
 

@@ -19,8 +19,6 @@ MI355X shape of the same idea:
 
 from __future__ import annotations
 
-from typing import Dict, List
-
 import torch
 
 from .. import parallel as mpu

@@ -13,7 +13,6 @@ Reference: megatron/optimizer/optimizer.py:100-783. Responsibilities:
 from __future__ import annotations
 
 from abc import ABC, abstractmethod
-from itertools import chain
 
 import torch
 

@@ -12,7 +12,6 @@ CUDA_DEVICE_MAX_CONNECTIONS=1 serialization trick (layers.py:344-351).
 
 from __future__ import annotations
 
-import math
 from typing import Callable, Optional
 
 import torch

@@ -17,8 +17,6 @@ MI355X differences vs the reference:
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 
 from . import state as ps

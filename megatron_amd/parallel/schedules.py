@@ -9,7 +9,6 @@ deallocated output's shape check is skipped, :55-88).
 
 from __future__ import annotations
 
-from typing import List, Optional
 
 import torch
 from torch.autograd.variable import Variable
@@ -18,7 +17,6 @@ from .. import microbatches
 from ..models.enums import ModelType
 from ..utils import unwrap_model
 from . import p2p, state as ps
-from .utils import make_viewless_tensor
 
 
 def get_forward_backward_func(cfg):

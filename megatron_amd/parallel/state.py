@@ -19,7 +19,6 @@ runs as direct point-to-point RCCL transfers rather than multi-hop rings.
 from __future__ import annotations
 
 import os
-from datetime import timedelta
 from typing import List, Optional
 
 import torch

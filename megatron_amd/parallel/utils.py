@@ -3,7 +3,7 @@ megatron/core/utils.py)."""
 
 from __future__ import annotations
 
-from typing import List, Sequence
+from typing import List
 
 import torch
 

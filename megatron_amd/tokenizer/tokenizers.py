@@ -8,7 +8,6 @@ from __future__ import annotations
 
 from abc import ABC, abstractmethod
 
-from ..utils import print_rank_0
 
 
 def build_tokenizer(cfg):

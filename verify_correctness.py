@@ -20,7 +20,6 @@ from __future__ import annotations
 import torch
 
 from megatron_amd.checkpointing import load_checkpoint
-from megatron_amd.config import get_config
 from megatron_amd.initialize import initialize_megatron
 from megatron_amd.models import MODEL_CLASSES, ModelType
 from megatron_amd.training import get_model

@@ -56,8 +56,7 @@ LLAMA_SPECS = {
 def load_hf_state_dict(cache_dir: str) -> dict:
     """Load a HF checkpoint directory (safetensors or bin shards)."""
     import glob
-    import json
-
+    
     weights = {}
     st_files = sorted(glob.glob(os.path.join(cache_dir, "*.safetensors")))
     if st_files:
