@@ -193,3 +193,81 @@ def test_falcon_roundtrip_megatron_to_hf(dist_single):
     back = megatron_to_hf_falcon(sd, 2, 64, 4, 1, 128)
     for k, v in back.items():
         assert torch.equal(v, orig[k]), k
+
+
+def test_merge_meta_llama(tmp_path):
+    """Meta consolidated.XX.pth shards -> merged HF-style dict: column-
+    parallel weights stitch on dim 0, row-parallel on dim 1, norms are
+    replicated, rope.freqs dropped, and q/k rotary rows convert from Meta's
+    interleaved pairing to HF's half-split so the standard pipeline
+    (rearrange_qkv + permute_qkv) applies unchanged."""
+    from weights_conversion.hf_to_megatron import _permute_rotary_rows
+    from weights_conversion.merge_llama import merge_meta_llama
+
+    n_heads, head_dim, hidden, ffn, vocab = 4, 8, 32, 48, 64
+    torch.manual_seed(11)
+
+    # Build the FULL model in HF convention first, then shard it the way
+    # Meta does (2 shards), converting q/k to Meta's interleaved rows.
+    full = {
+        "tok_embeddings.weight": torch.randn(vocab, hidden),
+        "norm.weight": torch.randn(hidden),
+        "output.weight": torch.randn(vocab, hidden),
+        "layers.0.attention.wq.weight": torch.randn(hidden, hidden),
+        "layers.0.attention.wk.weight": torch.randn(hidden, hidden),
+        "layers.0.attention.wv.weight": torch.randn(hidden, hidden),
+        "layers.0.attention.wo.weight": torch.randn(hidden, hidden),
+        "layers.0.feed_forward.w1.weight": torch.randn(ffn, hidden),
+        "layers.0.feed_forward.w2.weight": torch.randn(hidden, ffn),
+        "layers.0.feed_forward.w3.weight": torch.randn(ffn, hidden),
+        "layers.0.attention_norm.weight": torch.randn(hidden),
+        "layers.0.ffn_norm.weight": torch.randn(hidden),
+        "rope.freqs": torch.randn(head_dim // 2),
+    }
+    # q/k in the Meta files use interleaved rotary rows = what our kernel
+    # uses = _permute_rotary_rows applied to the HF layout
+    meta_full = dict(full)
+    for key in ("layers.0.attention.wq.weight",
+                "layers.0.attention.wk.weight"):
+        meta_full[key] = _permute_rotary_rows(full[key], n_heads)
+
+    dim0 = {"wq", "wk", "wv", "w1", "w3", "output"}
+    dim1 = {"tok_embeddings", "wo", "w2"}
+    shards = [{}, {}]
+    for k, w in meta_full.items():
+        short = k.split(".")[-2]
+        if short in dim0:
+            halves = torch.chunk(w, 2, dim=0)
+        elif short in dim1:
+            halves = torch.chunk(w, 2, dim=1)
+        else:
+            halves = (w, w)
+        shards[0][k], shards[1][k] = halves[0], halves[1]
+    torch.save(shards[0], tmp_path / "consolidated.00.pth")
+    torch.save(shards[1], tmp_path / "consolidated.01.pth")
+
+    hf = merge_meta_llama(str(tmp_path), n_heads=n_heads)
+
+    assert "rope.freqs" not in hf and not any("rope" in k for k in hf)
+    assert torch.equal(hf["model.embed_tokens.weight"],
+                       full["tok_embeddings.weight"])
+    assert torch.equal(hf["lm_head.weight"], full["output.weight"])
+    assert torch.equal(hf["model.norm.weight"], full["norm.weight"])
+    assert torch.equal(hf["model.layers.0.input_layernorm.weight"],
+                       full["layers.0.attention_norm.weight"])
+    assert torch.equal(hf["model.layers.0.mlp.gate_proj.weight"],
+                       full["layers.0.feed_forward.w1.weight"])
+    assert torch.equal(hf["model.layers.0.mlp.down_proj.weight"],
+                       full["layers.0.feed_forward.w2.weight"])
+    assert torch.equal(hf["model.layers.0.self_attn.o_proj.weight"],
+                       full["layers.0.attention.wo.weight"])
+    # rotary rows restored to HF half-split convention exactly: sharded
+    # interleaved Meta q/k merge back to the original HF-layout tensors.
+    # NOTE: with 2 shards of 4 heads, each shard holds 2 whole heads, so
+    # per-head row permutations commute with the sharding.
+    assert torch.equal(hf["model.layers.0.self_attn.q_proj.weight"],
+                       full["layers.0.attention.wq.weight"])
+    assert torch.equal(hf["model.layers.0.self_attn.k_proj.weight"],
+                       full["layers.0.attention.wk.weight"])
+    assert torch.equal(hf["model.layers.0.self_attn.v_proj.weight"],
+                       full["layers.0.attention.wv.weight"])

@@ -280,7 +280,20 @@ def main():
     dtype = {"bf16": torch.bfloat16, "fp16": torch.float16,
              "fp32": torch.float32}[args.dtype]
 
-    weights = load_hf_state_dict(args.cache_dir)
+    import glob as _glob
+
+    meta_shards = _glob.glob(os.path.join(args.cache_dir,
+                                          "consolidated.*.pth"))
+    if meta_shards and args.model != "falcon":
+        # Meta-format checkpoint dir: merge shards -> HF-style dict
+        # (reference hf_to_megatron.py:284 via utils/merge_llama.py)
+        from weights_conversion.merge_llama import merge_meta_llama
+
+        spec0 = LLAMA_SPECS[(args.model, args.size)]
+        weights = merge_meta_llama(args.cache_dir, n_heads=spec0[2],
+                                   n_kv_heads=spec0[3])
+    else:
+        weights = load_hf_state_dict(args.cache_dir)
     if args.model == "falcon":
         n_layers = 32 if args.size == 7 else 60
         hidden = 4544 if args.size == 7 else 8192
