@@ -89,3 +89,47 @@ def test_filter_ngrams(tmp_path):
     kept, total = filter_corpus(str(inp), str(out), [str(task)], n=5)
     assert (kept, total) == (1, 2)
     assert json.loads(out.read_text())["url"] == "u1"
+
+
+def test_cleanup_fix_dataset(tmp_path):
+    """Short docs dropped, short javascript docs dropped, mojibake fixed,
+    repeated punctuation collapsed; filtered docs carry their reason."""
+    import json
+    import subprocess
+    import sys as _sys
+
+    long_text = "All work and no play makes for dull text. " * 20  # > 512
+    docs = [
+        {"text": "too short"},
+        {"text": "enable javascript to view this page please " * 3},
+        {"text": long_text + " wow!!!!!!!!"},
+    ]
+    inp = tmp_path / "in.json"
+    inp.write_text("\n".join(json.dumps(d) for d in docs) + "\n")
+    out = str(tmp_path / "owt")
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [_sys.executable,
+         os.path.join(repo, "tools", "openwebtext",
+                      "cleanup_fix_dataset.py"),
+         "--input_glob", str(inp), "--output_prefix", out],
+        capture_output=True, text=True, timeout=120,
+    )
+    assert r.returncode == 0, r.stderr[-500:]
+    kept = [json.loads(l) for l in
+            open(out + "_cleaned.json").read().splitlines()]
+    dropped = [json.loads(l) for l in
+               open(out + "_filtered.json").read().splitlines()]
+    assert len(kept) == 1 and len(dropped) == 2
+    assert "!!!!" not in kept[0]["text"] and "!!!" in kept[0]["text"]
+    # remove_512 is checked first, so both short docs drop with that reason
+    assert all(d["filter_reason"] == ["remove_512"] for d in dropped)
+
+    # with remove_512 disabled the javascript filter fires
+    from tools.openwebtext.cleanup_fix_dataset import process_doc
+
+    flags, _, _, filtered = process_doc(
+        json.dumps({"text": "please enable javascript to continue"}),
+        {"remove_256_javascript"},
+    )
+    assert filtered and flags["remove_256_javascript"]
