@@ -246,3 +246,27 @@ def test_vocab_parallel_ce_label_smoothing(dist_single):
     assert torch.allclose(logits.grad, lr.grad, atol=1e-5), (
         (logits.grad - lr.grad).abs().max()
     )
+
+
+def test_rampup_batch_size_calculator():
+    """Global batch grows from start to target in even increments over the
+    rampup samples; every intermediate size divides by micro*dp."""
+    from megatron_amd.microbatches import RampupBatchsizeNumMicroBatches
+
+    calc = RampupBatchsizeNumMicroBatches(
+        start_batch_size=4, batch_size_increment=4, ramup_samples=100,
+        global_batch_size=16, micro_batch_size=2, data_parallel_size=1,
+    )
+    # 3 increments of 4 over 100 samples -> one every 33.3 samples
+    assert calc.get_current_global_batch_size() == 4
+    assert calc.get() == 2  # 4 / (2*1)
+    calc.update(34, consistency_check=True)
+    assert calc.get_current_global_batch_size() == 8
+    calc.update(67, consistency_check=True)
+    assert calc.get_current_global_batch_size() == 12
+    calc.update(101, consistency_check=True)
+    assert calc.get_current_global_batch_size() == 16
+    assert calc.get() == 8
+    # never overshoots the target
+    calc.update(10_000, consistency_check=True)
+    assert calc.get_current_global_batch_size() == 16
