@@ -23,7 +23,10 @@ def parallel_lm_logits(input_, word_embeddings_weight, parallel_output, cfg,
         and mpu.get_tensor_model_parallel_world_size() > 1
         and not cfg.sequence_parallel
     )
-    if cfg.sequence_parallel:
+    # When the async all-reduce path is taken, the linear's backward already
+    # all-reduces grad_input across TP ranks; inserting the copy region too
+    # would reduce twice (reference language_model.py:32-40).
+    if async_grad_allreduce or cfg.sequence_parallel:
         input_parallel = input_
     else:
         input_parallel = mappings.copy_to_tensor_model_parallel_region(input_)

@@ -408,7 +408,12 @@ def _attn_mask(sq, sk, causal, window_size, device):
     if causal:
         m |= torch.ones(sq, sk, dtype=torch.bool, device=device).triu(sk - sq + 1)
     if window_size is not None and window_size > 0:
-        # keys further than window_size behind the query are masked
+        # Sliding-window convention (canonical here = HF/Mistral): each query
+        # attends to exactly `window_size` keys INCLUDING itself, i.e. keys
+        # with q_idx - window_size + 1 <= k_idx <= q_idx. Note the reference
+        # passes window_size=(w, w) to flash-attn, which keeps w+1 keys — a
+        # deliberate one-key departure, documented in PARITY.md; Mistral
+        # checkpoints are trained with the HF convention.
         q_idx = torch.arange(sq, device=device).unsqueeze(1) + (sk - sq)
         k_idx = torch.arange(sk, device=device).unsqueeze(0)
         m |= k_idx < (q_idx - window_size + 1)
@@ -470,8 +475,11 @@ class FusedQKVSplitRope(torch.autograd.Function):
     per layer in the backward (narrow-backward per slice, then fan-in sums);
     here the stride-aware RoPE kernel writes d(q)/d(k) straight into the
     corresponding regions of d_mixed and d(v) is one strided copy. q/k are
-    read as strided views (no forward copies); v is cloned so no
-    view-of-input is returned from the Function.
+    read as strided views (no forward copies). v is deliberately returned as
+    a strided VIEW of the input (saves a full copy, ~2 GB at seq 32k);
+    autograd tracks it as a view output, so any in-place write to v
+    downstream raises loudly rather than corrupting gradients — all
+    consumers (FA kernel, CoreAttention, KV-cache fill) only read it.
     """
 
     @staticmethod

@@ -137,24 +137,27 @@ def forward_backward_no_pipelining(
     assert len(model) == 1
     model = model[0]
 
-    context_handler = torch.no_grad if forward_only else None
+    import contextlib
+    context_handler = torch.no_grad if forward_only else contextlib.nullcontext
     forward_data_store = []
     input_tensor, output_tensor_grad = None, None
 
     num_microbatches = microbatches.get_num_microbatches()
     for i in range(num_microbatches - 1):
-        output_tensor = forward_step(
-            forward_step_func, data_iterator, model, input_tensor,
-            forward_data_store, cfg, timers, collect_non_loss_data,
-        )
+        with context_handler():
+            output_tensor = forward_step(
+                forward_step_func, data_iterator, model, input_tensor,
+                forward_data_store, cfg, timers, collect_non_loss_data,
+            )
         if not forward_only:
             backward_step(optimizer, input_tensor, output_tensor,
                           output_tensor_grad, cfg, timers)
 
-    output_tensor = forward_step(
-        forward_step_func, data_iterator, model, input_tensor,
-        forward_data_store, cfg, timers, collect_non_loss_data,
-    )
+    with context_handler():
+        output_tensor = forward_step(
+            forward_step_func, data_iterator, model, input_tensor,
+            forward_data_store, cfg, timers, collect_non_loss_data,
+        )
     if not forward_only:
         # arm the bucketed async DP all-reduce for the final backward so
         # grad communication overlaps the remaining backward compute

@@ -732,3 +732,74 @@ def _body_sp_backward_parity(rank):
 
 def test_sp_backward_parity():
     _spawn("_body_sp_backward_parity", 29612)
+
+
+def _body_async_allreduce_backward_parity(rank):
+    """TP2 grads with the async TP all-reduce (the default) equal grads with
+    it disabled. Guards the parallel_lm_logits input-region logic: inserting
+    copy_to_tensor_model_parallel_region AND async_grad_allreduce=True would
+    all-reduce the backbone grad twice (caught by the round-1 advisor)."""
+    from megatron_amd import parallel as mpu
+    from megatron_amd.config import TrainingConfig, set_config
+    from megatron_amd.models import LlamaModel, ModelType
+    from megatron_amd.optim import get_megatron_optimizer
+    from megatron_amd.parallel.ddp import DistributedDataParallel as LocalDDP
+    from megatron_amd.utils import get_ltor_masks_and_position_ids
+
+    mpu.initialize_model_parallel(2, 1)
+    mpu.model_parallel_cuda_manual_seed(1234)
+    torch.manual_seed(1234)
+
+    def make(no_async):
+        cfg = TrainingConfig(
+            num_layers=2, hidden_size=64, num_attention_heads=4,
+            num_attention_heads_kv=2, seq_length=16,
+            max_position_embeddings=32, micro_batch_size=1,
+            hidden_dropout=0.0, attention_dropout=0.0,
+            use_cpu_initialization=True, use_flash_attn=False,
+            tensor_model_parallel_size=2, world_size=2,
+            lr=1e-3, clip_grad=0.0,
+            no_async_tensor_model_parallel_allreduce=no_async,
+        )
+        cfg.finalize()
+        cfg.pad_vocab_size(96)
+        set_config(cfg)
+        return cfg
+
+    tokens = torch.randint(0, 90, (1, 17))
+    torch.distributed.broadcast(tokens, 0)
+    inp = tokens[:, :-1].contiguous()
+    labels = tokens[:, 1:].contiguous()
+    am, _, pids = get_ltor_masks_and_position_ids(inp, 0, False, False,
+                                                  False)
+
+    def grads(no_async, ref_sd=None):
+        cfg = make(no_async)
+        m = LlamaModel(cfg)
+        if ref_sd is not None:
+            m.load_state_dict({k: v.clone() for k, v in ref_sd.items()})
+        sd = {k: v.detach().clone() for k, v in m.state_dict().items()}
+        m.model_type = ModelType.encoder_or_decoder
+        ddp = LocalDDP(m, True, True)
+        opt = get_megatron_optimizer([ddp], cfg)
+        ddp.zero_grad_buffer()
+        opt.zero_grad()
+        out = ddp(inp, pids, am, labels=labels)
+        out.float().mean().backward()
+        opt.reduce_model_grads()
+        names = [n for n, _ in m.named_parameters()]
+        return sd, dict(zip(names, (p.main_grad.clone()
+                                    for p in m.parameters())))
+
+    sd, g_sync = grads(True)
+    _, g_async = grads(False, ref_sd=sd)
+    assert g_sync.keys() == g_async.keys()
+    for name in g_sync:
+        a, b = g_sync[name], g_async[name]
+        assert torch.allclose(a, b, atol=5e-4), (
+            name, (a - b).abs().max()
+        )
+
+
+def test_async_allreduce_backward_parity():
+    _spawn("_body_async_allreduce_backward_parity", 29613)
