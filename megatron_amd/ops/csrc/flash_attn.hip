@@ -368,6 +368,282 @@ __global__ __launch_bounds__(NW * 64) void fa_fwd_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// forward v8: 8-wave workgroup (2 waves/SIMD — the tuned CDNA4 attention
+// regime) with the T14 async-stage split: K/V tile loads are ISSUED one full
+// compute phase before their vmcnt wait + LDS write, so the HBM latency
+// hides under the previous tile's MFMAs instead of stalling at the top of
+// the iteration (the v4 stage_tr_image load+wait+write ran inline: PMC
+// showed 44% SQ_WAIT_ANY). Additions: static s_setprio(1) for the
+// younger dispatch half, per-wave skip of fully-masked causal tiles, and an
+// LDS-packed dwordx4 epilogue (v4 stored 32 scalar bf16 per lane —
+// issue-bound store tail).
+
+template <int D>
+__global__ __launch_bounds__(512, 2) void fa_fwd_v8_kernel(
+    const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
+    const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ out,
+    float* __restrict__ lse, int B, int Sq, int Sk, int Hq, int Hkv,
+    float scale, int causal, int window, TStr qs, TStr ks, TStr vs, TStr os) {
+  constexpr int NW = 8;
+  constexpr int KFRAGS = D / 32;
+  constexpr int DTILES = D / 16;
+  constexpr int IMG = tr_elems<D>(kBlockN);
+  constexpr int BM = NW * 16;  // 128 q rows per workgroup
+  constexpr int NT = NW * 64;  // 512 threads
+  // staging: one K image + one V image = 2 * 64 rows * (D/8) uint4 chunks
+  constexpr int CHUNKS = kBlockN * (D / 8);   // per image
+  constexpr int PER_THREAD = CHUNKS / NT;     // 2 at D=128, 1 at D=64
+  static_assert(CHUNKS % NT == 0);
+
+  __shared__ __hip_bfloat16 p_lds[NW][16 * kStrip];
+  __shared__ __hip_bfloat16 k_img[2][IMG];
+  __shared__ __hip_bfloat16 v_img[2][IMG];
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int row_in_tile = lane & 15;
+  const int kgroup = lane >> 4;
+
+  const int qb = blockIdx.x;
+  const int h = blockIdx.y;
+  const int b = blockIdx.z;
+  const int hkv = h / (Hq / Hkv);
+
+  const long q_base = (long)b * qs.bs + (long)h * qs.hs;
+  const long k_base = (long)b * ks.bs + (long)hkv * ks.hs;
+  const long v_base = (long)b * vs.bs + (long)hkv * vs.hs;
+  const long o_base = (long)b * os.bs + (long)h * os.hs;
+  const long rq = qs.ss;
+  const long rk = ks.ss;
+  const long rv = vs.ss;
+
+  const int qrow0 = qb * BM + wave * 16;
+  const int skq = Sk - Sq;
+
+  frag_b16 qf[KFRAGS];
+  {
+    int qrow = qrow0 + row_in_tile;
+    int qr = qrow < Sq ? qrow : Sq - 1;
+#pragma unroll
+    for (int kk = 0; kk < KFRAGS; ++kk) {
+      qf[kk] = global_read16(q + q_base + (long)qr * rq + kk * 32 +
+                             kgroup * 8);
+    }
+  }
+
+  frag_f32 o_acc[DTILES];
+#pragma unroll
+  for (int t = 0; t < DTILES; ++t) o_acc[t] = frag_f32{0.f, 0.f, 0.f, 0.f};
+  float m_run[4], l_run[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    m_run[r] = -1e30f;
+    l_run[r] = 0.f;
+  }
+
+  int kb_end = (Sk + kBlockN - 1) / kBlockN;
+  if (causal) {
+    int max_qrow = qb * BM + BM - 1;
+    int max_key = max_qrow + skq;
+    kb_end = min(kb_end, max_key / kBlockN + 1);
+  }
+  int kb_start = 0;
+  if (window > 0) {
+    int min_qrow = qb * BM;
+    int min_key = min_qrow + skq - window + 1;
+    if (min_key > 0) kb_start = min_key / kBlockN;
+  }
+
+  // per-thread staging registers: tile loads issued one iteration early
+  uint4 kreg[PER_THREAD], vreg[PER_THREAD];
+  // chunk idx -> (row, col8) of the 64 x D tile; write to tr_off(row, col8)
+  auto stage_load = [&](int kb) {
+#pragma unroll
+    for (int j = 0; j < PER_THREAD; ++j) {
+      int idx = threadIdx.x + j * NT;
+      int r = idx / (D / 8);
+      int c8 = (idx % (D / 8)) * 8;
+      int row = kb * kBlockN + r;
+      int rr = row < Sk ? row : Sk - 1;
+      uint4 kv = *reinterpret_cast<const uint4*>(k + k_base + (long)rr * rk +
+                                                 c8);
+      uint4 vv = *reinterpret_cast<const uint4*>(v + v_base + (long)rr * rv +
+                                                 c8);
+      if (row >= Sk) {
+        kv = make_uint4(0, 0, 0, 0);
+        vv = make_uint4(0, 0, 0, 0);
+      }
+      kreg[j] = kv;
+      vreg[j] = vv;
+    }
+  };
+  auto stage_write = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < PER_THREAD; ++j) {
+      int idx = threadIdx.x + j * NT;
+      int r = idx / (D / 8);
+      int c8 = (idx % (D / 8)) * 8;
+      *reinterpret_cast<uint4*>(&k_img[buf][tr_off<D>(r, c8)]) = kreg[j];
+      *reinterpret_cast<uint4*>(&v_img[buf][tr_off<D>(r, c8)]) = vreg[j];
+    }
+  };
+
+  stage_load(kb_start);
+  stage_write(kb_start & 1);
+  if (kb_start + 1 < kb_end) stage_load(kb_start + 1);
+  __syncthreads();
+
+  // static priority for the younger dispatch half (T5 static form): at 2
+  // waves/SIMD the second half loses VALU arbitration on every segment
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= NT / 2) {
+    __builtin_amdgcn_s_setprio(1);
+  }
+
+  for (int kb = kb_start; kb < kb_end; ++kb) {
+    const int kstart = kb * kBlockN;
+    const __hip_bfloat16* k_cur = k_img[kb & 1];
+    const __hip_bfloat16* v_cur = v_img[kb & 1];
+    // T14 write-late: publish tile kb+1 (loads issued LAST iteration, a
+    // full compute phase of latency slack), then issue tile kb+2's loads
+    if (kb + 1 < kb_end) stage_write((kb + 1) & 1);
+    if (kb + 2 < kb_end) stage_load(kb + 2);
+
+    // per-wave skip: every row of this wave's 16-row band masked out
+    bool wave_live = true;
+    if (causal && kstart > qrow0 + 15 + skq) wave_live = false;
+    if (window > 0 && kstart + kBlockN - 1 < qrow0 + skq - window + 1) {
+      wave_live = false;
+    }
+    if (wave_live) {
+      // S = Q K^T : B-fragments = K d-runs from LDS
+      frag_f32 st[4];
+#pragma unroll
+      for (int t = 0; t < 4; ++t) st[t] = frag_f32{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kk = 0; kk < KFRAGS; ++kk) {
+#pragma unroll
+        for (int t = 0; t < 4; ++t) {
+          frag_b16 bf = img_dfrag<D>(k_cur, t * 16 + row_in_tile,
+                                     kk * 32 + kgroup * 8);
+          st[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[kk], bf, st[t],
+                                                          0, 0, 0);
+        }
+      }
+
+      float s_val[4][4];
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int col = kstart + t * 16 + row_in_tile;
+          int row = qrow0 + kgroup * 4 + r;
+          bool masked = (col >= Sk) || (row >= Sq);
+          if (causal && col > row + skq) masked = true;
+          if (window > 0 && col < row + skq - window + 1) masked = true;
+          s_val[t][r] = masked ? -1e30f : st[t][r] * scale;
+        }
+      }
+
+      float m_new[4], alpha[4];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float pm = fmaxf(fmaxf(s_val[0][r], s_val[1][r]),
+                         fmaxf(s_val[2][r], s_val[3][r]));
+        pm = group16_max(pm);
+        m_new[r] = fmaxf(m_run[r], pm);
+        alpha[r] = __expf(m_run[r] - m_new[r]);
+        m_run[r] = m_new[r];
+      }
+
+      float p_val[4][4];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float acc = 0.f;
+#pragma unroll
+        for (int t = 0; t < 4; ++t) {
+          float pv = (s_val[t][r] < -1e29f) ? 0.f
+                                            : __expf(s_val[t][r] - m_new[r]);
+          p_val[t][r] = pv;
+          acc += pv;
+        }
+        float rowsum = group16_sum(acc);
+        l_run[r] = l_run[r] * alpha[r] + rowsum;
+#pragma unroll
+        for (int t = 0; t < DTILES; ++t) {
+          o_acc[t][r] *= alpha[r];
+        }
+      }
+
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row = kgroup * 4 + r;
+          int col = t * 16 + row_in_tile;
+          p_lds[wave][row * kStrip + col] = __float2bfloat16(p_val[t][r]);
+        }
+      }
+      // strips are wave-private: lgkmcnt ordering suffices, no barrier
+
+      // O += P V : A = P strip, B = V row-runs (transpose reads)
+#pragma unroll
+      for (int kk2 = 0; kk2 < 2; ++kk2) {
+        frag_b16 pf = lds_read16(
+            &p_lds[wave][row_in_tile * kStrip + kk2 * 32 + kgroup * 8]);
+#pragma unroll
+        for (int t = 0; t < DTILES; ++t) {
+          frag_b16 vf = tr_bfrag<D>(v_cur, kk2 * 32, t, lane);
+          o_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vf, o_acc[t],
+                                                             0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // epilogue: round-trip the 16 x D wave tile through the P strip so the
+  // global stores are 16-byte dwordx4 instead of 32 scalar bf16 per lane
+  float inv_l[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    inv_l[r] = l_run[r] > 0.f ? 1.0f / l_run[r] : 0.f;
+    int row = qrow0 + kgroup * 4 + r;
+    if (row < Sq && row_in_tile == 0) {
+      lse[((long)b * Hq + h) * Sq + row] =
+          m_run[r] + logf(fmaxf(l_run[r], 1e-30f));
+    }
+  }
+#pragma unroll
+  for (int half = 0; half < D / 64; ++half) {
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int srow = kgroup * 4 + r;
+        int scol = t * 16 + row_in_tile;
+        p_lds[wave][srow * kStrip + scol] =
+            __float2bfloat16(o_acc[half * 4 + t][r] * inv_l[r]);
+      }
+    }
+    // wave-private strip: no barrier
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      int e = j * 64 + lane;       // 8-elem chunk index in the 16x64 strip
+      int srow = e / 8;
+      int scol = (e % 8) * 8;
+      int row = qrow0 + srow;
+      if (row < Sq) {
+        uint4 val = *reinterpret_cast<const uint4*>(
+            &p_lds[wave][srow * kStrip + scol]);
+        *reinterpret_cast<uint4*>(
+            out + o_base + (long)row * os.ss + half * 64 + scol) = val;
+      }
+    }
+    // all lanes re-write the strip next half: wave-private, in-order LDS
+  }
+}
+
+// ---------------------------------------------------------------------------
 // backward preprocess: delta[b,h,s] = sum_d dO * O (fp32)
 
 template <int D>
@@ -795,8 +1071,31 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
   int win = window_size > 0 ? (int)window_size : 0;
   static const int nw_env = []() {
     const char* e = getenv("MEGATRON_AMD_FA_FWD_WAVES");
-    return e ? atoi(e) : 12;
+    return e ? atoi(e) : 8;  // 8 = v8 T14-split kernel (default)
   }();
+  if (nw_env == 8) {
+    dim3 grid((Sq + 127) / 128, Hq, B);
+    if (D == 128) {
+      hipLaunchKernelGGL((fa_fwd_v8_kernel<128>), grid, dim3(512), 0, stream,
+                         (const __hip_bfloat16*)q.data_ptr(),
+                         (const __hip_bfloat16*)k.data_ptr(),
+                         (const __hip_bfloat16*)v.data_ptr(),
+                         (__hip_bfloat16*)out.data_ptr(),
+                         lse.data_ptr<float>(), B, Sq, Sk, Hq, Hkv,
+                         (float)softmax_scale, causal ? 1 : 0, win, qs, ks,
+                         vs, os);
+    } else {
+      hipLaunchKernelGGL((fa_fwd_v8_kernel<64>), grid, dim3(512), 0, stream,
+                         (const __hip_bfloat16*)q.data_ptr(),
+                         (const __hip_bfloat16*)k.data_ptr(),
+                         (const __hip_bfloat16*)v.data_ptr(),
+                         (__hip_bfloat16*)out.data_ptr(),
+                         lse.data_ptr<float>(), B, Sq, Sk, Hq, Hkv,
+                         (float)softmax_scale, causal ? 1 : 0, win, qs, ks,
+                         vs, os);
+    }
+    return {out, lse};
+  }
 #define LAUNCH_FWD(DD, NW)                                                      do {                                                                            dim3 grid((Sq + NW * 16 - 1) / (NW * 16), Hq, B);                             hipLaunchKernelGGL((fa_fwd_kernel<DD, NW>), grid, dim3(NW * 64), 0,                              stream, (const __hip_bfloat16*)q.data_ptr(),                                  (const __hip_bfloat16*)k.data_ptr(),                                          (const __hip_bfloat16*)v.data_ptr(),                                          (__hip_bfloat16*)out.data_ptr(),                                              lse.data_ptr<float>(), B, Sq, Sk, Hq, Hkv,                                    (float)softmax_scale, causal ? 1 : 0, win,                                    qs, ks, vs, os);              } while (0)
   if (D == 128) {
     if (nw_env >= 12) LAUNCH_FWD(128, 12);
