@@ -264,6 +264,18 @@ __global__ __launch_bounds__(NW * 64) void fa_fwd_kernel(
                                      (kb + 1) * kBlockN, Sk);
     }
 
+    // per-wave skip: with BM = NW*16 q rows per workgroup, kb_end covers the
+    // LAST wave's diagonal — earlier waves' tail tiles are fully masked
+    bool wave_live = true;
+    if (causal && kstart > qrow0 + 15 + skq) wave_live = false;
+    if (window > 0 && kstart + kBlockN - 1 < qrow0 + skq - window + 1) {
+      wave_live = false;
+    }
+    if (!wave_live) {
+      __syncthreads();
+      continue;
+    }
+
     // S = Q K^T : B-fragments = K d-runs from LDS
     frag_f32 st[4];
 #pragma unroll
@@ -774,6 +786,16 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dkdv_kernel(
       stage_iter(it + 1, (it + 1) & 1);
     }
 
+    // per-wave skip: this wave's 16 keys all causally after (or all
+    // SWA-expired before) every q row of this tile
+    bool wave_live = true;
+    if (causal && key0 > qstart + kBlockM - 1 + skq) wave_live = false;
+    if (window > 0 && key0 + 15 < qstart + skq - window + 1) wave_live = false;
+    if (!wave_live) {
+      __syncthreads();
+      continue;
+    }
+
     // S^T = K Q^T ; dP^T = V dO^T : B-fragments are Q/dO d-runs from LDS.
     // One tile's fragment pair live at a time (VGPR pressure), softmax +
     // strip write immediately after each tile's K-reduction.
@@ -949,6 +971,18 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dq_kernel(
                                      (kb + 1) * kBlockN, Sk);
     }
 
+    // per-wave skip (mirror of the forward: kb_end covers the last wave's
+    // diagonal, earlier waves' tail tiles are fully masked)
+    bool wave_live = true;
+    if (causal && kstart > qrow0 + 15 + skq) wave_live = false;
+    if (window > 0 && kstart + kBlockN - 1 < qrow0 + skq - window + 1) {
+      wave_live = false;
+    }
+    if (!wave_live) {
+      __syncthreads();
+      continue;
+    }
+
     // S = Q K^T ; dP = dO V^T : B-fragments are K/V d-runs from LDS.
     // Tile-at-a-time (one st/dp fragment pair live) for VGPR pressure —
     // dS needs only the stored lse/delta, no cross-tile softmax state.
@@ -1071,7 +1105,9 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
   int win = window_size > 0 ? (int)window_size : 0;
   static const int nw_env = []() {
     const char* e = getenv("MEGATRON_AMD_FA_FWD_WAVES");
-    return e ? atoi(e) : 8;  // 8 = v8 T14-split kernel (default)
+    // 12 = the measured-best 12-wave kernel; 8 selects the T14-split
+    // experiment (measured SLOWER: 172 vs 277 TF — kept for A/B)
+    return e ? atoi(e) : 12;
   }();
   if (nw_env == 8) {
     dim3 grid((Sq + 127) / 128, Hq, B);
