@@ -28,6 +28,11 @@ class RMSNorm(torch.nn.Module):
     def forward(self, x):
         return ops_f.rmsnorm(x, self.weight, self.eps)
 
+    def forward_res(self, x):
+        """(normed, residual_passthrough) — the residual branch's gradient
+        is folded into the norm backward in-kernel (no autograd fan-in add)."""
+        return ops_f.rmsnorm_res(x, self.weight, self.eps)
+
 
 class LayerNorm(torch.nn.Module):
     def __init__(self, dim: int, eps: float = 1e-5, sequence_parallel: bool = False,
@@ -42,6 +47,9 @@ class LayerNorm(torch.nn.Module):
 
     def forward(self, x):
         return ops_f.layernorm(x, self.weight, self.bias, self.eps)
+
+    def forward_res(self, x):
+        return ops_f.layernorm_res(x, self.weight, self.bias, self.eps)
 
 
 def get_norm(cfg, sequence_parallel=None):

@@ -511,12 +511,25 @@ class ParallelTransformerLayer(MegatronModule):
     def forward(self, hidden_states, attention_mask, position_ids=None,
                 inference_params=None, encoder_output=None,
                 enc_dec_attn_mask=None):
-        # [s, b, h]
-        residual = hidden_states
-
-        ln_out = self.input_layernorm(hidden_states)
+        # [s, b, h]. Where hidden_states feeds both a norm and the residual
+        # stream, forward_res returns a pass-through second output so the
+        # residual gradient is folded into the norm backward in-kernel
+        # (no autograd fan-in add — see RMSNormResFunction).
+        fuse_res = not self.apply_residual_connection_post_layernorm
         if self.parallel_layernorm:
-            mlp_ln_out = self.mlp_layernorm(hidden_states)
+            # Falcon-40B: chain both norms' pass-throughs so each tensor has
+            # a single extra consumer
+            mlp_ln_out, h_mid = self.mlp_layernorm.forward_res(hidden_states)
+            if fuse_res:
+                ln_out, residual = self.input_layernorm.forward_res(h_mid)
+            else:
+                ln_out = self.input_layernorm(h_mid)
+                residual = hidden_states
+        elif fuse_res:
+            ln_out, residual = self.input_layernorm.forward_res(hidden_states)
+        else:
+            residual = hidden_states
+            ln_out = self.input_layernorm(hidden_states)
 
         attn_out, attn_bias = self.self_attention(
             ln_out, attention_mask, position_ids=position_ids,
@@ -547,21 +560,30 @@ class ParallelTransformerLayer(MegatronModule):
         )
 
         if self.layer_type == LayerType.decoder:
-            ln_cross = self.post_attention_layernorm(attn_res)
+            ln_cross, attn_res_b = (
+                self.post_attention_layernorm.forward_res(attn_res)
+            )
             cross_out, cross_bias = self.inter_attention(
                 ln_cross, enc_dec_attn_mask, encoder_output=encoder_output
             )
             attn_res = ops_f.bias_dropout_add(
-                cross_out, cross_bias, attn_res, self.hidden_dropout,
+                cross_out, cross_bias, attn_res_b, self.hidden_dropout,
                 self.training,
             )
-            ln2_out = self.post_inter_attention_layernorm(attn_res)
+            if fuse_res:
+                ln2_out, residual2 = (
+                    self.post_inter_attention_layernorm.forward_res(attn_res)
+                )
+            else:
+                ln2_out = self.post_inter_attention_layernorm(attn_res)
+                residual2 = ln2_out
+        elif fuse_res:
+            ln2_out, residual2 = (
+                self.post_attention_layernorm.forward_res(attn_res)
+            )
         else:
             ln2_out = self.post_attention_layernorm(attn_res)
-        if self.apply_residual_connection_post_layernorm:
             residual2 = ln2_out
-        else:
-            residual2 = attn_res
 
         mlp_out, mlp_bias = self.mlp(ln2_out)
         out = ops_f.bias_dropout_add(

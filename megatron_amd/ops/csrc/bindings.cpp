@@ -9,13 +9,15 @@ std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor weight,
                                        double eps);
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
                                        torch::Tensor weight,
-                                       torch::Tensor inv);
+                                       torch::Tensor inv,
+                                       c10::optional<torch::Tensor> dres);
 std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor weight,
                                          torch::Tensor bias, double eps);
 std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
                                          torch::Tensor weight,
                                          torch::Tensor mean,
-                                         torch::Tensor inv);
+                                         torch::Tensor inv,
+                                         c10::optional<torch::Tensor> dres);
 
 // softmax.hip
 torch::Tensor scaled_masked_softmax_fwd(torch::Tensor x,
@@ -81,9 +83,13 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
-  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd, py::arg("dy"), py::arg("x"),
+        py::arg("weight"), py::arg("inv"),
+        py::arg("dres") = py::none());
   m.def("layernorm_fwd", &layernorm_fwd);
-  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("layernorm_bwd", &layernorm_bwd, py::arg("dy"), py::arg("x"),
+        py::arg("weight"), py::arg("mean"), py::arg("inv"),
+        py::arg("dres") = py::none());
   m.def("scaled_masked_softmax_fwd", &scaled_masked_softmax_fwd);
   m.def("scaled_masked_softmax_bwd", &scaled_masked_softmax_bwd);
   m.def("glu_fwd", &glu_fwd);

@@ -133,6 +133,7 @@ __global__ void norm_bwd_kernel_bf16v(const __hip_bfloat16* __restrict__ dy,
                                       const __hip_bfloat16* __restrict__ weight,
                                       const float* __restrict__ mean,
                                       const float* __restrict__ inv,
+                                      const __hip_bfloat16* __restrict__ dres,
                                       __hip_bfloat16* __restrict__ dx,
                                       float* __restrict__ dweight,
                                       float* __restrict__ dbias, long rows,
@@ -185,11 +186,14 @@ __global__ void norm_bwd_kernel_bf16v(const __hip_bfloat16* __restrict__ dy,
     }
     const float inv_H = 1.0f / H;
 
+    const uint4* drr =
+        dres ? reinterpret_cast<const uint4*>(dres + row * (long)H) : nullptr;
     for (int i = threadIdx.x, c = 0; i < HV; i += BLOCK, ++c) {
-      Bf16x8 vg, vx, vw, vo;
+      Bf16x8 vg, vx, vw, vo, vr;
       vg.u = dyr[i];
       vx.u = xr[i];
       vw.u = wv[i];
+      if (drr) vr.u = drr[i];
 #pragma unroll
       for (int k = 0; k < 8; ++k) {
         float g = __bfloat162float(vg.h[k]);
@@ -203,6 +207,8 @@ __global__ void norm_bwd_kernel_bf16v(const __hip_bfloat16* __restrict__ dy,
         } else {
           d = (gw - gsum * inv_H - xhat * dot * inv_H) * r;
         }
+        // fused residual-grad accumulate: dx_total = norm_dx + dres
+        if (drr) d += __bfloat162float(vr.h[k]);
         vo.h[k] = __float2bfloat16(d);
         acc_dw[c][k] += g * xhat;
         if (!RMS) acc_db[c][k] += g;
@@ -264,6 +270,7 @@ __global__ void norm_bwd_kernel(const T* __restrict__ dy,
                                 const T* __restrict__ weight,
                                 const float* __restrict__ mean,
                                 const float* __restrict__ inv,
+                                const T* __restrict__ dres,
                                 T* __restrict__ dx,
                                 float* __restrict__ dweight,
                                 float* __restrict__ dbias, long rows, int H) {
@@ -315,6 +322,7 @@ __global__ void norm_bwd_kernel(const T* __restrict__ dy,
       } else {
         d = (gw - gsum * inv_H - xhat * dot * inv_H) * r;
       }
+      if (dres) d += DTypeTraits<T>::to_float(dres[row * (long)H + i]);
       dxr[i] = DTypeTraits<T>::from_float(d);
     }
     __syncthreads();
@@ -377,8 +385,9 @@ void norm_fwd_launch(const torch::Tensor& x, const torch::Tensor& w,
 template <typename T, bool RMS>
 void norm_bwd_launch(const torch::Tensor& dy, const torch::Tensor& x,
                      const torch::Tensor& w, const torch::Tensor* mean,
-                     const torch::Tensor& inv, torch::Tensor& dx,
-                     torch::Tensor& dw, torch::Tensor* db) {
+                     const torch::Tensor& inv, const torch::Tensor* dres,
+                     torch::Tensor& dx, torch::Tensor& dw,
+                     torch::Tensor* db) {
   long rows = x.size(0);
   int H = x.size(1);
   TORCH_CHECK(H <= kBlock * kMaxAcc,
@@ -405,6 +414,8 @@ void norm_bwd_launch(const torch::Tensor& dy, const torch::Tensor& x,
                          (const __hip_bfloat16*)w.data_ptr(),
                          RMS ? nullptr : mean->data_ptr<float>(),
                          inv.data_ptr<float>(),
+                         dres ? (const __hip_bfloat16*)dres->data_ptr()
+                              : nullptr,
                          (__hip_bfloat16*)dx.data_ptr(),
                          dw_part.data_ptr<float>(),
                          RMS ? nullptr : db_part.data_ptr<float>(), rows, H);
@@ -421,7 +432,9 @@ void norm_bwd_launch(const torch::Tensor& dy, const torch::Tensor& x,
                      dim3(kBlock), 0, stream, (const T*)dy.data_ptr(),
                      (const T*)x.data_ptr(), (const T*)w.data_ptr(),
                      RMS ? nullptr : mean->data_ptr<float>(),
-                     inv.data_ptr<float>(), (T*)dx.data_ptr(),
+                     inv.data_ptr<float>(),
+                     dres ? (const T*)dres->data_ptr() : nullptr,
+                     (T*)dx.data_ptr(),
                      dw.data_ptr<float>(),
                      RMS ? nullptr : db->data_ptr<float>(), rows, H);
 }
@@ -446,19 +459,27 @@ std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor weight,
 }
 
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
-                                       torch::Tensor weight,
-                                       torch::Tensor inv) {
+                                       torch::Tensor weight, torch::Tensor inv,
+                                       c10::optional<torch::Tensor> dres) {
   TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && x.is_contiguous());
+  const torch::Tensor* dr = nullptr;
+  torch::Tensor dres_c;
+  if (dres.has_value()) {
+    dres_c = dres->contiguous();
+    TORCH_CHECK(dres_c.scalar_type() == x.scalar_type() &&
+                dres_c.numel() == x.numel());
+    dr = &dres_c;
+  }
   auto dx = torch::empty_like(x);
   auto dw32 = torch::zeros({x.size(1)}, x.options().dtype(torch::kFloat32));
   if (x.scalar_type() == torch::kBFloat16) {
-    norm_bwd_launch<__hip_bfloat16, true>(dy, x, weight, nullptr, inv, dx,
+    norm_bwd_launch<__hip_bfloat16, true>(dy, x, weight, nullptr, inv, dr, dx,
                                           dw32, nullptr);
   } else if (x.scalar_type() == torch::kFloat16) {
-    norm_bwd_launch<__half, true>(dy, x, weight, nullptr, inv, dx, dw32,
+    norm_bwd_launch<__half, true>(dy, x, weight, nullptr, inv, dr, dx, dw32,
                                   nullptr);
   } else {
-    norm_bwd_launch<float, true>(dy, x, weight, nullptr, inv, dx, dw32,
+    norm_bwd_launch<float, true>(dy, x, weight, nullptr, inv, dr, dx, dw32,
                                  nullptr);
   }
   return {dx, dw32.to(weight.scalar_type())};
@@ -483,18 +504,28 @@ std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor weight,
 
 std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
                                          torch::Tensor weight,
-                                         torch::Tensor mean,
-                                         torch::Tensor inv) {
+                                         torch::Tensor mean, torch::Tensor inv,
+                                         c10::optional<torch::Tensor> dres) {
+  const torch::Tensor* dr = nullptr;
+  torch::Tensor dres_c;
+  if (dres.has_value()) {
+    dres_c = dres->contiguous();
+    TORCH_CHECK(dres_c.scalar_type() == x.scalar_type() &&
+                dres_c.numel() == x.numel());
+    dr = &dres_c;
+  }
   auto dx = torch::empty_like(x);
   auto dw32 = torch::zeros({x.size(1)}, x.options().dtype(torch::kFloat32));
   auto db32 = torch::zeros({x.size(1)}, x.options().dtype(torch::kFloat32));
   if (x.scalar_type() == torch::kBFloat16) {
-    norm_bwd_launch<__hip_bfloat16, false>(dy, x, weight, &mean, inv, dx,
+    norm_bwd_launch<__hip_bfloat16, false>(dy, x, weight, &mean, inv, dr, dx,
                                            dw32, &db32);
   } else if (x.scalar_type() == torch::kFloat16) {
-    norm_bwd_launch<__half, false>(dy, x, weight, &mean, inv, dx, dw32, &db32);
+    norm_bwd_launch<__half, false>(dy, x, weight, &mean, inv, dr, dx, dw32,
+                                   &db32);
   } else {
-    norm_bwd_launch<float, false>(dy, x, weight, &mean, inv, dx, dw32, &db32);
+    norm_bwd_launch<float, false>(dy, x, weight, &mean, inv, dr, dx, dw32,
+                                  &db32);
   }
   return {dx, dw32.to(weight.scalar_type()), db32.to(weight.scalar_type())};
 }

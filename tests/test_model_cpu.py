@@ -360,3 +360,41 @@ def test_rope_scaling_tables():
     # theta change alters frequencies
     cos_t, _ = precompute_freqs(16, 64, theta=1e6)
     assert not torch.allclose(cos_t[10], cos1[10])
+
+
+def test_norm_res_matches_fanin():
+    """rmsnorm_res/layernorm_res (pass-through residual, dres folded into
+    norm backward) must match the plain norm + autograd fan-in add."""
+    import torch
+    from megatron_amd.ops import functional as ops_f
+
+    torch.manual_seed(0)
+    for fn, fused, nargs in [
+        (ops_f.rmsnorm, ops_f.rmsnorm_res, 1),
+        (ops_f.layernorm, ops_f.layernorm_res, 2),
+    ]:
+        x = torch.randn(6, 32, requires_grad=True)
+        w = torch.ones(32, requires_grad=True) + 0.1 * torch.randn(32)
+        w = w.detach().requires_grad_(True)
+        b = torch.randn(32, requires_grad=True)
+        args = (w,) if nargs == 1 else (w, b)
+
+        # eager: norm + residual fan-in
+        y = fn(x, *args)
+        out = (y * 1.7).sum() + (x * 0.3).sum()
+        out.backward()
+        gx, gw = x.grad.clone(), w.grad.clone()
+        gb = b.grad.clone() if nargs == 2 else None
+
+        x.grad = None
+        w.grad = None
+        if nargs == 2:
+            b.grad = None
+        y2, res = fused(x, *args)
+        out2 = (y2 * 1.7).sum() + (res * 0.3).sum()
+        out2.backward()
+        assert torch.allclose(y, y2, atol=1e-6)
+        assert torch.allclose(x.grad, gx, atol=1e-5), (x.grad - gx).abs().max()
+        assert torch.allclose(w.grad, gw, atol=1e-5)
+        if nargs == 2:
+            assert torch.allclose(b.grad, gb, atol=1e-5)

@@ -58,6 +58,12 @@ class TestRMSNorm:
         assert rel_err(dx, xr.grad) < tol
         assert rel_err(dw, wr.grad) < tol
 
+        # fused residual-grad epilogue: dx_total = norm_dx + dres
+        dres = torch.randn_like(x)
+        dx2, dw2 = ext.rmsnorm_bwd(dy, x, w, inv, dres)
+        assert rel_err(dx2.float(), dx.float() + dres.float()) < 1e-3
+        assert rel_err(dw2, dw) < 1e-6
+
 
 class TestLayerNorm:
     @pytest.mark.parametrize("H", [4096, 1024])
