@@ -79,9 +79,19 @@ __device__ __forceinline__ float group16_sum(float v) {
 
 // --- padded tr-image -------------------------------------------------------
 
+// Bank-conflict-free tile-row geometry (derived analytically from the r02
+// PMC finding of +39% SQ_LDS_BANK_CONFLICT with the old D*4+8 stride, then
+// re-measured):
+//  - stride D*4+128 elems (== 0 mod 128) makes a d-run ds_read_b128's 16
+//    hardware-group lanes (rows t*16+0..15 at one col) land on 16 DISTINCT
+//    4-dword bank ranges = all 64 banks exactly once, via
+//  - the alternating +64-elem shift per tile-row PAIR (((rt>>1)&1)*64),
+//    which also puts a tr-read's partner tile-rows (ktile, ktile+2) exactly
+//    32 banks apart, so the 2x32-lane ds_read_b64_tr_b16 groups split the
+//    bank space instead of colliding.
 template <int D>
 constexpr int tr_stride() {
-  return D * 4 + 8;  // one 4-row tile-row + 8-element pad
+  return D * 4 + 128;
 }
 
 template <int D>
@@ -89,10 +99,14 @@ constexpr int tr_elems(int rows) {
   return (rows / 4) * tr_stride<D>();
 }
 
+__device__ __forceinline__ int tr_shift(int rowtile) {
+  return ((rowtile >> 1) & 1) * 64;
+}
+
 template <int D>
 __device__ __forceinline__ int tr_off(int row, int col) {
-  return (row >> 2) * tr_stride<D>() + (col >> 4) * 64 + (row & 3) * 16 +
-         (col & 15);
+  return (row >> 2) * tr_stride<D>() + tr_shift(row >> 2) + (col >> 4) * 64 +
+         (row & 3) * 16 + (col & 15);
 }
 
 __device__ __forceinline__ frag_b16 lds_read16(const __hip_bfloat16* p) {
@@ -128,8 +142,9 @@ __device__ __forceinline__ frag_b16 tr_bfrag(const __hip_bfloat16* img,
                                              int row0, int dtile, int lane) {
   const int c = lane & 15;
   const int ktile = (row0 >> 2) + (lane >> 4) * 2;
-  const int b1 = ktile * tr_stride<D>() + dtile * 64;
-  const int b2 = b1 + tr_stride<D>();
+  const int b1 = ktile * tr_stride<D>() + tr_shift(ktile) + dtile * 64;
+  const int b2 =
+      (ktile + 1) * tr_stride<D>() + tr_shift(ktile + 1) + dtile * 64;
   const __attribute__((address_space(3))) v4s* p1 =
       (const __attribute__((address_space(3))) v4s*)&img[b1] + c;
   const __attribute__((address_space(3))) v4s* p2 =
