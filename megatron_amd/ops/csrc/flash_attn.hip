@@ -395,6 +395,243 @@ __global__ __launch_bounds__(NW * 64) void fa_fwd_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// forward mb2: 8 waves x TWO 16-row bands per wave (BM = 256). Every K
+// d-run fragment and V tr-read pair feeds TWO MFMAs (one per band), halving
+// LDS fragment traffic per FLOP — the PMC-identified bottleneck (LDS array
+// ~48% busy vs MFMA pipe 14.5% at one band). 2 waves/SIMD (VGPR ~220).
+
+template <int D>
+__global__ __launch_bounds__(512, 2) void fa_fwd_mb2_kernel(
+    const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
+    const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ out,
+    float* __restrict__ lse, int B, int Sq, int Sk, int Hq, int Hkv,
+    float scale, int causal, int window, TStr qs, TStr ks, TStr vs, TStr os) {
+  constexpr int NW = 8;
+  constexpr int MB = 2;
+  constexpr int KFRAGS = D / 32;
+  constexpr int DTILES = D / 16;
+  constexpr int IMG = tr_elems<D>(kBlockN);
+  constexpr int BM = NW * 16 * MB;  // 256 q rows per workgroup
+  constexpr int NT = NW * 64;
+
+  __shared__ __hip_bfloat16 p_lds[NW][MB * 16 * kStrip];
+  __shared__ __hip_bfloat16 k_img[2][IMG];
+  __shared__ __hip_bfloat16 v_img[2][IMG];
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int row_in_tile = lane & 15;
+  const int kgroup = lane >> 4;
+
+  const int qb = blockIdx.x;
+  const int h = blockIdx.y;
+  const int b = blockIdx.z;
+  const int hkv = h / (Hq / Hkv);
+
+  const long q_base = (long)b * qs.bs + (long)h * qs.hs;
+  const long k_base = (long)b * ks.bs + (long)hkv * ks.hs;
+  const long v_base = (long)b * vs.bs + (long)hkv * vs.hs;
+  const long o_base = (long)b * os.bs + (long)h * os.hs;
+  const long rq = qs.ss;
+  const long rk = ks.ss;
+  const long rv = vs.ss;
+
+  // band mb covers rows [qrow0(mb), qrow0(mb)+16)
+  const int wrow0 = qb * BM + wave * (16 * MB);
+  const int skq = Sk - Sq;
+
+  frag_b16 qf[MB][KFRAGS];
+#pragma unroll
+  for (int mb = 0; mb < MB; ++mb) {
+    int qrow = wrow0 + mb * 16 + row_in_tile;
+    int qr = qrow < Sq ? qrow : Sq - 1;
+#pragma unroll
+    for (int kk = 0; kk < KFRAGS; ++kk) {
+      qf[mb][kk] = global_read16(q + q_base + (long)qr * rq + kk * 32 +
+                                 kgroup * 8);
+    }
+  }
+
+  frag_f32 o_acc[MB][DTILES];
+  float m_run[MB][4], l_run[MB][4];
+#pragma unroll
+  for (int mb = 0; mb < MB; ++mb) {
+#pragma unroll
+    for (int t = 0; t < DTILES; ++t) {
+      o_acc[mb][t] = frag_f32{0.f, 0.f, 0.f, 0.f};
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      m_run[mb][r] = -1e30f;
+      l_run[mb][r] = 0.f;
+    }
+  }
+
+  int kb_end = (Sk + kBlockN - 1) / kBlockN;
+  if (causal) {
+    int max_key = qb * BM + BM - 1 + skq;
+    kb_end = min(kb_end, max_key / kBlockN + 1);
+  }
+  int kb_start = 0;
+  if (window > 0) {
+    int min_key = qb * BM + skq - window + 1;
+    if (min_key > 0) kb_start = min_key / kBlockN;
+  }
+
+  stage_tr_image<D, kBlockN, NT>(k_img[kb_start & 1], k + k_base, rk,
+                                 kb_start * kBlockN, Sk);
+  stage_tr_image<D, kBlockN, NT>(v_img[kb_start & 1], v + v_base, rv,
+                                 kb_start * kBlockN, Sk);
+  __syncthreads();
+
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= NT / 2) {
+    __builtin_amdgcn_s_setprio(1);
+  }
+
+  for (int kb = kb_start; kb < kb_end; ++kb) {
+    const int kstart = kb * kBlockN;
+    const __hip_bfloat16* k_cur = k_img[kb & 1];
+    const __hip_bfloat16* v_cur = v_img[kb & 1];
+    if (kb + 1 < kb_end) {
+      stage_tr_image<D, kBlockN, NT>(k_img[(kb + 1) & 1], k + k_base, rk,
+                                     (kb + 1) * kBlockN, Sk);
+      stage_tr_image<D, kBlockN, NT>(v_img[(kb + 1) & 1], v + v_base, rv,
+                                     (kb + 1) * kBlockN, Sk);
+    }
+
+    bool wave_live = true;
+    if (causal && kstart > wrow0 + 16 * MB - 1 + skq) wave_live = false;
+    if (window > 0 &&
+        kstart + kBlockN - 1 < wrow0 + skq - window + 1) {
+      wave_live = false;
+    }
+    if (!wave_live) {
+      __syncthreads();
+      continue;
+    }
+
+    // S = Q K^T, both bands per B-fragment read
+    frag_f32 st[MB][4];
+#pragma unroll
+    for (int mb = 0; mb < MB; ++mb) {
+#pragma unroll
+      for (int t = 0; t < 4; ++t) st[mb][t] = frag_f32{0.f, 0.f, 0.f, 0.f};
+    }
+#pragma unroll
+    for (int kk = 0; kk < KFRAGS; ++kk) {
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        frag_b16 bf = img_dfrag<D>(k_cur, t * 16 + row_in_tile,
+                                   kk * 32 + kgroup * 8);
+#pragma unroll
+        for (int mb = 0; mb < MB; ++mb) {
+          st[mb][t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              qf[mb][kk], bf, st[mb][t], 0, 0, 0);
+        }
+      }
+    }
+
+#pragma unroll
+    for (int mb = 0; mb < MB; ++mb) {
+      const int qrow0 = wrow0 + mb * 16;
+      float s_val[4][4];
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int col = kstart + t * 16 + row_in_tile;
+          int row = qrow0 + kgroup * 4 + r;
+          bool masked = (col >= Sk) || (row >= Sq);
+          if (causal && col > row + skq) masked = true;
+          if (window > 0 && col < row + skq - window + 1) masked = true;
+          s_val[t][r] = masked ? -1e30f : st[mb][t][r] * scale;
+        }
+      }
+      float m_new[4], alpha[4];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float pm = fmaxf(fmaxf(s_val[0][r], s_val[1][r]),
+                         fmaxf(s_val[2][r], s_val[3][r]));
+        pm = group16_max(pm);
+        m_new[r] = fmaxf(m_run[mb][r], pm);
+        alpha[r] = __expf(m_run[mb][r] - m_new[r]);
+        m_run[mb][r] = m_new[r];
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float acc = 0.f;
+#pragma unroll
+        for (int t = 0; t < 4; ++t) {
+          float pv = (s_val[t][r] < -1e29f)
+                         ? 0.f
+                         : __expf(s_val[t][r] - m_new[r]);
+          s_val[t][r] = pv;
+          acc += pv;
+        }
+        float rowsum = group16_sum(acc);
+        l_run[mb][r] = l_run[mb][r] * alpha[r] + rowsum;
+#pragma unroll
+        for (int t = 0; t < DTILES; ++t) {
+          o_acc[mb][t][r] *= alpha[r];
+        }
+      }
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int srow = mb * 16 + kgroup * 4 + r;
+          p_lds[wave][srow * kStrip + t * 16 + row_in_tile] =
+              __float2bfloat16(s_val[t][r]);
+        }
+      }
+    }
+    // strips wave-private: no barrier
+
+    // O += P V, both bands per V tr-read pair
+#pragma unroll
+    for (int kk2 = 0; kk2 < 2; ++kk2) {
+      frag_b16 pf[MB];
+#pragma unroll
+      for (int mb = 0; mb < MB; ++mb) {
+        pf[mb] = lds_read16(&p_lds[wave][(mb * 16 + row_in_tile) * kStrip +
+                                         kk2 * 32 + kgroup * 8]);
+      }
+#pragma unroll
+      for (int t = 0; t < DTILES; ++t) {
+        frag_b16 vf = tr_bfrag<D>(v_cur, kk2 * 32, t, lane);
+#pragma unroll
+        for (int mb = 0; mb < MB; ++mb) {
+          o_acc[mb][t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              pf[mb], vf, o_acc[mb][t], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int mb = 0; mb < MB; ++mb) {
+    const int qrow0 = wrow0 + mb * 16;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = qrow0 + kgroup * 4 + r;
+      float inv_l = l_run[mb][r] > 0.f ? 1.0f / l_run[mb][r] : 0.f;
+      if (row < Sq) {
+#pragma unroll
+        for (int t = 0; t < DTILES; ++t) {
+          out[o_base + (long)row * os.ss + t * 16 + row_in_tile] =
+              __float2bfloat16(o_acc[mb][t][r] * inv_l);
+        }
+        if (row_in_tile == 0) {
+          lse[((long)b * Hq + h) * Sq + row] =
+              m_run[mb][r] + logf(fmaxf(l_run[mb][r], 1e-30f));
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // forward v8: 8-wave workgroup (2 waves/SIMD — the tuned CDNA4 attention
 // regime) with the T14 async-stage split: K/V tile loads are ISSUED one full
 // compute phase before their vmcnt wait + LDS write, so the HBM latency
@@ -1124,6 +1361,30 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
     // experiment (measured SLOWER: 172 vs 277 TF — kept for A/B)
     return e ? atoi(e) : 12;
   }();
+  if (nw_env == 82) {
+    // 8 waves x 2 bands (BM = 256)
+    dim3 grid((Sq + 255) / 256, Hq, B);
+    if (D == 128) {
+      hipLaunchKernelGGL((fa_fwd_mb2_kernel<128>), grid, dim3(512), 0, stream,
+                         (const __hip_bfloat16*)q.data_ptr(),
+                         (const __hip_bfloat16*)k.data_ptr(),
+                         (const __hip_bfloat16*)v.data_ptr(),
+                         (__hip_bfloat16*)out.data_ptr(),
+                         lse.data_ptr<float>(), B, Sq, Sk, Hq, Hkv,
+                         (float)softmax_scale, causal ? 1 : 0, win, qs, ks,
+                         vs, os);
+    } else {
+      hipLaunchKernelGGL((fa_fwd_mb2_kernel<64>), grid, dim3(512), 0, stream,
+                         (const __hip_bfloat16*)q.data_ptr(),
+                         (const __hip_bfloat16*)k.data_ptr(),
+                         (const __hip_bfloat16*)v.data_ptr(),
+                         (__hip_bfloat16*)out.data_ptr(),
+                         lse.data_ptr<float>(), B, Sq, Sk, Hq, Hkv,
+                         (float)softmax_scale, causal ? 1 : 0, win, qs, ks,
+                         vs, os);
+    }
+    return {out, lse};
+  }
   if (nw_env == 8) {
     dim3 grid((Sq + 127) / 128, Hq, B);
     if (D == 128) {
