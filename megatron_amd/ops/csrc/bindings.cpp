@@ -74,12 +74,16 @@ void build_blending_indices(torch::Tensor dataset_index,
 std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
                                           torch::Tensor v, bool causal,
                                           double softmax_scale,
-                                          int64_t window_size);
+                                          int64_t window_size,
+                                          double dropout_p, int64_t drop_seed,
+                                          int64_t drop_offset);
 std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
                                           torch::Tensor k, torch::Tensor v,
                                           torch::Tensor out, torch::Tensor lse,
                                           bool causal, double softmax_scale,
-                                          int64_t window_size);
+                                          int64_t window_size,
+                                          double dropout_p, int64_t drop_seed,
+                                          int64_t drop_offset);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
@@ -105,6 +109,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("build_sample_idx", &build_sample_idx);
   m.def("build_blending_indices", &build_blending_indices);
   m.def("fp8_quantize", &fp8_quantize);
-  m.def("flash_attn_fwd", &flash_attn_fwd);
-  m.def("flash_attn_bwd", &flash_attn_bwd);
+  m.def("flash_attn_fwd", &flash_attn_fwd, py::arg("q"),
+        py::arg("k"), py::arg("v"), py::arg("causal"),
+        py::arg("softmax_scale"), py::arg("window_size"),
+        py::arg("dropout_p") = 0.0, py::arg("drop_seed") = 0,
+        py::arg("drop_offset") = 0);
+  m.def("flash_attn_bwd", &flash_attn_bwd, py::arg("dout"),
+        py::arg("q"), py::arg("k"), py::arg("v"),
+        py::arg("out"), py::arg("lse"), py::arg("causal"),
+        py::arg("softmax_scale"), py::arg("window_size"),
+        py::arg("dropout_p") = 0.0, py::arg("drop_seed") = 0,
+        py::arg("drop_offset") = 0);
 }

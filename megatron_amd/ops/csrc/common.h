@@ -156,3 +156,17 @@ __device__ __forceinline__ float uint_to_uniform(unsigned x) {
   // (0,1] uniform
   return (x >> 8) * (1.0f / 16777216.0f);
 }
+
+// 64-bit-counter variant for index spaces past 2^32 (FA attention dropout:
+// counter n = element/4, ctr = (n_lo, n_hi, offset_lo, offset_hi), key =
+// seed). Mirrored bit-exactly by tests/philox_ref.py.
+__device__ __forceinline__ uint4 philox10_ctr64(unsigned long long seed,
+                                                unsigned long long offset,
+                                                unsigned long long n) {
+  uint2 key = make_uint2((unsigned)seed, (unsigned)(seed >> 32));
+  uint4 ctr = make_uint4((unsigned)n, (unsigned)(n >> 32), (unsigned)offset,
+                         (unsigned)(offset >> 32));
+#pragma unroll
+  for (int i = 0; i < 10; ++i) philox_round(ctr, key);
+  return ctr;
+}

@@ -77,6 +77,20 @@ __device__ __forceinline__ float group16_sum(float v) {
   return v;
 }
 
+// Attention-dropout keep decision for element (row=bh_row%..., col). The
+// (b,h,row,col) -> philox mapping is shared by forward and both backward
+// kernels (and mirrored in tests/philox_ref.py + the CPU reference), so the
+// backward regenerates the forward's mask exactly.
+__device__ __forceinline__ bool attn_drop_keep(unsigned long long seed,
+                                               unsigned long long offset,
+                                               long long bh_row, long long Skp,
+                                               int col, float p) {
+  unsigned long long e = (unsigned long long)bh_row * Skp + col;
+  uint4 r = philox10_ctr64(seed, offset, e >> 2);
+  unsigned v = (&r.x)[e & 3];
+  return uint_to_uniform(v) > p;
+}
+
 // --- padded tr-image -------------------------------------------------------
 
 // Bank-conflict-free tile-row geometry (derived analytically from the r02
@@ -189,12 +203,14 @@ __device__ __forceinline__ void stage_tr_image(
 // ---------------------------------------------------------------------------
 // forward
 
-template <int D, int NW>
+template <int D, int NW, bool DROP = false>
 __global__ __launch_bounds__(NW * 64) void fa_fwd_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ out,
     float* __restrict__ lse, int B, int Sq, int Sk, int Hq, int Hkv,
-    float scale, int causal, int window, TStr qs, TStr ks, TStr vs, TStr os) {
+    float scale, int causal, int window, TStr qs, TStr ks, TStr vs, TStr os,
+    float drop_p = 0.f, unsigned long long drop_seed = 0,
+    unsigned long long drop_offset = 0) {
   constexpr int KFRAGS = D / 32;
   constexpr int DTILES = D / 16;
   constexpr int IMG = tr_elems<D>(kBlockN);
@@ -356,7 +372,17 @@ __global__ __launch_bounds__(NW * 64) void fa_fwd_kernel(
       for (int r = 0; r < 4; ++r) {
         int row = kgroup * 4 + r;
         int col = t * 16 + row_in_tile;
-        p_lds[wave][row * kStrip + col] = __float2bfloat16(p_val[t][r]);
+        float pw = p_val[t][r];
+        if (DROP) {
+          // dropout applies to P feeding O (the lse/l_run softmax state
+          // stays dropout-independent)
+          bool keep = attn_drop_keep(
+              drop_seed, drop_offset,
+              ((long long)b * Hq + h) * Sq + (qrow0 + row),
+              ((long long)Sk + 3) & ~3LL, kstart + col, drop_p);
+          pw = keep ? pw * (1.0f / (1.0f - drop_p)) : 0.f;
+        }
+        p_lds[wave][row * kStrip + col] = __float2bfloat16(pw);
       }
     }
     // strips are wave-private: lgkmcnt ordering suffices, no barrier
@@ -935,7 +961,7 @@ __global__ void fa_bwd_delta_kernel(const __hip_bfloat16* __restrict__ dout,
 // backward dK/dV (k-parallel; flattened (gqa head, q-block) loop with
 // double-buffered dO/Q images)
 
-template <int D, int NW>
+template <int D, int NW, bool DROP = false>
 __global__ __launch_bounds__(NW * 64) void fa_bwd_dkdv_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v,
@@ -943,7 +969,8 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dkdv_kernel(
     const float* __restrict__ delta, __hip_bfloat16* __restrict__ dk,
     __hip_bfloat16* __restrict__ dv, int B, int Sq, int Sk, int Hq, int Hkv,
     float scale, int causal, int window, TStr qs, TStr ks, TStr vs, TStr ds,
-    TStr dks, TStr dvs) {
+    TStr dks, TStr dvs, float drop_p = 0.f, unsigned long long drop_seed = 0,
+    unsigned long long drop_offset = 0) {
   constexpr int KFRAGS = D / 32;
   constexpr int DTILES = D / 16;
   constexpr int IMG = tr_elems<D>(kBlockM);
@@ -1077,7 +1104,19 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dkdv_kernel(
         if (!masked) {
           float l = lse_h[qrow];
           pt = __expf(stt[r] * scale - l);
-          dst = pt * (dpt[r] - delta_h[qrow]) * scale;
+          float dp_eff = dpt[r];
+          if (DROP) {
+            bool keep = attn_drop_keep(
+                drop_seed, drop_offset,
+                ((long long)b * Hq + hq) * Sq + qrow,
+                ((long long)Sk + 3) & ~3LL, key, drop_p);
+            float rs = 1.0f / (1.0f - drop_p);
+            dp_eff = keep ? dp_eff * rs : 0.f;
+            dst = pt * (dp_eff - delta_h[qrow]) * scale;
+            pt = keep ? pt * rs : 0.f;  // dV uses the dropped P
+          } else {
+            dst = pt * (dp_eff - delta_h[qrow]) * scale;
+          }
         }
         int lrow = kgroup * 4 + r;
         int lcol = t * 16 + row_in_tile;
@@ -1129,14 +1168,15 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dkdv_kernel(
 // ---------------------------------------------------------------------------
 // backward dQ (q-parallel)
 
-template <int D, int NW>
+template <int D, int NW, bool DROP = false>
 __global__ __launch_bounds__(NW * 64) void fa_bwd_dq_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v,
     const __hip_bfloat16* __restrict__ dout, const float* __restrict__ lse,
     const float* __restrict__ delta, __hip_bfloat16* __restrict__ dq, int B,
     int Sq, int Sk, int Hq, int Hkv, float scale, int causal, int window,
-    TStr qs, TStr ks, TStr vs, TStr ds, TStr dqs) {
+    TStr qs, TStr ks, TStr vs, TStr ds, TStr dqs, float drop_p = 0.f,
+    unsigned long long drop_seed = 0, unsigned long long drop_offset = 0) {
   constexpr int KFRAGS = D / 32;
   constexpr int DTILES = D / 16;
   constexpr int IMG = tr_elems<D>(kBlockN);
@@ -1263,7 +1303,15 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dq_kernel(
         float ds = 0.f;
         if (!masked) {
           float p = __expf(st[r] * scale - lse_r[r]);
-          ds = p * (dp[r] - delta_r[r]) * scale;
+          float dp_eff = dp[r];
+          if (DROP) {
+            bool keep = attn_drop_keep(
+                drop_seed, drop_offset,
+                ((long long)b * Hq + h) * Sq + row,
+                ((long long)Sk + 3) & ~3LL, col, drop_p);
+            dp_eff = keep ? dp_eff * (1.0f / (1.0f - drop_p)) : 0.f;
+          }
+          ds = p * (dp_eff - delta_r[r]) * scale;
         }
         int lrow = kgroup * 4 + r;
         int lcol = t * 16 + row_in_tile;
@@ -1339,7 +1387,9 @@ static torch::Tensor empty_like_strided(const torch::Tensor& ref) {
 std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
                                           torch::Tensor v, bool causal,
                                           double softmax_scale,
-                                          int64_t window_size) {
+                                          int64_t window_size,
+                                          double dropout_p, int64_t drop_seed,
+                                          int64_t drop_offset) {
   check_bshd(q, "q");
   check_bshd(k, "k");
   check_bshd(v, "v");
@@ -1424,7 +1474,9 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
                                           torch::Tensor k, torch::Tensor v,
                                           torch::Tensor out, torch::Tensor lse,
                                           bool causal, double softmax_scale,
-                                          int64_t window_size) {
+                                          int64_t window_size,
+                                          double dropout_p, int64_t drop_seed,
+                                          int64_t drop_offset) {
   check_bshd(dout, "dout");
   check_bshd(q, "q");
   check_bshd(k, "k");
@@ -1443,14 +1495,18 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
              os = tstr(out), dqs = tstr(dq), dks = tstr(dk), dvs = tstr(dv);
 
   long rows = (long)B * Hq * Sq;
-#define LAUNCH_BWD(DD)                                                        \
+  const float dp = (float)dropout_p;
+  const unsigned long long dseed = (unsigned long long)drop_seed;
+  const unsigned long long doff = (unsigned long long)drop_offset;
+#define LAUNCH_BWD(DD, DROP)                                                  \
   do {                                                                        \
     hipLaunchKernelGGL((fa_bwd_delta_kernel<DD>), dim3(rows), dim3(64), 0,    \
                        stream, (const __hip_bfloat16*)dout.data_ptr(),        \
                        (const __hip_bfloat16*)out.data_ptr(),                 \
                        delta.data_ptr<float>(), B, Sq, Hq, ds, os);           \
     dim3 gridk((Sk + 12 * 16 - 1) / (12 * 16), Hkv, B);                       \
-    hipLaunchKernelGGL((fa_bwd_dkdv_kernel<DD, 12>), gridk, dim3(12 * 64), 0, \
+    hipLaunchKernelGGL((fa_bwd_dkdv_kernel<DD, 12, DROP>), gridk,             \
+                       dim3(12 * 64), 0,                                      \
                        stream, (const __hip_bfloat16*)q.data_ptr(),           \
                        (const __hip_bfloat16*)k.data_ptr(),                   \
                        (const __hip_bfloat16*)v.data_ptr(),                   \
@@ -1459,9 +1515,10 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
                        (__hip_bfloat16*)dk.data_ptr(),                        \
                        (__hip_bfloat16*)dv.data_ptr(), B, Sq, Sk, Hq, Hkv,    \
                        (float)softmax_scale, causal ? 1 : 0, win, qs, ks, vs, \
-                       ds, dks, dvs);                                         \
+                       ds, dks, dvs, dp, dseed, doff);                        \
     dim3 gridq((Sq + 12 * 16 - 1) / (12 * 16), Hq, B);                        \
-    hipLaunchKernelGGL((fa_bwd_dq_kernel<DD, 12>), gridq, dim3(12 * 64), 0,   \
+    hipLaunchKernelGGL((fa_bwd_dq_kernel<DD, 12, DROP>), gridq,               \
+                       dim3(12 * 64), 0,                                      \
                        stream, (const __hip_bfloat16*)q.data_ptr(),           \
                        (const __hip_bfloat16*)k.data_ptr(),                   \
                        (const __hip_bfloat16*)v.data_ptr(),                   \
@@ -1469,14 +1526,17 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
                        lse.data_ptr<float>(), delta.data_ptr<float>(),        \
                        (__hip_bfloat16*)dq.data_ptr(), B, Sq, Sk, Hq, Hkv,    \
                        (float)softmax_scale, causal ? 1 : 0, win, qs, ks, vs, \
-                       ds, dqs);                                              \
+                       ds, dqs, dp, dseed, doff);                             \
   } while (0)
 
+  const bool drop = dropout_p > 0.0;
   if (D == 128) {
-    LAUNCH_BWD(128);
+    if (drop) LAUNCH_BWD(128, true);
+    else LAUNCH_BWD(128, false);
   } else {
     TORCH_CHECK(D == 64);
-    LAUNCH_BWD(64);
+    if (drop) LAUNCH_BWD(64, true);
+    else LAUNCH_BWD(64, false);
   }
   return {dq, dk, dv};
 }

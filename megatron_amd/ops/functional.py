@@ -443,19 +443,32 @@ class FlashAttnFunction(torch.autograd.Function):
     def forward(ctx, q, k, v, causal, softmax_scale, window_size, dropout_p,
                 training):
         ext = _C(q)
-        if training and dropout_p and dropout_p > 0.0:
-            # fail loudly: the CDNA4 FA kernels do not implement attention
-            # dropout (use attention_dropout 0, or the CoreAttention path)
-            raise NotImplementedError(
-                "flash attention with attention_dropout > 0 is not "
-                "supported; set --attention_dropout 0 or disable "
-                "--use_flash_attn"
-            )
+        use_drop = bool(training and dropout_p and dropout_p > 0.0)
         if softmax_scale is None:
             softmax_scale = 1.0 / math.sqrt(q.shape[-1])
         ctx.causal = causal
         ctx.softmax_scale = softmax_scale
         ctx.window_size = window_size
+        # philox-in-kernel attention dropout on the P strip, regenerated
+        # bit-exactly in backward (reference transformer.py:544 dropout via
+        # flash-attn)
+        ctx.dropout_p = float(dropout_p) if use_drop else 0.0
+        ctx.drop_seed = 0
+        ctx.drop_offset = 0
+        if use_drop:
+            if q.is_cuda:
+                gen = torch.cuda.default_generators[q.device.index]
+                ctx.drop_seed = gen.initial_seed()
+                try:
+                    ctx.drop_offset = gen.get_offset()
+                    gen.set_offset(ctx.drop_offset + 4)
+                except Exception:
+                    ctx.drop_offset = int(
+                        torch.randint(0, 2**31, (1,)).item()
+                    )
+            else:
+                ctx.drop_seed = torch.initial_seed() & (2**63 - 1)
+                ctx.drop_offset = int(torch.randint(0, 2**31, (1,)).item())
         if ext is not None:
             # kernels are stride-aware over batch/seq/head (head_dim must be
             # dense, strides 16B-aligned): sbhd transposes and QKV-projection
@@ -472,10 +485,15 @@ class FlashAttnFunction(torch.autograd.Function):
                 q, k, v,
                 bool(causal), float(softmax_scale),
                 int(window_size) if window_size is not None else -1,
+                ctx.dropout_p, ctx.drop_seed, ctx.drop_offset,
             )
             ctx.save_for_backward(q, k, v, out, lse)
             return out
-        out, lse = _sdpa_reference(q, k, v, causal, softmax_scale, window_size)
+        out, lse = _sdpa_reference(
+            q, k, v, causal, softmax_scale, window_size,
+            dropout_p=ctx.dropout_p, drop_seed=ctx.drop_seed,
+            drop_offset=ctx.drop_offset,
+        )
         ctx.save_for_backward(q, k, v, out, lse)
         return out
 
@@ -491,10 +509,13 @@ class FlashAttnFunction(torch.autograd.Function):
                 dout, q, k, v, out, lse,
                 bool(ctx.causal), float(ctx.softmax_scale),
                 int(ctx.window_size) if ctx.window_size is not None else -1,
+                ctx.dropout_p, ctx.drop_seed, ctx.drop_offset,
             )
             return dq, dk, dv, None, None, None, None, None
         dq, dk, dv = _sdpa_reference_bwd(
-            dout, q, k, v, out, lse, ctx.causal, ctx.softmax_scale, ctx.window_size
+            dout, q, k, v, out, lse, ctx.causal, ctx.softmax_scale,
+            ctx.window_size, dropout_p=ctx.dropout_p,
+            drop_seed=ctx.drop_seed, drop_offset=ctx.drop_offset,
         )
         return dq, dk, dv, None, None, None, None, None
 
@@ -516,7 +537,26 @@ def _attn_mask(sq, sk, causal, window_size, device):
     return m
 
 
-def _sdpa_reference(q, k, v, causal, scale, window_size):
+def _drop_keep_mask(bsz, heads, sq, sk, dropout_p, drop_seed, drop_offset,
+                    device):
+    """[b, n, sq, sk] bool keep-mask, bit-exact vs the HIP kernels."""
+    import numpy as np
+
+    from .philox_ref import attn_dropout_mask
+
+    m = np.stack([
+        np.stack([
+            attn_dropout_mask(drop_seed, drop_offset, b, h, heads, sq, sk,
+                              dropout_p)
+            for h in range(heads)
+        ])
+        for b in range(bsz)
+    ])
+    return torch.from_numpy(m).to(device)
+
+
+def _sdpa_reference(q, k, v, causal, scale, window_size, dropout_p=0.0,
+                    drop_seed=0, drop_offset=0):
     b, sq, n, h = q.shape
     n_kv = k.shape[2]
     rep = n // n_kv
@@ -528,11 +568,16 @@ def _sdpa_reference(q, k, v, causal, scale, window_size):
     scores = scores.masked_fill(mask, float("-inf"))
     lse = torch.logsumexp(scores, dim=-1)  # b n s
     p = torch.softmax(scores, dim=-1)
+    if dropout_p and dropout_p > 0.0:
+        keep = _drop_keep_mask(b, n, sq, k.shape[1], dropout_p, drop_seed,
+                               drop_offset, q.device)
+        p = p * keep.to(p.dtype) / (1.0 - dropout_p)
     of = torch.matmul(p, vf)
     return of.permute(0, 2, 1, 3).to(q.dtype), lse
 
 
-def _sdpa_reference_bwd(dout, q, k, v, out, lse, causal, scale, window_size):
+def _sdpa_reference_bwd(dout, q, k, v, out, lse, causal, scale, window_size,
+                        dropout_p=0.0, drop_seed=0, drop_offset=0):
     b, sq, n, h = q.shape
     n_kv = k.shape[2]
     rep = n // n_kv
@@ -546,6 +591,10 @@ def _sdpa_reference_bwd(dout, q, k, v, out, lse, causal, scale, window_size):
         mask = _attn_mask(sq, k.shape[1], causal, window_size, q.device)
         scores = scores.masked_fill(mask, float("-inf"))
         p = torch.softmax(scores, dim=-1)
+        if dropout_p and dropout_p > 0.0:
+            keep = _drop_keep_mask(b, n, sq, k.shape[1], dropout_p,
+                                   drop_seed, drop_offset, q.device)
+            p = p * keep.to(p.dtype) / (1.0 - dropout_p)
         of = torch.matmul(p, vr)
     df = dout.float().permute(0, 2, 1, 3)
     dq, dk, dv = torch.autograd.grad(of, (qf, kf, vf), grad_outputs=df)

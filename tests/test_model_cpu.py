@@ -264,12 +264,15 @@ def test_param_group_conds():
     assert not groups[1]["params"]
 
 
-def test_flash_dropout_raises():
+def test_flash_dropout_supported():
+    # dropout through flash attention is implemented (philox in-kernel on
+    # GPU, same-mask numpy reference on CPU) — see test_flash_attention_
+    # dropout_cpu_path / tests/test_ops_gpu.py::test_dropout_parity
     from megatron_amd.ops.functional import flash_attention
 
     q = torch.randn(1, 8, 2, 16)
-    with pytest.raises(NotImplementedError):
-        flash_attention(q, q, q, dropout_p=0.1, training=True)
+    out = flash_attention(q, q, q, dropout_p=0.1, training=True)
+    assert out.shape == q.shape
 
 
 def test_chunked_lm_loss_matches_unchunked():
@@ -398,3 +401,40 @@ def test_norm_res_matches_fanin():
         assert torch.allclose(w.grad, gw, atol=1e-5)
         if nargs == 2:
             assert torch.allclose(b.grad, gb, atol=1e-5)
+
+
+def test_attn_dropout_philox_ref():
+    """numpy philox mask: deterministic, keep-rate ~ 1-p, offset-sensitive."""
+    import numpy as np
+    from megatron_amd.ops.philox_ref import attn_dropout_mask
+
+    m1 = attn_dropout_mask(1234, 0, 0, 1, 4, 64, 64, 0.25)
+    m2 = attn_dropout_mask(1234, 0, 0, 1, 4, 64, 64, 0.25)
+    m3 = attn_dropout_mask(1234, 4, 0, 1, 4, 64, 64, 0.25)
+    assert (m1 == m2).all()
+    assert not (m1 == m3).all()
+    rate = m1.mean()
+    assert 0.70 < rate < 0.80, rate
+
+
+def test_flash_attention_dropout_cpu_path():
+    """FA autograd Function with dropout (CPU reference path): runs fwd+bwd,
+    output expectation is preserved (~unbiased), p=0 path unchanged."""
+    import torch
+    from megatron_amd.ops.functional import flash_attention
+
+    torch.manual_seed(7)
+    q = torch.randn(2, 32, 2, 64, requires_grad=True)
+    k = torch.randn(2, 32, 2, 64, requires_grad=True)
+    v = torch.randn(2, 32, 2, 64, requires_grad=True)
+
+    out0 = flash_attention(q, k, v, causal=True, dropout_p=0.5,
+                           training=False)  # eval: no dropout
+    ref = flash_attention(q, k, v, causal=True, dropout_p=0.0, training=True)
+    assert torch.allclose(out0, ref)
+
+    out = flash_attention(q, k, v, causal=True, dropout_p=0.3, training=True)
+    assert not torch.allclose(out, ref)
+    out.float().mean().backward()
+    assert q.grad is not None and torch.isfinite(q.grad).all()
+    assert torch.isfinite(k.grad).all() and torch.isfinite(v.grad).all()

@@ -290,6 +290,51 @@ class TestFlashAttention:
         assert rel_err(dk, rdk) < 8e-2
         assert rel_err(dq, rdq) < 8e-2
 
+    @pytest.mark.parametrize("p", [0.1, 0.5])
+    def test_dropout_parity(self, p):
+        """Philox attention dropout: HIP kernels vs fp32 reference using the
+        SAME bit-exact mask (megatron_amd/ops/philox_ref.py) — forward out
+        and all three backward grads."""
+        from megatron_amd.ops.functional import (
+            _sdpa_reference, _sdpa_reference_bwd,
+        )
+
+        ext = _ext()
+        b, s, n, d = 2, 192, 4, 128
+        seed, off = 1234, 16
+        q = torch.randn(b, s, n, d, device="cuda", dtype=torch.bfloat16)
+        k = torch.randn(b, s, n, d, device="cuda", dtype=torch.bfloat16)
+        v = torch.randn(b, s, n, d, device="cuda", dtype=torch.bfloat16)
+        scale = 1.0 / math.sqrt(d)
+        out, lse = ext.flash_attn_fwd(q, k, v, True, scale, -1, p, seed, off)
+        ref, ref_lse = _sdpa_reference(q, k, v, True, scale, None,
+                                       dropout_p=p, drop_seed=seed,
+                                       drop_offset=off)
+        assert rel_err(out, ref) < 8e-2
+        # lse is dropout-independent
+        assert (lse - ref_lse).abs().max().item() < 1e-2
+
+        dout = torch.randn_like(out)
+        dq, dk, dv = ext.flash_attn_bwd(dout, q, k, v, out, lse, True, scale,
+                                        -1, p, seed, off)
+        rdq, rdk, rdv = _sdpa_reference_bwd(dout, q, k, v, out, lse, True,
+                                            scale, None, dropout_p=p,
+                                            drop_seed=seed, drop_offset=off)
+        assert rel_err(dv, rdv) < 1e-1
+        assert rel_err(dk, rdk) < 1e-1
+        assert rel_err(dq, rdq) < 1e-1
+
+    def test_dropout_determinism_and_seed_sensitivity(self):
+        ext = _ext()
+        b, s, n, d = 1, 128, 2, 64
+        q = torch.randn(b, s, n, d, device="cuda", dtype=torch.bfloat16)
+        scale = 1.0 / math.sqrt(d)
+        o1, _ = ext.flash_attn_fwd(q, q, q, True, scale, -1, 0.3, 7, 0)
+        o2, _ = ext.flash_attn_fwd(q, q, q, True, scale, -1, 0.3, 7, 0)
+        o3, _ = ext.flash_attn_fwd(q, q, q, True, scale, -1, 0.3, 7, 4)
+        assert torch.equal(o1, o2)
+        assert not torch.equal(o1, o3)
+
     def test_sbhd_strided_matches_contiguous(self):
         """sbhd-transposed views (the model's native layout, no transpose
         copies) must produce bitwise-identical results to bshd-contiguous
