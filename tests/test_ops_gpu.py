@@ -324,6 +324,29 @@ class TestFlashAttention:
         assert rel_err(dk, rdk) < 1e-1
         assert rel_err(dq, rdq) < 1e-1
 
+    def test_dropout_mask_bit_exact(self):
+        """Extract the kernel's keep-mask directly (q=0 -> uniform P;
+        V=identity -> out[i,d] = keep(i,d)/((i+1)(1-p))) and compare it
+        BOOLEAN-exactly with the numpy philox mirror."""
+        import numpy as np
+
+        from megatron_amd.ops.philox_ref import attn_dropout_mask
+
+        ext = _ext()
+        b, s, n, d = 1, 128, 2, 128
+        p, seed, off = 0.4, 99, 8
+        q = torch.zeros(b, s, n, d, device="cuda", dtype=torch.bfloat16)
+        k = torch.zeros(b, s, n, d, device="cuda", dtype=torch.bfloat16)
+        v = torch.eye(d, device="cuda", dtype=torch.bfloat16).view(
+            1, s, 1, d).expand(b, s, n, d).contiguous()
+        out, _ = ext.flash_attn_fwd(q, k, v, True, 1.0, -1, p, seed, off)
+        got = (out.float() > 0).cpu().numpy()  # [b,s,n,d] == keep[row, col<d]
+        for h in range(n):
+            ref = attn_dropout_mask(seed, off, 0, h, n, s, s, p)
+            for i in range(s):
+                cols = min(i + 1, d)
+                assert (got[0, i, h, :cols] == ref[i, :cols]).all(), (h, i)
+
     def test_dropout_determinism_and_seed_sensitivity(self):
         ext = _ext()
         b, s, n, d = 1, 128, 2, 64

@@ -379,3 +379,73 @@ def test_finetune_instruction_loss_mask(dist_single):
     # 6 predicts the pad token -> 0
     expect = torch.tensor([[0.25, 0.25, 0.25, 1.0, 1.0, 1.0, 0.0]])
     assert torch.allclose(loss_mask, expect), loss_mask
+
+
+def test_stream_reshard_bounded_and_correct(tmp_path):
+    """Streaming reshard (tp1,pp2 -> tp2,pp1): results equal the full-merge
+    reference, and the shard cache holds only ONE source PP stage at a time
+    (the bounded-memory property replacing the reference's loader/saver
+    subprocess streaming, reference tools/checkpoint_util.py:13-86)."""
+    import argparse as ap
+    import sys
+    sys.path.insert(0, os.path.join(REPO, "tools"))
+    import checkpoint_util as cu
+
+    torch.manual_seed(5)
+    H, V = 32, 64
+    margs = ap.Namespace(
+        num_layers=4, hidden_size=H, num_attention_heads=4,
+        num_attention_heads_kv=2, ffn_hidden_size=3 * H,
+        padded_vocab_size=V, tensor_model_parallel_size=1,
+        pipeline_model_parallel_size=2, glu_activation="swiglu",
+    )
+
+    def layer_sd(prefix):
+        return {
+            f"{prefix}.input_layernorm.weight": torch.randn(H),
+            f"{prefix}.self_attention.query_key_value.weight":
+                torch.randn(2 * H, H),
+            f"{prefix}.self_attention.dense.weight": torch.randn(H, H),
+            f"{prefix}.post_attention_layernorm.weight": torch.randn(H),
+            f"{prefix}.mlp.dense_h_to_4h.weight": torch.randn(6 * H, H),
+            f"{prefix}.mlp.dense_4h_to_h.weight": torch.randn(H, 3 * H),
+        }
+
+    stage0 = {"embedding.word_embeddings.weight": torch.randn(V, H)}
+    for i in range(2):
+        stage0.update(layer_sd(f"encoder.layers.{i}"))
+    stage1 = {}
+    for i in range(2):
+        stage1.update(layer_sd(f"encoder.layers.{i}"))
+    stage1["encoder.final_layernorm.weight"] = torch.randn(H)
+    stage1["lm_head"] = torch.randn(V, H)
+
+    src = tmp_path / "src"
+    for pp, sd in [(0, stage0), (1, stage1)]:
+        d = src / "release" / f"mp_rank_00_{pp:03d}"
+        d.mkdir(parents=True)
+        torch.save({"args": margs, "checkpoint_version": 3.0, "iteration": 0,
+                    "model": {"language_model": sd}},
+                   d / "model_optim_rng.pt")
+    (src / "latest_checkpointed_iteration.txt").write_text("release")
+
+    out = tmp_path / "out"
+    cache = cu.stream_reshard(str(src), str(out), tp=2, pp=1, glu=True,
+                              progress=lambda *_: None)
+    assert cache.max_loaded == 1  # one tp shard per stage, one stage live
+
+    # reference: full merge then split
+    shards = {(0, 0): {"model": {"language_model": stage0}},
+              (0, 1): {"model": {"language_model": stage1}}}
+    full = cu.merge_full_state(shards, 1, 2, 4, glu=True)
+    ref_split = cu.split_full_state(full, 2, 1, 4, glu=True)
+
+    for tpr in range(2):
+        got = torch.load(
+            out / "release" / f"mp_rank_{tpr:02d}" / "model_optim_rng.pt",
+            map_location="cpu", weights_only=False,
+        )["model"]["language_model"]
+        ref = ref_split[(tpr, 0)]
+        assert set(got.keys()) == set(ref.keys())
+        for k in ref:
+            assert torch.equal(got[k], ref[k]), k

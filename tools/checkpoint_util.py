@@ -178,6 +178,184 @@ def split_full_state(full, tp, pp, num_layers, glu):
     return out
 
 
+# ---------------------------------------------------------------------------
+# Streaming (bounded-memory) reshard. The reference bounds memory with a
+# loader subprocess + mp.Queue streaming one named tensor group at a time
+# (tools/checkpoint_util.py:13-86); here a single process achieves the same
+# bound: source shard files are memory-mapped (clean, OS-evictable pages)
+# and opened one source-PP stage at a time, and each TARGET rank's state is
+# assembled and written before the next target starts — peak anonymous RSS
+# is ~one target shard (model/(tp*pp)) + transient merged tensors, never the
+# full unsharded model.
+
+
+class _ShardCache:
+    """mmap-backed loader of source shard payloads, one PP stage resident at
+    a time. `max_loaded` is exposed for the bounded-memory test."""
+
+    def __init__(self, paths, tp_size):
+        self.paths = paths
+        self.tp_size = tp_size
+        self._stage = None
+        self._stage_models = None
+        self.max_loaded = 0
+        self.load_calls = 0
+
+    def _load(self, path):
+        self.load_calls += 1
+        try:
+            return torch.load(path, map_location="cpu", weights_only=False,
+                              mmap=True)
+        except Exception:
+            return torch.load(path, map_location="cpu", weights_only=False)
+
+    def meta(self):
+        """args/iteration payload from shard (0,0) — loaded and dropped."""
+        payload = self._load(self.paths[(0, 0)])
+        meta = {"args": payload.get("args"),
+                "iteration": payload.get("iteration", 0)}
+        del payload
+        return meta
+
+    def stage_models(self, pp):
+        if self._stage != pp:
+            self._stage_models = None
+            models = []
+            for tp in range(self.tp_size):
+                payload = self._load(self.paths[(tp, pp)])
+                m = payload["model"]
+                if "language_model" in m:
+                    m = m["language_model"]
+                models.append(m)
+            self._stage = pp
+            self._stage_models = models
+            self.max_loaded = max(self.max_loaded, len(models))
+        return self._stage_models
+
+
+def _shard_paths(load_dir):
+    tracker = get_checkpoint_tracker_filename(load_dir)
+    with open(tracker) as f:
+        meta = f.read().strip()
+    release = meta == "release"
+    iteration = 0 if release else int(meta)
+    sub = "release" if release else f"iter_{iteration:07d}"
+    base = os.path.join(load_dir, sub)
+    paths = {}
+    for d in sorted(os.listdir(base)):
+        if not d.startswith("mp_rank_"):
+            continue
+        parts = d.split("_")
+        tp = int(parts[2])
+        pp = int(parts[3]) if len(parts) > 3 else 0
+        path = os.path.join(base, d, "model_optim_rng.pt")
+        if not os.path.exists(path):
+            path = os.path.join(base, d, "model_rng.pt")
+        paths[(tp, pp)] = path
+    tp_size = max(k[0] for k in paths) + 1
+    pp_size = max(k[1] for k in paths) + 1
+    return paths, tp_size, pp_size, iteration, release
+
+
+def _merge_key(models, key, glu):
+    kind = _merge_kind(key)
+    tensors = [m[key] for m in models]
+    if kind == 0:
+        return _glu_aware_cat(key, tensors, None, glu)
+    if kind == 1:
+        return torch.cat(tensors, dim=1)
+    return tensors[0]
+
+
+def stream_reshard(load_dir, save_dir, tp, pp, glu, progress=print):
+    """Bounded-memory reshard loop: one target rank's dict in flight."""
+    paths, tp_src, pp_src, iteration, release = _shard_paths(load_dir)
+    cache = _ShardCache(paths, tp_src)
+    meta = cache.meta()
+    margs = meta["args"]
+    num_layers = getattr(margs, "num_layers")
+    assert num_layers % pp == 0
+    layers_per_tgt = num_layers // pp
+
+    # source stage layer spans (count local layers per source stage lazily)
+    src_spans = []
+    offset = 0
+    for sp in range(pp_src):
+        models = cache.stage_models(sp)
+        n_local = len({k.split(".")[2] for k in models[0].keys()
+                       if k.startswith("encoder.layers.")})
+        src_spans.append((offset, offset + n_local))
+        offset += n_local
+    assert offset == num_layers, (offset, num_layers)
+
+    sub = "release" if release else f"iter_{iteration:07d}"
+    if margs is not None:
+        margs.tensor_model_parallel_size = tp
+        margs.pipeline_model_parallel_size = pp
+
+    def split_piece(key, merged, tpr):
+        kind = _merge_kind(key)
+        if kind == 0:
+            return _glu_aware_split(key, merged, tp, glu)[tpr]
+        if kind == 1:
+            return torch.chunk(merged, tp, dim=1)[tpr]
+        return merged
+
+    # iterate targets outer pp (sequential source stages), inner tp
+    for ppr in range(pp):
+        lo, hi = ppr * layers_per_tgt, (ppr + 1) * layers_per_tgt
+        for tpr in range(tp):
+            sd = {}
+            # non-layer keys (fetch the source stage only when this target
+            # actually needs it — the cache holds ONE stage at a time)
+            if ppr == 0:
+                first_models = cache.stage_models(0)
+                for key in list(first_models[0].keys()):
+                    if key.startswith("embedding."):
+                        sd[key] = split_piece(key, _merge_key(first_models,
+                                                              key, glu), tpr)
+            if ppr == pp - 1:
+                last_models = cache.stage_models(pp_src - 1)
+                for key in list(last_models[0].keys()):
+                    if (key.startswith("encoder.final_layernorm")
+                            or key == "lm_head"):
+                        sd[key] = split_piece(key, _merge_key(last_models,
+                                                              key, glu), tpr)
+            # layers in [lo, hi) from their source stages
+            for sp, (slo, shi) in enumerate(src_spans):
+                if shi <= lo or slo >= hi:
+                    continue
+                models = cache.stage_models(sp)
+                for key in list(models[0].keys()):
+                    if not key.startswith("encoder.layers."):
+                        continue
+                    parts = key.split(".")
+                    gidx = int(parts[2]) + slo
+                    if not (lo <= gidx < hi):
+                        continue
+                    parts[2] = str(gidx - lo)
+                    local_key = ".".join(parts)
+                    sd[local_key] = split_piece(key, _merge_key(models, key,
+                                                                glu), tpr)
+            if pp == 1:
+                d = os.path.join(save_dir, sub, f"mp_rank_{tpr:02d}")
+            else:
+                d = os.path.join(save_dir, sub, f"mp_rank_{tpr:02d}_{ppr:03d}")
+            os.makedirs(d, exist_ok=True)
+            state = {
+                "args": margs,
+                "checkpoint_version": 3.0,
+                "iteration": iteration,
+                "model": {"language_model": sd},
+            }
+            torch.save(state, os.path.join(d, "model_optim_rng.pt"))
+            progress(f"  wrote tp{tpr} pp{ppr} ({len(sd)} tensors)")
+            del sd, state
+    with open(get_checkpoint_tracker_filename(save_dir), "w") as f:
+        f.write("release" if release else str(iteration))
+    return cache
+
+
 def main():
     parser = argparse.ArgumentParser()
     parser.add_argument("--model_type", default="llama2",
@@ -190,46 +368,21 @@ def main():
                         default=1)
     args = parser.parse_args()
 
-    shards, tp_size, pp_size, iteration, release = _load_shards(args.load_dir)
-    first = shards[(0, 0)]
-    margs = first.get("args")
-    num_layers = getattr(margs, "num_layers")
-    glu = getattr(margs, "glu_activation", None) is not None or (
-        args.model_type in ("llama", "llama2", "codellama", "mistral")
-    )
+    paths, tp_size, pp_size, _, _ = _shard_paths(args.load_dir)
+    glu = args.model_type in ("llama", "llama2", "codellama", "mistral")
+    # honor the saved args' glu flag when present
+    probe = torch.load(paths[(0, 0)], map_location="cpu", weights_only=False)
+    margs = probe.get("args")
+    if margs is not None and getattr(margs, "glu_activation", None):
+        glu = True
+    del probe
 
-    print(f"merging from tp={tp_size} pp={pp_size} "
+    print(f"streaming reshard from tp={tp_size} pp={pp_size} "
           f"-> tp={args.target_tensor_parallel_size} "
           f"pp={args.target_pipeline_parallel_size}")
-    full = merge_full_state(shards, tp_size, pp_size, num_layers, glu)
-    split = split_full_state(
-        full, args.target_tensor_parallel_size,
-        args.target_pipeline_parallel_size, num_layers, glu,
-    )
-
-    # write target checkpoint
-    sub = "release" if release else f"iter_{iteration:07d}"
-    tp = args.target_tensor_parallel_size
-    pp = args.target_pipeline_parallel_size
-    if margs is not None:
-        margs.tensor_model_parallel_size = tp
-        margs.pipeline_model_parallel_size = pp
-    for (tpr, ppr), sd in split.items():
-        if pp == 1:
-            d = os.path.join(args.save_dir, sub, f"mp_rank_{tpr:02d}")
-        else:
-            d = os.path.join(args.save_dir, sub,
-                             f"mp_rank_{tpr:02d}_{ppr:03d}")
-        os.makedirs(d, exist_ok=True)
-        state = {
-            "args": margs,
-            "checkpoint_version": 3.0,
-            "iteration": iteration,
-            "model": {"language_model": sd},
-        }
-        torch.save(state, os.path.join(d, "model_optim_rng.pt"))
-    with open(get_checkpoint_tracker_filename(args.save_dir), "w") as f:
-        f.write("release" if release else str(iteration))
+    stream_reshard(args.load_dir, args.save_dir,
+                   args.target_tensor_parallel_size,
+                   args.target_pipeline_parallel_size, glu)
     print(f"saved resharded checkpoint to {args.save_dir}")
 
 
