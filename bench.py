@@ -68,12 +68,13 @@ MODEL_SPECS = {
                        num_attention_heads=64, num_attention_heads_kv=8,
                        vocab=32000, seq=4096, model_name="llama2"),
     # Mistral-7B: 32 layers, h=4096, ffn=14336, 32 heads, 8 kv, SWA 4096.
-    # seq-32k activations cap the micro-batch at 1 on 288 GB
+    # chunked LM-head loss (loss_chunk_size) bounds the fp32 logits peak at
+    # seq 32k, reopening micro-batch 2 on 288 GB (round-1 ran mbs1)
     "mistral-7b": dict(num_layers=32, hidden_size=4096, ffn_hidden_size=14336,
                        num_attention_heads=32, num_attention_heads_kv=8,
                        vocab=32000, seq=32768, model_name="mistral",
                        sliding_window_size=4096, rope_scaling_factor=4.0,
-                       mbs=1),
+                       mbs=2, loss_chunk_size=8192),
     # Falcon-7B: 32 layers, h=4544, 71 heads, MQA (1 kv head), parallel attn
     "falcon-7b": dict(num_layers=32, hidden_size=4544, ffn_hidden_size=18176,
                       num_attention_heads=71, num_attention_heads_kv=1,
@@ -160,6 +161,7 @@ def main():
         model_name=spec["model_name"],
         sliding_window_size=spec.get("sliding_window_size"),
         rope_scaling_factor=spec.get("rope_scaling_factor", 1.0),
+        loss_chunk_size=spec.get("loss_chunk_size", 0),
         rank=rank, world_size=world_size, local_rank=local_rank,
         **dtype_flags,
     )
