@@ -74,7 +74,7 @@ MODEL_SPECS = {
                        num_attention_heads=32, num_attention_heads_kv=8,
                        vocab=32000, seq=32768, model_name="mistral",
                        sliding_window_size=4096, rope_scaling_factor=4.0,
-                       mbs=2, loss_chunk_size=8192),
+                       mbs=1, loss_chunk_size=8192),
     # Falcon-7B: 32 layers, h=4544, 71 heads, MQA (1 kv head), parallel attn
     "falcon-7b": dict(num_layers=32, hidden_size=4544, ffn_hidden_size=18176,
                       num_attention_heads=71, num_attention_heads_kv=1,

@@ -1411,6 +1411,34 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
     // experiment (measured SLOWER: 172 vs 277 TF — kept for A/B)
     return e ? atoi(e) : 12;
   }();
+  if (dropout_p > 0.0) {
+    // attention dropout: 12-wave DROP instantiation only (the experimental
+    // variants fall through to it)
+    dim3 grid((Sq + 12 * 16 - 1) / (12 * 16), Hq, B);
+    const float dp = (float)dropout_p;
+    const unsigned long long dseed = (unsigned long long)drop_seed;
+    const unsigned long long doff = (unsigned long long)drop_offset;
+    if (D == 128) {
+      hipLaunchKernelGGL((fa_fwd_kernel<128, 12, true>), grid, dim3(12 * 64),
+                         0, stream, (const __hip_bfloat16*)q.data_ptr(),
+                         (const __hip_bfloat16*)k.data_ptr(),
+                         (const __hip_bfloat16*)v.data_ptr(),
+                         (__hip_bfloat16*)out.data_ptr(),
+                         lse.data_ptr<float>(), B, Sq, Sk, Hq, Hkv,
+                         (float)softmax_scale, causal ? 1 : 0, win, qs, ks,
+                         vs, os, dp, dseed, doff);
+    } else {
+      hipLaunchKernelGGL((fa_fwd_kernel<64, 12, true>), grid, dim3(12 * 64),
+                         0, stream, (const __hip_bfloat16*)q.data_ptr(),
+                         (const __hip_bfloat16*)k.data_ptr(),
+                         (const __hip_bfloat16*)v.data_ptr(),
+                         (__hip_bfloat16*)out.data_ptr(),
+                         lse.data_ptr<float>(), B, Sq, Sk, Hq, Hkv,
+                         (float)softmax_scale, causal ? 1 : 0, win, qs, ks,
+                         vs, os, dp, dseed, doff);
+    }
+    return {out, lse};
+  }
   if (nw_env == 82) {
     // 8 waves x 2 bands (BM = 256)
     dim3 grid((Sq + 255) / 256, Hq, B);
