@@ -4,6 +4,9 @@
 
 #include <vector>
 
+bool wgrad_gemm_hand(torch::Tensor input, torch::Tensor grad_output,
+                     torch::Tensor main_grad);
+
 // norms.hip
 std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor weight,
                                        double eps);
@@ -87,6 +90,7 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("wgrad_gemm_hand", &wgrad_gemm_hand);
   m.def("rmsnorm_bwd", &rmsnorm_bwd, py::arg("dy"), py::arg("x"),
         py::arg("weight"), py::arg("inv"),
         py::arg("dres") = py::none());

@@ -178,6 +178,11 @@ LtPlan make_plan(long in_dim, long out_dim, long K, hipDataType ab_type,
 
 }  // namespace
 
+// wgrad_gemm.hip: hand-written 256x256-tile MFMA GEMM for the dividing
+// shapes; returns false -> fall through to the hipBLASLt path
+bool wgrad_gemm_hand(torch::Tensor input, torch::Tensor grad_output,
+                     torch::Tensor main_grad);
+
 void wgrad_gemm_accum_fp32(torch::Tensor input, torch::Tensor grad_output,
                            torch::Tensor main_grad) {
   TORCH_CHECK(input.is_cuda() && input.dim() == 2 && input.is_contiguous());
@@ -199,6 +204,14 @@ void wgrad_gemm_accum_fp32(torch::Tensor input, torch::Tensor grad_output,
     ab_type = HIP_R_32F;
   }
   TORCH_CHECK(grad_output.scalar_type() == input.scalar_type());
+
+  static const int use_hand = []() {
+    const char* e = getenv("MEGATRON_AMD_WGRAD_HAND");
+    return e ? atoi(e) : 0;  // flips to 1 once measured faster
+  }();
+  if (use_hand && wgrad_gemm_hand(input, grad_output, main_grad)) {
+    return;
+  }
 
   hipStream_t stream = c10::hip::getCurrentHIPStream();
   ShapeKey key{in_dim, out_dim, K, (int)ab_type};

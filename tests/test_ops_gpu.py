@@ -487,3 +487,27 @@ class TestFp8Quantize:
         recon = q2.float() * s2
         rel = ((recon - x.float()).norm() / x.float().norm()).item()
         assert rel < 0.05
+
+
+class TestWgradHandKernel:
+    @pytest.mark.parametrize("mn", [(256, 256), (512, 256), (768, 512)])
+    def test_matches_fp32_reference(self, mn):
+        """Hand 256x256-tile MFMA wgrad GEMM vs fp32 torch reference,
+        including the += accumulate semantics."""
+        ext = _ext()
+        M, N = mn
+        K = 256
+        g = torch.randn(K, M, device="cuda", dtype=torch.bfloat16)
+        x = torch.randn(K, N, device="cuda", dtype=torch.bfloat16)
+        mg = torch.randn(M, N, device="cuda", dtype=torch.float32)
+        expected = mg + g.float().t() @ x.float()
+        ok = ext.wgrad_gemm_hand(x, g, mg)
+        assert ok
+        assert rel_err(mg, expected) < 2e-2
+
+    def test_rejects_nondividing(self):
+        ext = _ext()
+        g = torch.randn(64, 100, device="cuda", dtype=torch.bfloat16)
+        x = torch.randn(64, 256, device="cuda", dtype=torch.bfloat16)
+        mg = torch.zeros(100, 256, device="cuda", dtype=torch.float32)
+        assert not ext.wgrad_gemm_hand(x, g, mg)
