@@ -212,6 +212,152 @@ __global__ __launch_bounds__(kThreads, 2) void wgrad_gemm_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// v2: counted-vmcnt half-tile pipeline (the T3+T4 structure). Two phases per
+// K-step, each {issue 4 glds for tile t+1's half h; 32 MFMAs on tile t's
+// half h; s_waitcnt vmcnt(4) — the NEXT phase's half has landed, the one
+// just issued stays in flight ACROSS the barrier; raw s_barrier}. Never
+// vmcnt(0) in the main loop (the 2-phase v1 above drains the whole glds
+// queue at every barrier — its measured ~900 TF ceiling).
+
+__global__ __launch_bounds__(kThreads, 2) void wgrad_gemm_pipe_kernel(
+    const __hip_bfloat16* __restrict__ a,  // [K, M]
+    const __hip_bfloat16* __restrict__ b,  // [K, N]
+    float* __restrict__ c, int M, int N, int K, int group_m) {
+  __shared__ __hip_bfloat16 a_img[2][kImgElems];
+  __shared__ __hip_bfloat16 b_img[2][kImgElems];
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wm = wave >> 2;
+  const int wn = wave & 3;
+
+  const int tiles_m = M / kBM;
+  int bid = blockIdx.x;
+  int group_sz = group_m * (N / kBN);
+  int g = bid / group_sz;
+  int r = bid % group_sz;
+  int gm = min(group_m, tiles_m - g * group_m);
+  int tm = g * group_m + r % gm;
+  int tn = r / gm;
+  const int m0 = tm * kBM;
+  const int n0 = tn * kBN;
+
+  frag_f32 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = frag_f32{0.f, 0.f, 0.f, 0.f};
+  }
+
+  // stage_half: half h of tile kt = tile-rows [h*8, h*8+8) = k in
+  // [h*32, h*32+32); 16 glds instructions over 8 waves = 2 per wave
+  auto stage_half = [&](const __hip_bfloat16* src, long ld,
+                        __hip_bfloat16* img, int k_base, int h) {
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      int u = h * 16 + wave * 2 + p;  // unit in [h*16, h*16+16)
+      int rt = u >> 1;
+      int half = u & 1;
+      int k = rt * 4 + ((lane >> 1) & 3);
+      int col = half * 128 + (lane >> 3) * 16 + (lane & 1) * 8;
+      const __hip_bfloat16* gsrc = src + (long)(k_base + k) * ld + col;
+      __hip_bfloat16* dst = img + rt * kTRS + tr_shift(rt) + half * 512;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gsrc,
+          (__attribute__((address_space(3))) unsigned int*)dst, 16, 0, 0);
+    }
+  };
+  // one compute phase: 32 MFMAs on k-chunk kc of the given images
+  auto compute_half = [&](const __hip_bfloat16* ai, const __hip_bfloat16* bi,
+                          int kc) {
+    const int k0 = kc * 32;
+    frag_b16 bf[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      bf[j] = tr_frag(bi, k0, (wn * 64 + j * 16) >> 4, lane);
+    }
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      frag_b16 af = tr_frag(ai, k0, (wm * 128 + i * 16) >> 4, lane);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf[j],
+                                                            acc[i][j], 0, 0,
+                                                            0);
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+  };
+
+  const int ksteps = K / kBK;
+  // prologue: tile 0 fully, tile 1 half 0; drain everything once
+  stage_half(a + m0, M, a_img[0], 0, 0);
+  stage_half(b + n0, N, b_img[0], 0, 0);
+  stage_half(a + m0, M, a_img[0], 0, 1);
+  stage_half(b + n0, N, b_img[0], 0, 1);
+  if (ksteps > 1) {
+    stage_half(a + m0, M, a_img[1], kBK, 0);
+    stage_half(b + n0, N, b_img[1], kBK, 0);
+  }
+  if (ksteps > 1) {
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");  // tile 0 landed
+  } else {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  }
+  __builtin_amdgcn_s_barrier();
+
+  for (int kt = 0; kt < ksteps; ++kt) {
+    const int cur = kt & 1;
+    const int nxt = cur ^ 1;
+    const bool more = kt + 1 < ksteps;
+    // phase A: issue t+1 half1, compute t half0
+    if (more) {
+      stage_half(a + m0, M, a_img[nxt], (kt + 1) * kBK, 1);
+      stage_half(b + n0, N, b_img[nxt], (kt + 1) * kBK, 1);
+    }
+    compute_half(a_img[cur], b_img[cur], 0);
+    // t's halves landed long ago; t+1.h0 must land before NEXT tile's
+    // phase A, but phase B only needs t.h1 (resident) — keep everything
+    // in flight here, no wait, LDS-safety barrier only
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    // phase B: issue t+2 half0, compute t half1
+    if (kt + 2 < ksteps) {
+      stage_half(a + m0, M, a_img[cur], (kt + 2) * kBK, 0);
+      stage_half(b + n0, N, b_img[cur], (kt + 2) * kBK, 0);
+    }
+    compute_half(a_img[cur], b_img[cur], 1);
+    // before the buffer flip: t+1's halves (issued 2 and 1 phases ago)
+    // must be LDS-visible; t+2.h0 (4 glds/wave) may stay in flight
+    if (more) {
+      if (kt + 2 < ksteps) {
+        asm volatile("s_waitcnt vmcnt(4) lgkmcnt(0)" ::: "memory");
+      } else {
+        asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+      }
+    } else {
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+  }
+
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    int row_base = m0 + wm * 128 + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int r2 = 0; r2 < 4; ++r2) {
+      long row = row_base + r2;
+      float* crow = c + row * (long)N + n0 + wn * 64 + (lane & 15);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        crow[j * 16] += acc[i][j][r2];
+      }
+    }
+  }
+}
+
 }  // namespace
 
 // Host entry: returns false when the shape does not fit the fast path
@@ -231,12 +377,25 @@ bool wgrad_gemm_hand(torch::Tensor input, torch::Tensor grad_output,
     const char* e = getenv("MEGATRON_AMD_WGRAD_GROUP_M");
     return e ? atoi(e) : 8;
   }();
+  static const int variant = []() {
+    const char* e = getenv("MEGATRON_AMD_WGRAD_V");
+    return e ? atoi(e) : 2;  // 2 = counted-vmcnt pipeline, 1 = simple
+  }();
   auto stream = c10::hip::getCurrentHIPStream();
   dim3 grid((M / kBM) * (N / kBN));
-  hipLaunchKernelGGL(wgrad_gemm_kernel, grid, dim3(kThreads), 0, stream,
-                     (const __hip_bfloat16*)grad_output.data_ptr(),
-                     (const __hip_bfloat16*)input.data_ptr(),
-                     main_grad.data_ptr<float>(), (int)M, (int)N, (int)K,
-                     group_m);
+  if (variant == 2) {
+    hipLaunchKernelGGL(wgrad_gemm_pipe_kernel, grid, dim3(kThreads), 0,
+                       stream,
+                       (const __hip_bfloat16*)grad_output.data_ptr(),
+                       (const __hip_bfloat16*)input.data_ptr(),
+                       main_grad.data_ptr<float>(), (int)M, (int)N, (int)K,
+                       group_m);
+  } else {
+    hipLaunchKernelGGL(wgrad_gemm_kernel, grid, dim3(kThreads), 0, stream,
+                       (const __hip_bfloat16*)grad_output.data_ptr(),
+                       (const __hip_bfloat16*)input.data_ptr(),
+                       main_grad.data_ptr<float>(), (int)M, (int)N, (int)K,
+                       group_m);
+  }
   return true;
 }
