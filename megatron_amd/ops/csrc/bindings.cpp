@@ -4,6 +4,7 @@
 
 #include <vector>
 
+torch::Tensor gemv_bf16(torch::Tensor weight, torch::Tensor x);
 bool wgrad_gemm_hand(torch::Tensor input, torch::Tensor grad_output,
                      torch::Tensor main_grad);
 
@@ -91,6 +92,7 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("wgrad_gemm_hand", &wgrad_gemm_hand);
+  m.def("gemv_bf16", &gemv_bf16);
   m.def("rmsnorm_bwd", &rmsnorm_bwd, py::arg("dy"), py::arg("x"),
         py::arg("weight"), py::arg("inv"),
         py::arg("dres") = py::none());

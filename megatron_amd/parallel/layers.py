@@ -103,7 +103,24 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
                 out2d = fp8_linear_fwd(ti2d, weight)
             output = out2d.view(*total_input.shape[:-1], weight.shape[0])
         else:
-            output = torch.matmul(total_input, weight.t())
+            output = None
+            if (total_input.is_cuda and not total_input.requires_grad
+                    and total_input.dtype == torch.bfloat16
+                    and weight.dtype == torch.bfloat16
+                    and weight.shape[1] % 8 == 0):
+                ti2d = total_input.reshape(-1, total_input.shape[-1])
+                if ti2d.shape[0] <= 4:
+                    # decode-time GEMV: the hand weight-stream kernel beats
+                    # the library's M<=4 GEMM selection (ops/csrc/gemv.hip)
+                    from ..ops import ext as _oext
+
+                    out2d = _oext.load(required=True).gemv_bf16(
+                        weight, ti2d.contiguous()
+                    )
+                    output = out2d.view(*total_input.shape[:-1],
+                                        weight.shape[0])
+            if output is None:
+                output = torch.matmul(total_input, weight.t())
         if bias is not None:
             output = output + bias
         return output

@@ -511,3 +511,18 @@ class TestWgradHandKernel:
         x = torch.randn(64, 256, device="cuda", dtype=torch.bfloat16)
         mg = torch.zeros(100, 256, device="cuda", dtype=torch.float32)
         assert not ext.wgrad_gemm_hand(x, g, mg)
+
+
+class TestGemv:
+    @pytest.mark.parametrize("t", [1, 2, 4])
+    @pytest.mark.parametrize("mk", [(4096, 4096), (12288, 4096),
+                                    (32000, 4096), (100, 1024)])
+    def test_matches_matmul(self, t, mk):
+        ext = _ext()
+        M, K = mk
+        w = torch.randn(M, K, device="cuda", dtype=torch.bfloat16) * 0.05
+        x = torch.randn(t, K, device="cuda", dtype=torch.bfloat16)
+        y = ext.gemv_bf16(w, x)
+        ref = x.float() @ w.float().t()
+        assert y.shape == (t, M)
+        assert rel_err(y, ref) < 2e-2
