@@ -30,15 +30,17 @@ __global__ __launch_bounds__(256) void gemv_kernel(
   const int row = blockIdx.x * 4 + wave;
   if (row >= M) return;
 
-  const uint4* wrow = reinterpret_cast<const uint4*>(w + (long)row * K);
+  using v4u = __attribute__((ext_vector_type(4))) unsigned int;
+  const v4u* wrow = reinterpret_cast<const v4u*>(w + (long)row * K);
   float acc[T];
 #pragma unroll
   for (int t = 0; t < T; ++t) acc[t] = 0.f;
 
   const int kv = K / 8;  // dwordx4 chunks per row
   for (int i = lane; i < kv; i += 64) {
+    v4u wraw = __builtin_nontemporal_load(&wrow[i]);
     Bf16x8 wv;
-    wv.u = __builtin_nontemporal_load(&wrow[i]);
+    wv.u = make_uint4(wraw.x, wraw.y, wraw.z, wraw.w);
 #pragma unroll
     for (int t = 0; t < T; ++t) {
       Bf16x8 xv;
