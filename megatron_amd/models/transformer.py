@@ -424,6 +424,18 @@ class ParallelAttention(MegatronModule):
 
         b = query.shape[1]
         n, hn = query.shape[2], query.shape[3]
+
+        if query.is_cuda and hn in (64, 128) and query.dtype == torch.bfloat16:
+            # fused decode attention (ops/csrc/decode_attn.hip): one kernel
+            # replaces the fp32-cast + bmm + mask + softmax + bmm chain
+            from ..ops import ext as _oext
+
+            ctx_row = _oext.load(required=True).decode_attn(
+                query[0].contiguous(), k_cache, v_cache, ip.graph_pos,
+                1.0 / math.sqrt(hn),
+            )
+            output, bias = self.dense(ctx_row.unsqueeze(0))
+            return output, bias
         nkv = k_cache.shape[2]
         L = k_cache.shape[0]
         q = query.permute(1, 2, 0, 3).reshape(b * n, 1, hn)  # [b*n,1,h]

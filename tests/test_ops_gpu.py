@@ -526,3 +526,32 @@ class TestGemv:
         ref = x.float() @ w.float().t()
         assert y.shape == (t, M)
         assert rel_err(y, ref) < 2e-2
+
+
+class TestDecodeAttn:
+    @pytest.mark.parametrize("gqa", [1, 4])
+    @pytest.mark.parametrize("h", [128, 64])
+    def test_matches_eager(self, gqa, h):
+        """Fused decode attention vs the eager fp32 cache-softmax chain."""
+        ext = _ext()
+        b, n, L = 2, 8, 96
+        nkv = n // gqa
+        pos_val = 57
+        q = torch.randn(b, n, h, device="cuda", dtype=torch.bfloat16)
+        kc = torch.randn(L, b, nkv, h, device="cuda", dtype=torch.bfloat16)
+        vc = torch.randn(L, b, nkv, h, device="cuda", dtype=torch.bfloat16)
+        pos = torch.tensor([pos_val], device="cuda", dtype=torch.long)
+        scale = 1.0 / math.sqrt(h)
+        out = ext.decode_attn(q, kc, vc, pos, scale)
+
+        # eager reference
+        kr = kc.permute(1, 2, 0, 3).float()  # [b,nkv,L,h]
+        vr = vc.permute(1, 2, 0, 3).float()
+        if gqa > 1:
+            kr = kr.repeat_interleave(gqa, dim=1)
+            vr = vr.repeat_interleave(gqa, dim=1)
+        scores = torch.einsum("bnh,bnlh->bnl", q.float(), kr) * scale
+        scores[:, :, pos_val + 1:] = float("-inf")
+        p = torch.softmax(scores, dim=-1)
+        ref = torch.einsum("bnl,bnlh->bnh", p, vr).reshape(b, n * h)
+        assert rel_err(out, ref) < 3e-2
