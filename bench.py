@@ -42,8 +42,9 @@ def parse_args():
     p.add_argument("--steps", type=int, default=8)
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--model", type=str, default="llama2-7b",
-                   choices=["llama2-7b", "llama2-70b", "mistral-7b",
-                            "falcon-7b", "gpt-125m", "llama2-tiny"])
+                   choices=["llama2-7b", "llama2-70b", "llama2-70b-shard8",
+                            "mistral-7b", "falcon-7b", "gpt-125m",
+                            "llama2-tiny"])
     p.add_argument("--seq-len", type=int, default=None)
     p.add_argument("--micro-batch-size", type=int, default=None)
     p.add_argument("--global-batch", type=int, default=None)
@@ -79,6 +80,14 @@ MODEL_SPECS = {
     "falcon-7b": dict(num_layers=32, hidden_size=4544, ffn_hidden_size=18176,
                       num_attention_heads=71, num_attention_heads_kv=1,
                       vocab=65024, seq=2048, model_name="falcon"),
+    # one TP4xPP2 rank's shard of Llama-2-70B: 40 layers (one PP stage),
+    # heads/kv-heads/ffn divided by tp=4, hidden kept full (TP shards only
+    # the head and ffn dimensions) — single-GPU memory/timing proxy for the
+    # per-rank footprint of the 8-GPU 70B config (VERDICT #3 validation)
+    "llama2-70b-shard8": dict(num_layers=40, hidden_size=8192,
+                              ffn_hidden_size=7168, num_attention_heads=16,
+                              num_attention_heads_kv=2, vocab=32000,
+                              seq=4096, model_name="llama2"),
     "gpt-125m": dict(num_layers=12, hidden_size=768, ffn_hidden_size=3072,
                      num_attention_heads=12, num_attention_heads_kv=12,
                      vocab=50304, seq=1024, model_name="gpt"),
@@ -101,7 +110,8 @@ def main():
     have_gpu = torch.cuda.is_available()
     spec = dict(MODEL_SPECS[args.model])
     if not have_gpu and args.model in ("llama2-7b", "llama2-70b",
-                                       "mistral-7b", "falcon-7b"):
+                                       "llama2-70b-shard8", "mistral-7b",
+                                       "falcon-7b"):
         # CPU plumbing check shrinks the model but keeps the code path
         spec = dict(MODEL_SPECS["llama2-tiny"])
         spec["model_name"] = MODEL_SPECS[args.model]["model_name"]
@@ -264,6 +274,11 @@ def main():
                                      op=torch.distributed.ReduceOp.MAX)
     elapsed = elapsed_t.item()
 
+    if have_gpu:
+        print(f"peak allocated {torch.cuda.max_memory_allocated() / 2**30:.1f}"
+              f" GiB / reserved "
+              f"{torch.cuda.max_memory_reserved() / 2**30:.1f} GiB",
+              file=sys.stderr)
     ms_per_step = elapsed / args.steps * 1000.0
     tokens_per_step = gbs * seq
     tokens_per_sec = tokens_per_step / (elapsed / args.steps)
