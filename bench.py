@@ -87,7 +87,8 @@ MODEL_SPECS = {
     "llama2-70b-shard8": dict(num_layers=40, hidden_size=8192,
                               ffn_hidden_size=7168, num_attention_heads=16,
                               num_attention_heads_kv=2, vocab=32000,
-                              seq=4096, model_name="llama2"),
+                              seq=4096, model_name="llama2",
+                              kv_channels=128),
     "gpt-125m": dict(num_layers=12, hidden_size=768, ffn_hidden_size=3072,
                      num_attention_heads=12, num_attention_heads_kv=12,
                      vocab=50304, seq=1024, model_name="gpt"),
@@ -148,6 +149,7 @@ def main():
         ffn_hidden_size=spec["ffn_hidden_size"],
         num_attention_heads=spec["num_attention_heads"],
         num_attention_heads_kv=spec["num_attention_heads_kv"],
+        kv_channels=spec.get("kv_channels"),
         max_position_embeddings=max(seq, 4096),
         seq_length=seq, micro_batch_size=mbs, global_batch_size=gbs,
         tensor_model_parallel_size=tp, pipeline_model_parallel_size=pp,
