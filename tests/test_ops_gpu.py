@@ -58,11 +58,15 @@ class TestRMSNorm:
         assert rel_err(dx, xr.grad) < tol
         assert rel_err(dw, wr.grad) < tol
 
-        # fused residual-grad epilogue: dx_total = norm_dx + dres
+        # fused residual-grad epilogue: dx_total = norm_dx + dres. The
+        # kernel adds dres BEFORE the output round (one rounding), the
+        # reference here rounds dx first (two roundings) -> bf16 tolerance;
+        # dw comparison allows fp32 atomic-order nondeterminism.
         dres = torch.randn_like(x)
         dx2, dw2 = ext.rmsnorm_bwd(dy, x, w, inv, dres)
-        assert rel_err(dx2.float(), dx.float() + dres.float()) < 1e-3
-        assert rel_err(dw2, dw) < 1e-6
+        tol2 = 2e-2 if dtype == torch.bfloat16 else 1e-5
+        assert rel_err(dx2.float(), dx.float() + dres.float()) < tol2
+        assert rel_err(dw2, dw) < 1e-4
 
 
 class TestLayerNorm:
