@@ -357,6 +357,35 @@ class ParallelAttention(MegatronModule):
             key = kv[..., :hn].contiguous()
             value = kv[..., hn:].contiguous()
 
+        # fused decode step: {rope + cache-append}, {attention} — three
+        # kernels total (with the dense gemv) instead of the ~20-op eager
+        # chain; all positions device-held so the step is graph-capturable
+        if (
+            self.rope_cos is not None
+            and inference_params is not None
+            and getattr(inference_params, "use_graph", False)
+            and sq == 1
+            and self.layer_number in inference_params.key_value_memory_dict
+            and self.attention_type == AttnType.self_attn
+            and query.is_cuda
+            and query.dtype == torch.bfloat16
+            and hn in (64, 128)
+        ):
+            from ..ops import ext as _oext
+
+            _m = _oext.load(required=True)
+            ip = inference_params
+            k_cache, v_cache = ip.key_value_memory_dict[self.layer_number]
+            cos = self.rope_cos.to(device=query.device, dtype=torch.float32)
+            sin = self.rope_sin.to(device=query.device, dtype=torch.float32)
+            q_rot = _m.decode_rope_append(
+                query[0], key[0], value[0], cos, sin, ip.graph_pos,
+                k_cache, v_cache,
+            )
+            ctx_row = _m.decode_attn(q_rot, k_cache, v_cache, ip.graph_pos,
+                                     1.0 / math.sqrt(hn))
+            return self.dense(ctx_row.unsqueeze(0))
+
         # rotary embedding
         if self.rope_cos is not None:
             # tables stay fp32 regardless of module dtype conversion

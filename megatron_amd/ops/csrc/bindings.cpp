@@ -8,6 +8,11 @@ torch::Tensor gemv_bf16(torch::Tensor weight, torch::Tensor x);
 torch::Tensor decode_attn(torch::Tensor q, torch::Tensor k_cache,
                           torch::Tensor v_cache, torch::Tensor pos,
                           double scale);
+torch::Tensor decode_rope_append(torch::Tensor q, torch::Tensor k,
+                                 torch::Tensor v, torch::Tensor cos,
+                                 torch::Tensor sin, torch::Tensor pos,
+                                 torch::Tensor k_cache,
+                                 torch::Tensor v_cache);
 bool wgrad_gemm_hand(torch::Tensor input, torch::Tensor grad_output,
                      torch::Tensor main_grad);
 
@@ -97,6 +102,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("wgrad_gemm_hand", &wgrad_gemm_hand);
   m.def("gemv_bf16", &gemv_bf16);
   m.def("decode_attn", &decode_attn);
+  m.def("decode_rope_append", &decode_rope_append);
   m.def("rmsnorm_bwd", &rmsnorm_bwd, py::arg("dy"), py::arg("x"),
         py::arg("weight"), py::arg("inv"),
         py::arg("dres") = py::none());

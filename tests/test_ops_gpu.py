@@ -559,3 +559,34 @@ class TestDecodeAttn:
         p = torch.softmax(scores, dim=-1)
         ref = torch.einsum("bnl,bnlh->bnh", p, vr).reshape(b, n * h)
         assert rel_err(out, ref) < 3e-2
+
+
+class TestDecodeRopeAppend:
+    def test_matches_eager(self):
+        """Fused rope+append vs eager rope + index_copy."""
+        from megatron_amd.models.rope import precompute_freqs
+
+        ext = _ext()
+        b, n, nkv, h, L = 2, 8, 2, 128, 64
+        pos_val = 37
+        q = torch.randn(b, n, h, device="cuda", dtype=torch.bfloat16)
+        k = torch.randn(b, nkv, h, device="cuda", dtype=torch.bfloat16)
+        v = torch.randn(b, nkv, h, device="cuda", dtype=torch.bfloat16)
+        cos, sin = precompute_freqs(h, L, device="cuda")
+        pos = torch.tensor([pos_val], device="cuda", dtype=torch.long)
+        kc = torch.zeros(L, b, nkv, h, device="cuda", dtype=torch.bfloat16)
+        vc = torch.zeros(L, b, nkv, h, device="cuda", dtype=torch.bfloat16)
+        q_rot = ext.decode_rope_append(q, k, v, cos, sin, pos, kc, vc)
+
+        def rope(x):
+            xf = x.float()
+            x1, x2 = xf[..., 0::2], xf[..., 1::2]
+            c = cos[pos_val].view(1, 1, -1)
+            s = sin[pos_val].view(1, 1, -1)
+            return torch.stack([x1 * c - x2 * s, x2 * c + x1 * s],
+                               -1).flatten(-2)
+
+        assert rel_err(q_rot, rope(q)) < 3e-2
+        assert rel_err(kc[pos_val], rope(k)) < 3e-2
+        assert torch.equal(vc[pos_val], v)
+        assert kc[pos_val + 1].abs().max().item() == 0  # only one row written
