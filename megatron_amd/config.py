@@ -138,7 +138,7 @@ class TrainingConfig:
     # hipGraph-captured single-token decode (inference/forward_step.py).
     # Opt-in: one replay replaces the per-token launch storm; enable for
     # serving (tools/run_text_generation_server.py turns it on)
-    use_hip_graph_decode: bool = False
+    use_hip_graph_decode: bool = True   # hipGraph-replayed decode (TP1/PP1)
 
     # -- parallelism --
     tensor_model_parallel_size: int = 1

@@ -348,3 +348,52 @@ def test_fp8_linear_and_training(dist_single):
         assert skipped == 0
         losses.append(loss_dict["lm loss"].item())
     assert losses[-1] < losses[0] * 0.9, losses
+
+
+def test_multi_graph_lifetime(gpu_cfg):
+    """Round-1 left hipGraph decode opt-in due to a suspected interaction
+    when several CUDAGraphs are created/destroyed in one process. Guard the
+    now-default-on path: three capture/replay/destroy cycles (different
+    batch sizes) must each produce logits matching a fresh eager run."""
+    import gc
+
+    from megatron_amd import global_state
+    from megatron_amd.config import set_config
+    from megatron_amd.inference.forward_step import ForwardStep
+    from megatron_amd.models import LlamaModel
+    from megatron_amd.tokenizer.tokenizers import FakeTokenizer
+    from megatron_amd.utils import get_ltor_masks_and_position_ids
+
+    cfg = gpu_cfg
+    set_config(cfg)
+    global_state.set_tokenizer(FakeTokenizer(1000))
+    torch.manual_seed(11)
+    m = LlamaModel(cfg, parallel_output=False).cuda().bfloat16()
+    m.eval()
+
+    for cycle, b in enumerate([1, 2, 1]):
+        total, prompt_len = 12, 6
+        torch.manual_seed(100 + cycle)
+        tokens = torch.randint(1, 999, (b, total), device="cuda")
+        am, _, pids = get_ltor_masks_and_position_ids(tokens, 0, False,
+                                                      False, False)
+
+        def run(use_graph):
+            cfg.use_hip_graph_decode = use_graph
+            fs = ForwardStep(m, b, total)
+            with torch.no_grad():
+                fs(tokens[:, :prompt_len], pids[:, :prompt_len],
+                   am[..., :prompt_len, :prompt_len])
+                outs = []
+                for i in range(prompt_len, total):
+                    lg = fs(tokens[:, i:i + 1], pids[:, i:i + 1],
+                            am[..., i:i + 1, :i + 1])
+                    outs.append(lg[:, -1, :].float().clone())
+            del fs
+            gc.collect()
+            return outs
+
+        eager = run(False)
+        graph = run(True)
+        for i, (e, g) in enumerate(zip(eager, graph)):
+            assert (e - g).abs().max().item() < 5e-2, (cycle, i)
