@@ -382,8 +382,11 @@ class ParallelAttention(MegatronModule):
                 query[0], key[0], value[0], cos, sin, ip.graph_pos,
                 k_cache, v_cache,
             )
+            win = -1
+            if self.use_flash_attn and self.flash_attention.window_size:
+                win = int(self.flash_attention.window_size)
             ctx_row = _m.decode_attn(q_rot, k_cache, v_cache, ip.graph_pos,
-                                     1.0 / math.sqrt(hn))
+                                     1.0 / math.sqrt(hn), win)
             return self.dense(ctx_row.unsqueeze(0))
 
         # rotary embedding
@@ -480,6 +483,10 @@ class ParallelAttention(MegatronModule):
         scale = 1.0 / math.sqrt(hn)
         scores = torch.bmm(q.float(), k.float().transpose(1, 2)) * scale
         invalid = ip.graph_arange > ip.graph_pos  # [L] device-side length
+        if self.use_flash_attn and self.flash_attention.window_size:
+            # sliding window (HF/Mistral: keep w keys incl. current)
+            w = int(self.flash_attention.window_size)
+            invalid = invalid | (ip.graph_arange < ip.graph_pos - w + 1)
         scores = scores.masked_fill(invalid.view(1, 1, L), float("-inf"))
         probs = torch.softmax(scores, dim=-1).to(v.dtype)
         ctx = torch.bmm(probs, v)  # [b*n,1,h]

@@ -7,7 +7,7 @@
 torch::Tensor gemv_bf16(torch::Tensor weight, torch::Tensor x);
 torch::Tensor decode_attn(torch::Tensor q, torch::Tensor k_cache,
                           torch::Tensor v_cache, torch::Tensor pos,
-                          double scale);
+                          double scale, int64_t window);
 torch::Tensor decode_rope_append(torch::Tensor q, torch::Tensor k,
                                  torch::Tensor v, torch::Tensor cos,
                                  torch::Tensor sin, torch::Tensor pos,
@@ -101,7 +101,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("wgrad_gemm_hand", &wgrad_gemm_hand);
   m.def("gemv_bf16", &gemv_bf16);
-  m.def("decode_attn", &decode_attn);
+  m.def("decode_attn", &decode_attn, py::arg("q"),
+        py::arg("k_cache"), py::arg("v_cache"), py::arg("pos"),
+        py::arg("scale"), py::arg("window") = -1);
   m.def("decode_rope_append", &decode_rope_append);
   m.def("rmsnorm_bwd", &rmsnorm_bwd, py::arg("dy"), py::arg("x"),
         py::arg("weight"), py::arg("inv"),

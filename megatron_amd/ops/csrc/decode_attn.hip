@@ -30,7 +30,7 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
     const __hip_bfloat16* __restrict__ k_cache,
     const __hip_bfloat16* __restrict__ v_cache,
     const long* __restrict__ pos_ptr, __hip_bfloat16* __restrict__ out,
-    int B, int N, int NKV, int L, float scale) {
+    int B, int N, int NKV, int L, float scale, int window) {
   constexpr int kPerLane = H / 64;  // q/k/v elements per lane (1 or 2)
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -56,7 +56,11 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
 #pragma unroll
   for (int e = 0; e < kPerLane; ++e) o[e] = 0.f;
 
-  for (long l = wave; l <= pos; l += 4) {
+  // sliding window: keep `window` keys INCLUDING the current position
+  // (HF/Mistral convention, see ops/functional.py _attn_mask)
+  long lo = 0;
+  if (window > 0 && pos - window + 1 > 0) lo = pos - window + 1;
+  for (long l = lo + wave; l <= pos; l += 4) {
     const __hip_bfloat16* krow = kb + l * row_stride;
     float dot = 0.f;
 #pragma unroll
@@ -114,7 +118,7 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
 
 torch::Tensor decode_attn(torch::Tensor q, torch::Tensor k_cache,
                           torch::Tensor v_cache, torch::Tensor pos,
-                          double scale) {
+                          double scale, int64_t window) {
   TORCH_CHECK(q.is_cuda() && q.dim() == 3 && q.is_contiguous());
   TORCH_CHECK(k_cache.dim() == 4 && k_cache.is_contiguous());
   TORCH_CHECK(v_cache.is_contiguous());
@@ -136,14 +140,14 @@ torch::Tensor decode_attn(torch::Tensor q, torch::Tensor k_cache,
                        (const __hip_bfloat16*)k_cache.data_ptr(),
                        (const __hip_bfloat16*)v_cache.data_ptr(),
                        pos.data_ptr<long>(), (__hip_bfloat16*)out.data_ptr(),
-                       B, N, NKV, L, (float)scale);
+                       B, N, NKV, L, (float)scale, (int)window);
   } else {
     hipLaunchKernelGGL((decode_attn_kernel<64>), grid, dim3(256), 0, stream,
                        (const __hip_bfloat16*)q.data_ptr(),
                        (const __hip_bfloat16*)k_cache.data_ptr(),
                        (const __hip_bfloat16*)v_cache.data_ptr(),
                        pos.data_ptr<long>(), (__hip_bfloat16*)out.data_ptr(),
-                       B, N, NKV, L, (float)scale);
+                       B, N, NKV, L, (float)scale, (int)window);
   }
   return out;
 }

@@ -590,3 +590,26 @@ class TestDecodeRopeAppend:
         assert rel_err(kc[pos_val], rope(k)) < 3e-2
         assert torch.equal(vc[pos_val], v)
         assert kc[pos_val + 1].abs().max().item() == 0  # only one row written
+
+
+class TestDecodeAttnWindow:
+    def test_sliding_window(self):
+        """decode_attn with window w attends only the last w cache rows
+        (incl. current) — HF/Mistral convention."""
+        ext = _ext()
+        b, n, h, L, w = 1, 2, 128, 64, 8
+        pos_val = 40
+        q = torch.randn(b, n, h, device="cuda", dtype=torch.bfloat16)
+        kc = torch.randn(L, b, n, h, device="cuda", dtype=torch.bfloat16)
+        vc = torch.randn(L, b, n, h, device="cuda", dtype=torch.bfloat16)
+        pos = torch.tensor([pos_val], device="cuda", dtype=torch.long)
+        scale = 1.0 / math.sqrt(h)
+        out = ext.decode_attn(q, kc, vc, pos, scale, w)
+        kr = kc.permute(1, 2, 0, 3).float()
+        vr = vc.permute(1, 2, 0, 3).float()
+        scores = torch.einsum("bnh,bnlh->bnl", q.float(), kr) * scale
+        scores[:, :, :pos_val - w + 1] = float("-inf")
+        scores[:, :, pos_val + 1:] = float("-inf")
+        p = torch.softmax(scores, dim=-1)
+        ref = torch.einsum("bnl,bnlh->bnh", p, vr).reshape(b, n * h)
+        assert rel_err(out, ref) < 3e-2
