@@ -54,6 +54,10 @@ def parse_args():
                    help="override layer count (shape validation only; "
                    "results with this flag are not official numbers)")
     p.add_argument("--recompute", action="store_true")
+    p.add_argument("--no-grad-accum-fusion", action="store_true",
+                   help="autograd bf16 wgrad + fp32 hook accumulate (the "
+                   "reference's non-apex path) instead of the fused "
+                   "fp32-accum wgrad GEMM")
     p.add_argument("--dtype", type=str, default="bf16",
                    choices=["bf16", "fp16", "fp8"])
     return p.parse_args()
@@ -160,6 +164,7 @@ def main():
         lr_warmup_iters=0, clip_grad=1.0,
         hidden_dropout=0.0, attention_dropout=0.0,
         use_flash_attn=True,
+        gradient_accumulation_fusion=not args.no_grad_accum_fusion,
         recompute_granularity=(
             "full" if (args.recompute or spec.get("recompute"))
             else "selective"

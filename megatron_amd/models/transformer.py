@@ -370,6 +370,7 @@ class ParallelAttention(MegatronModule):
             and query.is_cuda
             and query.dtype == torch.bfloat16
             and hn in (64, 128)
+            and not torch.is_grad_enabled()
         ):
             from ..ops import ext as _oext
 
