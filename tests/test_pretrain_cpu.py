@@ -10,6 +10,8 @@ import pytest
 import torch
 import torch.multiprocessing as mp
 
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
@@ -102,3 +104,25 @@ def test_pretrain_end_to_end_and_resume(tmp_path):
     d = str(tmp_path)
     mp.spawn(_worker, args=(29641, d), nprocs=1, join=True)
     mp.spawn(_worker_resume, args=(29642, d), nprocs=1, join=True)
+
+
+def test_bench_torchrun_two_ranks(tmp_path):
+    """The driver's multi-GPU invocation shape — torch.distributed.run with
+    one rank per 'GPU' — must work end to end (cpu/gloo here): JSON contract
+    line parses, n_gpus == world size, dp2 geometry chosen."""
+    import json
+    import subprocess
+    import sys
+
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29917", os.path.join(REPO, "bench.py"),
+         "--gpus", "2", "--steps", "1", "--warmup", "0"],
+        capture_output=True, text=True, timeout=600, cwd=REPO,
+    )
+    assert r.returncode == 0, r.stderr[-1500:]
+    line = [ln for ln in r.stdout.splitlines() if ln.startswith("{")][-1]
+    res = json.loads(line)
+    assert res["n_gpus"] == 2
+    assert res["config"]["parallelism"] == "tp1_pp1_dp2"
