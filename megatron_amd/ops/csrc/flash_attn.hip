@@ -347,7 +347,8 @@ __global__ __launch_bounds__(NW * 64) void fa_fwd_kernel(
       m_run[r] = m_new[r];
     }
 
-    float p_val[4][4];
+    // p overwrites s in place (overlapping lifetimes — keeps the register
+    // count under the 4-waves/SIMD budget for the 16-wave variant)
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       float acc = 0.f;
@@ -355,7 +356,7 @@ __global__ __launch_bounds__(NW * 64) void fa_fwd_kernel(
       for (int t = 0; t < 4; ++t) {
         float pv = (s_val[t][r] < -1e29f) ? 0.f
                                           : __expf(s_val[t][r] - m_new[r]);
-        p_val[t][r] = pv;
+        s_val[t][r] = pv;
         acc += pv;
       }
       float rowsum = group16_sum(acc);
@@ -372,7 +373,7 @@ __global__ __launch_bounds__(NW * 64) void fa_fwd_kernel(
       for (int r = 0; r < 4; ++r) {
         int row = kgroup * 4 + r;
         int col = t * 16 + row_in_tile;
-        float pw = p_val[t][r];
+        float pw = s_val[t][r];
         if (DROP) {
           // dropout applies to P feeding O (the lse/l_run softmax state
           // stays dropout-independent)
@@ -1488,10 +1489,12 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
   }
 #define LAUNCH_FWD(DD, NW)                                                      do {                                                                            dim3 grid((Sq + NW * 16 - 1) / (NW * 16), Hq, B);                             hipLaunchKernelGGL((fa_fwd_kernel<DD, NW>), grid, dim3(NW * 64), 0,                              stream, (const __hip_bfloat16*)q.data_ptr(),                                  (const __hip_bfloat16*)k.data_ptr(),                                          (const __hip_bfloat16*)v.data_ptr(),                                          (__hip_bfloat16*)out.data_ptr(),                                              lse.data_ptr<float>(), B, Sq, Sk, Hq, Hkv,                                    (float)softmax_scale, causal ? 1 : 0, win,                                    qs, ks, vs, os);              } while (0)
   if (D == 128) {
-    if (nw_env >= 12) LAUNCH_FWD(128, 12);
+    if (nw_env == 16) LAUNCH_FWD(128, 16);
+    else if (nw_env >= 12) LAUNCH_FWD(128, 12);
     else LAUNCH_FWD(128, 4);
   } else {
-    if (nw_env >= 12) LAUNCH_FWD(64, 12);
+    if (nw_env == 16) LAUNCH_FWD(64, 16);
+    else if (nw_env >= 12) LAUNCH_FWD(64, 12);
     else LAUNCH_FWD(64, 4);
   }
 #undef LAUNCH_FWD
