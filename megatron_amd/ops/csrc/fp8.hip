@@ -29,15 +29,40 @@ __global__ void fp8_quantize_kernel(const __hip_bfloat16* __restrict__ x,
                                     const float* __restrict__ scale_inv_src,
                                     uint8_t* __restrict__ q,
                                     float* __restrict__ amax, long n) {
-  // scale_inv_src holds the DEQUANT scale s (x ~= q * s); quantize by 1/s
+  // scale_inv_src holds the DEQUANT scale s (x ~= q * s); quantize by 1/s.
+  // Vectorized 8-wide: 16-B bf16 loads, hardware packed e4m3 converts
+  // (v_cvt_pk_fp8_f32 — OCP e4m3fn on gfx950), 8-B stores. The original
+  // scalar loop measured ~700 GB/s (2-B loads + 1-B stores); this is the
+  // plain streaming shape.
   const float qscale = 1.0f / scale_inv_src[0];
   float local_amax = 0.f;
-  for (long i = (long)blockIdx.x * kBlock + threadIdx.x; i < n;
+  const long nv = n / 8;
+  const uint4* xv = reinterpret_cast<const uint4*>(x);
+  uint2* qv = reinterpret_cast<uint2*>(q);
+  for (long i = (long)blockIdx.x * kBlock + threadIdx.x; i < nv;
+       i += (long)gridDim.x * kBlock) {
+    Bf16x8 vx;
+    vx.u = xv[i];
+    float f[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      f[e] = __bfloat162float(vx.h[e]);
+      local_amax = fmaxf(local_amax, fabsf(f[e]));
+      f[e] = fminf(fmaxf(f[e] * qscale, -kE4M3Max), kE4M3Max);
+    }
+    int w0 = 0, w1 = 0;
+    w0 = __builtin_amdgcn_cvt_pk_fp8_f32(f[0], f[1], w0, false);
+    w0 = __builtin_amdgcn_cvt_pk_fp8_f32(f[2], f[3], w0, true);
+    w1 = __builtin_amdgcn_cvt_pk_fp8_f32(f[4], f[5], w1, false);
+    w1 = __builtin_amdgcn_cvt_pk_fp8_f32(f[6], f[7], w1, true);
+    qv[i] = make_uint2((unsigned)w0, (unsigned)w1);
+  }
+  // tail (n % 8)
+  for (long i = nv * 8 + (long)blockIdx.x * kBlock + threadIdx.x; i < n;
        i += (long)gridDim.x * kBlock) {
     float v = __bfloat162float(x[i]);
     local_amax = fmaxf(local_amax, fabsf(v));
-    float scaled = v * qscale;
-    scaled = fminf(fmaxf(scaled, -kE4M3Max), kE4M3Max);
+    float scaled = fminf(fmaxf(v * qscale, -kE4M3Max), kE4M3Max);
     q[i] = c10::Float8_e4m3fn(scaled).x;
   }
   __shared__ float lds[kBlock / WAVE_SIZE];

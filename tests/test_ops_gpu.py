@@ -613,3 +613,19 @@ class TestDecodeAttnWindow:
         p = torch.softmax(scores, dim=-1)
         ref = torch.einsum("bnl,bnlh->bnh", p, vr).reshape(b, n * h)
         assert rel_err(out, ref) < 3e-2
+
+
+class TestFp8Quantize:
+    @pytest.mark.parametrize("n", [4096 * 512, 1000])  # vector + tail paths
+    def test_matches_torch_cast(self, n):
+        """Hardware packed e4m3 convert vs torch's cast at the same scale."""
+        ext = _ext()
+        x = (torch.randn(n, device="cuda", dtype=torch.bfloat16) * 3).contiguous()
+        scale = torch.tensor([0.5], device="cuda")  # dequant scale
+        q, amax = ext.fp8_quantize(x, scale)
+        ref = (x.float() * 2.0).clamp(-448, 448).to(torch.float8_e4m3fn)
+        qq = q.view(torch.float8_e4m3fn).float()
+        # RNE agreement except possibly a ULP at representable boundaries
+        diff = (qq - ref.float()).abs()
+        assert (diff == 0).float().mean().item() > 0.999
+        assert abs(amax.item() - x.float().abs().max().item()) < 1e-2
