@@ -131,6 +131,10 @@ def main():
     # mbs sweep on MI355X (profiles/r01): 15.5k @ mbs4, 16.7k @ mbs8 with
     # the wide-workgroup FA kernels
     mbs = args.micro_batch_size or spec.get("mbs") or (8 if have_gpu else 1)
+    if args.dtype == "fp8" and args.micro_batch_size is None and have_gpu:
+        # fp8's weight copies + quantized activations don't fit mbs8 x 2
+        # accumulation microbatches in 288 GB
+        mbs = 6
     # pp > 1 needs several in-flight microbatches to fill the 1F1B pipeline;
     # at pp == 1, 2 accumulation microbatches amortize the optimizer step
     # (measured +1.5% tokens/s at 7B mbs8, gpurun_out/r2_8.log)
