@@ -271,9 +271,17 @@ def main():
             sort_by="self_cuda_time_total", row_limit=30,
         ), file=sys.stderr)
     else:
-        for _ in range(args.steps):
-            train_step(forward_step_func, data_iter, model, optimizer,
-                       opt_sched, cfg)
+        for i in range(args.steps):
+            ld = train_step(forward_step_func, data_iter, model, optimizer,
+                            opt_sched, cfg)
+            if os.environ.get("MEGATRON_AMD_PRINT_LOSS") and rank == 0:
+                # convergence evidence mode (adds a sync per step — not for
+                # timing-quality numbers)
+                try:
+                    loss = ld[0]["lm loss"].item()
+                    print(f"step {i}: lm loss {loss:.4f}", file=sys.stderr)
+                except Exception:
+                    pass
         barrier_sync()
     elapsed = time.time() - t0
 
