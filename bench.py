@@ -60,6 +60,8 @@ def parse_args():
                    "fp32-accum wgrad GEMM")
     p.add_argument("--dtype", type=str, default="bf16",
                    choices=["bf16", "fp16", "fp8"])
+    p.add_argument("--fp8-wgrad", action="store_true",
+                   help="with --dtype fp8: run the wgrad GEMM in e4m3 too")
     return p.parse_args()
 
 
@@ -150,6 +152,8 @@ def main():
         # separate recipe, never the headline bf16 number
         dtype_flags["bf16"] = True
         dtype_flags["fp8"] = True
+        if args.fp8_wgrad:
+            dtype_flags["fp8_wgrad"] = True
 
     from megatron_amd.config import TrainingConfig, set_config
     from megatron_amd.initialize import initialize_megatron

@@ -139,6 +139,9 @@ class TrainingConfig:
     # Opt-in: one replay replaces the per-token launch storm; enable for
     # serving (tools/run_text_generation_server.py turns it on)
     use_hip_graph_decode: bool = True   # hipGraph-replayed decode (TP1/PP1)
+    fp8_wgrad: bool = False  # with fp8: also run the wgrad GEMM in e4m3
+    #   (cast-transpose operands, fp32 GEMM output accumulated into
+    #   main_grad) — experimental, off by default even under fp8
 
     # -- parallelism --
     tensor_model_parallel_size: int = 1

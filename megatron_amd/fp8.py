@@ -151,3 +151,49 @@ def fp8_linear_dgrad_delayed(dy2d, weight, meta_dy):
     qt, st = _cached_weight_t_fp8(weight)
     return torch._scaled_mm(a8, qt.t(), scale_a=sa, scale_b=st,
                             out_dtype=dy2d.dtype)
+
+
+def fp8_wgrad_enabled() -> bool:
+    from .config import get_config
+
+    try:
+        cfg = get_config()
+    except Exception:
+        return False
+    return bool(getattr(cfg, "fp8_wgrad", False))
+
+
+def fp8_linear_wgrad(x2d, dy2d, main_grad, meta_x=None, meta_dy=None):
+    """main_grad[N, K] += dy^T @ x with both GEMM operands in e4m3.
+
+    Uses the one-pass cast-transpose kernel (ops/csrc/fp8.hip) to produce
+    the transposed fp8 layouts hipBLASLt needs (A = dy^T row-major,
+    B = x as [M,K] column-major = x^T row-major viewed .t()); the GEMM
+    emits fp32 which accumulates into main_grad. Falls back to False
+    (caller runs the bf16 fp32-accum wgrad) without the extension or
+    _scaled_mm fp32-out support."""
+    mod = _ext()
+    if mod is None or not hasattr(mod, "fp8_cast_transpose"):
+        return False
+
+    def scale_for(t, meta):
+        if meta is not None:
+            return meta.scale, meta
+        amax = t.abs().amax().float().clamp(min=1e-12)
+        return (amax / E4M3_MAX).reshape(1), None
+
+    s_dy, m_dy = scale_for(dy2d, meta_dy)
+    s_x, m_x = scale_for(x2d, meta_x)
+    _, dyt8, amax_dy = mod.fp8_cast_transpose(dy2d.contiguous(), s_dy)
+    _, xt8, amax_x = mod.fp8_cast_transpose(x2d.contiguous(), s_x)
+    if m_dy is not None:
+        m_dy.update(amax_dy)
+    if m_x is not None:
+        m_x.update(amax_x)
+    try:
+        out = torch._scaled_mm(dyt8, xt8.t(), scale_a=s_dy, scale_b=s_x,
+                               out_dtype=torch.float32)
+    except RuntimeError:
+        return False
+    main_grad += out
+    return True

@@ -16,6 +16,9 @@ torch::Tensor decode_rope_append(torch::Tensor q, torch::Tensor k,
 bool wgrad_gemm_hand(torch::Tensor input, torch::Tensor grad_output,
                      torch::Tensor main_grad);
 
+std::vector<torch::Tensor> fp8_cast_transpose(torch::Tensor x,
+                                              torch::Tensor scale);
+
 // norms.hip
 std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor weight,
                                        double eps);
@@ -100,6 +103,7 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("wgrad_gemm_hand", &wgrad_gemm_hand);
+  m.def("fp8_cast_transpose", &fp8_cast_transpose);
   m.def("gemv_bf16", &gemv_bf16);
   m.def("decode_attn", &decode_attn, py::arg("q"),
         py::arg("k_cache"), py::arg("v_cache"), py::arg("pos"),

@@ -83,6 +83,103 @@ inline int grid_for(long total) {
   return (int)std::min<long>(g, 8192);
 }
 
+// Cast-transpose (TransformerEngine-style): one read of bf16 [R, C]
+// produces BOTH e4m3 layouts — q [R, C] and qT [C, R] — plus the tensor
+// amax. The fp8 wgrad GEMM needs its operands in the transposed layout
+// (hipBLASLt wants A row-major / B column-major), and a separate eager
+// transpose of an fp8 tensor would cost another full pass.
+//
+// 64x64 bf16 tiles staged through LDS with a +8 pad (conflict-free
+// transposed reads); 256 threads: each thread loads 8x (8-elem rows) and
+// stores 8x 8-elem columns-as-rows.
+__global__ __launch_bounds__(256) void fp8_cast_transpose_kernel(
+    const __hip_bfloat16* __restrict__ x,
+    const float* __restrict__ scale_inv_src, uint8_t* __restrict__ q,
+    uint8_t* __restrict__ qt, float* __restrict__ amax, long R, long C) {
+  const float qscale = 1.0f / scale_inv_src[0];
+  __shared__ uint8_t tile[64][72];  // quantized tile, padded
+  const long tiles_c = (C + 63) >> 6;
+  const long tile_id = blockIdx.x;
+  const long tr = tile_id / tiles_c;
+  const long tc = tile_id % tiles_c;
+  const long r0 = tr * 64, c0 = tc * 64;
+
+  float local_amax = 0.f;
+  // load+quantize: thread t handles row r0 + t/8 (+32 rows second half),
+  // 8 cols at (t%8)*8
+  const int lr = threadIdx.x >> 3;          // 0..31
+  const int lc = (threadIdx.x & 7) * 8;     // 0..56
+#pragma unroll
+  for (int half = 0; half < 2; ++half) {
+    long r = r0 + lr + half * 32;
+    if (r < R) {
+      const __hip_bfloat16* src = x + r * C + c0 + lc;
+      int valid = (int)min((long)8, C - (c0 + lc));
+      float f[8];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        f[e] = (e < valid) ? __bfloat162float(src[e]) : 0.f;
+        local_amax = fmaxf(local_amax, fabsf(f[e]));
+        f[e] = fminf(fmaxf(f[e] * qscale, -kE4M3Max), kE4M3Max);
+      }
+      int w0 = 0, w1 = 0;
+      w0 = __builtin_amdgcn_cvt_pk_fp8_f32(f[0], f[1], w0, false);
+      w0 = __builtin_amdgcn_cvt_pk_fp8_f32(f[2], f[3], w0, true);
+      w1 = __builtin_amdgcn_cvt_pk_fp8_f32(f[4], f[5], w1, false);
+      w1 = __builtin_amdgcn_cvt_pk_fp8_f32(f[6], f[7], w1, true);
+      // row-major output (8B when fully in-bounds)
+      uint8_t bytes[8];
+      *reinterpret_cast<int*>(bytes) = w0;
+      *reinterpret_cast<int*>(bytes + 4) = w1;
+      if (valid == 8) {
+        *reinterpret_cast<uint2*>(q + r * C + c0 + lc) =
+            make_uint2((unsigned)w0, (unsigned)w1);
+      } else {
+        for (int e = 0; e < valid; ++e) q[r * C + c0 + lc + e] = bytes[e];
+      }
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        tile[lr + half * 32][lc + e] = bytes[e];
+      }
+    } else {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) tile[lr + half * 32][lc + e] = 0;
+    }
+  }
+  __syncthreads();
+  // transposed store: thread t handles OUTPUT row (= input col) c0 + lr
+  // (+32 second half), 8 input rows at lc
+#pragma unroll
+  for (int half = 0; half < 2; ++half) {
+    long c = c0 + lr + half * 32;
+    if (c < C) {
+      uint8_t bytes[8];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) bytes[e] = tile[lc + e][lr + half * 32];
+      long base = c * R + r0 + lc;
+      int valid = (int)min((long)8, R - (r0 + lc));
+      if (valid == 8) {
+        *reinterpret_cast<uint2*>(qt + base) =
+            *reinterpret_cast<uint2*>(bytes);
+      } else {
+        for (int e = 0; e < valid; ++e) qt[base + e] = bytes[e];
+      }
+    }
+  }
+
+  __shared__ float lds[kBlock / WAVE_SIZE];
+  local_amax = wave_reduce_max(local_amax);
+  const int lane = threadIdx.x & (WAVE_SIZE - 1);
+  const int wave = threadIdx.x / WAVE_SIZE;
+  if (lane == 0) lds[wave] = local_amax;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float m = 0.f;
+    for (int w = 0; w < kBlock / WAVE_SIZE; ++w) m = fmaxf(m, lds[w]);
+    atomic_max_pos_float(amax, m);
+  }
+}
+
 }  // namespace
 
 std::vector<torch::Tensor> fp8_quantize(torch::Tensor x,
@@ -101,4 +198,25 @@ std::vector<torch::Tensor> fp8_quantize(torch::Tensor x,
                      scale.data_ptr<float>(), (uint8_t*)q.data_ptr(),
                      amax.data_ptr<float>(), n);
   return {q, amax};
+}
+
+std::vector<torch::Tensor> fp8_cast_transpose(torch::Tensor x,
+                                              torch::Tensor scale) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous()
+              && x.scalar_type() == torch::kBFloat16,
+              "fp8_cast_transpose: contiguous 2-D bf16 input required");
+  TORCH_CHECK(scale.is_cuda() && scale.numel() == 1
+              && scale.scalar_type() == torch::kFloat32);
+  long R = x.size(0), C = x.size(1);
+  auto q = torch::empty({R, C}, x.options().dtype(torch::kFloat8_e4m3fn));
+  auto qt = torch::empty({C, R}, x.options().dtype(torch::kFloat8_e4m3fn));
+  auto amax = torch::zeros({1}, x.options().dtype(torch::kFloat32));
+  auto stream = c10::hip::getCurrentHIPStream();
+  long tiles = ((R + 63) / 64) * ((C + 63) / 64);
+  hipLaunchKernelGGL(fp8_cast_transpose_kernel, dim3((unsigned)tiles),
+                     dim3(256), 0, stream,
+                     (const __hip_bfloat16*)x.data_ptr(),
+                     scale.data_ptr<float>(), (uint8_t*)q.data_ptr(),
+                     (uint8_t*)qt.data_ptr(), amax.data_ptr<float>(), R, C);
+  return {q, qt, amax};
 }

@@ -184,7 +184,19 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
             )
 
         if ctx.gradient_accumulation_fusion and hasattr(weight, "main_grad"):
-            _wgrad_accum_fp32(total_input_2d, grad_output_2d, weight.main_grad)
+            done = False
+            if ctx.fp8:
+                from ..fp8 import fp8_linear_wgrad, fp8_wgrad_enabled
+
+                if fp8_wgrad_enabled():
+                    meta = ctx.fp8_meta or {}
+                    done = fp8_linear_wgrad(
+                        total_input_2d, grad_output_2d, weight.main_grad,
+                        meta_x=meta.get("wx"), meta_dy=meta.get("wdy"),
+                    )
+            if not done:
+                _wgrad_accum_fp32(total_input_2d, grad_output_2d,
+                                  weight.main_grad)
             grad_weight = None
         else:
             grad_weight = grad_output_2d.t().matmul(total_input_2d)
@@ -224,6 +236,11 @@ def _linear_fp8_meta(module, ref_tensor):
         module._fp8_meta = {
             "x": Fp8TensorMeta(ref_tensor.device),
             "dy": Fp8TensorMeta(ref_tensor.device),
+            # wgrad operand roles (fp8_wgrad): same tensors, separate
+            # delayed-scaling state so the cast-transpose pass keeps its
+            # own history
+            "wx": Fp8TensorMeta(ref_tensor.device),
+            "wdy": Fp8TensorMeta(ref_tensor.device),
         }
     return module._fp8_meta
 
