@@ -48,11 +48,16 @@ def main():
             w8 = w.to(torch.float8_e4m3fn)
             wt8 = w.t().contiguous().to(torch.float8_e4m3fn)
             g8 = g.to(torch.float8_e4m3fn)
+            # wgrad orientation (fp8_wgrad): dy^T [N,M] @ x[M,K]-col, f32 out
+            gt8 = g.t().contiguous().to(torch.float8_e4m3fn)
+            xt8_w = x.t().contiguous().to(torch.float8_e4m3fn)
             for _ in range(3):
                 torch._scaled_mm(x8, w8.t(), scale_a=one, scale_b=one,
                                  out_dtype=torch.bfloat16)
                 torch._scaled_mm(g8, wt8.t(), scale_a=one, scale_b=one,
                                  out_dtype=torch.bfloat16)
+                torch._scaled_mm(gt8, xt8_w.t(), scale_a=one, scale_b=one,
+                                 out_dtype=torch.float32)
             torch.cuda.synchronize()
             print(f"tuned fp8 {m}x{k}x{n}", flush=True)
     print("done")
