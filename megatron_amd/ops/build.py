@@ -84,7 +84,7 @@ def build(verbose: bool = True, force: bool = False) -> str:
             + objs
             + [f"-L{p}" for p in lib_paths]
             + ["-ltorch", "-ltorch_cpu", "-ltorch_hip", "-lc10", "-lc10_hip",
-               "-ltorch_python", "-lhipblas", "-lhipblaslt", "-lamdhip64", "-o", OUT]
+               "-ltorch_python", "-lhipblas", "-lhipblaslt", "-lrocblas", "-lamdhip64", "-o", OUT]
         )
         if verbose:
             print("[ops.build] linking _C.so ...", flush=True)

@@ -19,6 +19,8 @@
 #include <hip/hip_runtime.h>
 #include <hipblas/hipblas.h>
 #include <hipblaslt/hipblaslt.h>
+#define ROCBLAS_BETA_FEATURES_API 1
+#include <rocblas/rocblas.h>
 
 #include <torch/extension.h>
 #include <ATen/ATen.h>
@@ -79,7 +81,86 @@ struct LtPlan {
   hipblasLtMatrixLayout_t a_lay, b_lay, c_lay;
   hipblasLtMatmulAlgo_t algo;
   bool valid = false;
+  // rocBLAS arm: when its best solution beats hipBLASLt's, run it instead
+  bool use_rocblas = false;
+  int rocblas_index = 0;
 };
+
+rocblas_handle get_rb_handle() {
+  static rocblas_handle h = nullptr;
+  if (h == nullptr) rocblas_create_handle(&h);
+  return h;
+}
+
+rocblas_datatype rb_dtype(hipDataType t) {
+  if (t == HIP_R_16BF) return rocblas_datatype_bf16_r;
+  if (t == HIP_R_16F) return rocblas_datatype_f16_r;
+  return rocblas_datatype_f32_r;
+}
+
+// benchmark rocBLAS's solution pool for the same (N,T) f32-accum GEMM;
+// returns best (index, ms) or ms=1e30 when none work
+std::pair<int, float> rocblas_sweep(long in_dim, long out_dim, long K,
+                                    hipDataType ab_type, const void* a,
+                                    const void* b, void* c_scratch,
+                                    hipStream_t stream) {
+  auto h = get_rb_handle();
+  rocblas_set_stream(h, stream);
+  float alpha = 1.0f, beta = 1.0f;
+  rocblas_datatype abt = rb_dtype(ab_type);
+  rocblas_int n_sol = 0;
+  auto q = rocblas_gemm_ex_get_solutions(
+      h, rocblas_operation_none, rocblas_operation_transpose, (int)in_dim,
+      (int)out_dim, (int)K, &alpha, a, abt, (int)in_dim, b, abt,
+      (int)out_dim, &beta, c_scratch, rocblas_datatype_f32_r, (int)in_dim,
+      c_scratch, rocblas_datatype_f32_r, (int)in_dim,
+      rocblas_datatype_f32_r, rocblas_gemm_algo_solution_index,
+      rocblas_gemm_flags_none, nullptr, &n_sol);
+  if (q != rocblas_status_success || n_sol <= 0) return {0, 1e30f};
+  if (n_sol > 64) n_sol = 64;
+  std::vector<rocblas_int> sols(n_sol);
+  rocblas_int got = n_sol;
+  q = rocblas_gemm_ex_get_solutions(
+      h, rocblas_operation_none, rocblas_operation_transpose, (int)in_dim,
+      (int)out_dim, (int)K, &alpha, a, abt, (int)in_dim, b, abt,
+      (int)out_dim, &beta, c_scratch, rocblas_datatype_f32_r, (int)in_dim,
+      c_scratch, rocblas_datatype_f32_r, (int)in_dim,
+      rocblas_datatype_f32_r, rocblas_gemm_algo_solution_index,
+      rocblas_gemm_flags_none, sols.data(), &got);
+  if (q != rocblas_status_success || got <= 0) return {0, 1e30f};
+
+  hipEvent_t ev0, ev1;
+  hipEventCreate(&ev0);
+  hipEventCreate(&ev1);
+  int best = 0;
+  float best_ms = 1e30f;
+  for (int i = 0; i < got; ++i) {
+    auto run = [&]() {
+      return rocblas_gemm_ex(
+          h, rocblas_operation_none, rocblas_operation_transpose,
+          (int)in_dim, (int)out_dim, (int)K, &alpha, a, abt, (int)in_dim, b,
+          abt, (int)out_dim, &beta, c_scratch, rocblas_datatype_f32_r,
+          (int)in_dim, c_scratch, rocblas_datatype_f32_r, (int)in_dim,
+          rocblas_datatype_f32_r, rocblas_gemm_algo_solution_index, sols[i],
+          rocblas_gemm_flags_none);
+    };
+    if (run() != rocblas_status_success) continue;
+    hipEventRecord(ev0, stream);
+    bool ok = true;
+    for (int it = 0; it < 3; ++it) ok = ok && run() == rocblas_status_success;
+    hipEventRecord(ev1, stream);
+    hipEventSynchronize(ev1);
+    float ms = 1e30f;
+    hipEventElapsedTime(&ms, ev0, ev1);
+    if (ok && ms < best_ms) {
+      best_ms = ms;
+      best = sols[i];
+    }
+  }
+  hipEventDestroy(ev0);
+  hipEventDestroy(ev1);
+  return {best, best_ms};
+}
 
 std::unordered_map<ShapeKey, LtPlan, ShapeKeyHash>& plan_cache() {
   static std::unordered_map<ShapeKey, LtPlan, ShapeKeyHash> cache;
@@ -164,11 +245,23 @@ LtPlan make_plan(long in_dim, long out_dim, long K, hipDataType ab_type,
     }
     hipEventDestroy(ev0);
     hipEventDestroy(ev1);
+    // rocBLAS arm: its pool wins several TN bf16 shapes on this chip
+    auto rb = rocblas_sweep(in_dim, out_dim, K, ab_type, a, b, c_scratch,
+                            stream);
+    if (rb.second < best_ms) {
+      p.use_rocblas = true;
+      p.rocblas_index = rb.first;
+    }
     if (getenv("MEGATRON_AMD_WGRAD_VERBOSE")) {
-      double tf = 2.0 * in_dim * out_dim * K / (best_ms / 3 * 1e-3) / 1e12;
+      double ms = (p.use_rocblas ? rb.second : best_ms) / 3;
+      double tf = 2.0 * in_dim * out_dim * K / (ms * 1e-3) / 1e12;
       fprintf(stderr,
-              "[wgrad tune] %ldx%ldxK%ld: algo %d/%d  %.3f ms  %.0f TF/s\n",
-              out_dim, in_dim, K, best, got, best_ms / 3, tf);
+              "[wgrad tune] %ldx%ldxK%ld: %s %d  %.3f ms  %.0f TF/s "
+              "(lt best %.3f ms, rb best %.3f ms)\n",
+              out_dim, in_dim, K,
+              p.use_rocblas ? "rocblas sol" : "lt algo",
+              p.use_rocblas ? p.rocblas_index : best, ms, best_ms / 3,
+              rb.second / 3);
     }
   }
   p.algo = results[best].algo;
@@ -231,7 +324,22 @@ void wgrad_gemm_accum_fp32(torch::Tensor input, torch::Tensor grad_output,
   }
 
   float alpha = 1.0f, beta = 1.0f;
-  if (plan.valid) {
+  if (plan.valid && plan.use_rocblas) {
+    auto h = get_rb_handle();
+    rocblas_set_stream(h, stream);
+    rocblas_datatype abt = rb_dtype(ab_type);
+    auto st = rocblas_gemm_ex(
+        h, rocblas_operation_none, rocblas_operation_transpose, (int)in_dim,
+        (int)out_dim, (int)K, &alpha, input.data_ptr(), abt, (int)in_dim,
+        grad_output.data_ptr(), abt, (int)out_dim, &beta,
+        main_grad.data_ptr(), rocblas_datatype_f32_r, (int)in_dim,
+        main_grad.data_ptr(), rocblas_datatype_f32_r, (int)in_dim,
+        rocblas_datatype_f32_r, rocblas_gemm_algo_solution_index,
+        plan.rocblas_index, rocblas_gemm_flags_none);
+    if (st == rocblas_status_success) return;
+    std::lock_guard<std::mutex> g(cache_mutex());
+    plan_cache()[key].valid = false;
+  } else if (plan.valid) {
     auto st = hipblasLtMatmul(
         get_lt_handle(), plan.desc, &alpha, input.data_ptr(), plan.a_lay,
         grad_output.data_ptr(), plan.b_lay, &beta, main_grad.data_ptr(),
