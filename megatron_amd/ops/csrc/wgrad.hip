@@ -116,7 +116,13 @@ std::pair<int, float> rocblas_sweep(long in_dim, long out_dim, long K,
       c_scratch, rocblas_datatype_f32_r, (int)in_dim,
       rocblas_datatype_f32_r, rocblas_gemm_algo_solution_index,
       rocblas_gemm_flags_none, nullptr, &n_sol);
-  if (q != rocblas_status_success || n_sol <= 0) return {0, 1e30f};
+  if (q != rocblas_status_success || n_sol <= 0) {
+    if (getenv("MEGATRON_AMD_WGRAD_VERBOSE")) {
+      fprintf(stderr, "[wgrad tune] rocblas get_solutions: status %d n %d\n",
+              (int)q, (int)n_sol);
+    }
+    return {0, 1e30f};
+  }
   if (n_sol > 64) n_sol = 64;
   std::vector<rocblas_int> sols(n_sol);
   rocblas_int got = n_sol;
@@ -260,7 +266,7 @@ LtPlan make_plan(long in_dim, long out_dim, long K, hipDataType ab_type,
               "(lt best %.3f ms, rb best %.3f ms)\n",
               out_dim, in_dim, K,
               p.use_rocblas ? "rocblas sol" : "lt algo",
-              p.use_rocblas ? p.rocblas_index : best, ms, best_ms / 3,
+              p.use_rocblas ? p.rocblas_index : best, ms, tf, best_ms / 3,
               rb.second / 3);
     }
   }
