@@ -25,7 +25,6 @@ sys.path.append(os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 from megatron_amd.checkpointing import (  # noqa: E402
     get_checkpoint_tracker_filename,
-    read_metadata,
 )
 
 
