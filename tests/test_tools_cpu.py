@@ -449,3 +449,76 @@ def test_stream_reshard_bounded_and_correct(tmp_path):
         assert set(got.keys()) == set(ref.keys())
         for k in ref:
             assert torch.equal(got[k], ref[k]), k
+
+
+def test_stream_reshard_tp2pp2_to_tp4pp1(tmp_path):
+    """Streaming reshard with BOTH source axes sharded (tp2 x pp2 ->
+    tp4 x pp1): equality vs the full-merge reference on every target."""
+    import argparse as ap
+    import sys
+    sys.path.insert(0, os.path.join(REPO, "tools"))
+    import checkpoint_util as cu
+
+    torch.manual_seed(9)
+    H, V = 32, 64
+    margs = ap.Namespace(
+        num_layers=4, hidden_size=H, num_attention_heads=8,
+        num_attention_heads_kv=4, ffn_hidden_size=3 * H,
+        padded_vocab_size=V, tensor_model_parallel_size=2,
+        pipeline_model_parallel_size=2, glu_activation="swiglu",
+    )
+
+    def layer_sd(prefix, tp):
+        # per-TP-shard shapes (dim0 keys halved, dim1 keys halved on dim1)
+        return {
+            f"{prefix}.input_layernorm.weight": torch.randn(H),
+            f"{prefix}.self_attention.query_key_value.weight":
+                torch.randn(H, H),
+            f"{prefix}.self_attention.dense.weight": torch.randn(H, H // 2),
+            f"{prefix}.mlp.dense_h_to_4h.weight": torch.randn(3 * H, H),
+            f"{prefix}.mlp.dense_4h_to_h.weight":
+                torch.randn(H, (3 * H) // 4),
+        }
+
+    shards = {}
+    for pp in range(2):
+        for tp in range(2):
+            sd = {}
+            if pp == 0:
+                sd["embedding.word_embeddings.weight"] = (
+                    torch.randn(V // 2, H)
+                )
+            for i in range(2):
+                sd.update(layer_sd(f"encoder.layers.{i}", tp))
+            if pp == 1:
+                sd["encoder.final_layernorm.weight"] = torch.randn(H)
+                sd["lm_head"] = torch.randn(V // 2, H)
+            shards[(tp, pp)] = sd
+
+    src = tmp_path / "src"
+    for (tp, pp), sd in shards.items():
+        d = src / "release" / f"mp_rank_{tp:02d}_{pp:03d}"
+        d.mkdir(parents=True)
+        torch.save({"args": margs, "checkpoint_version": 3.0, "iteration": 0,
+                    "model": {"language_model": sd}},
+                   d / "model_optim_rng.pt")
+    (src / "latest_checkpointed_iteration.txt").write_text("release")
+
+    out = tmp_path / "out"
+    cache = cu.stream_reshard(str(src), str(out), tp=4, pp=1, glu=True,
+                              progress=lambda *_: None)
+    assert cache.max_loaded == 2  # one stage of 2 TP shards resident
+
+    wrapped = {k: {"model": {"language_model": v}}
+               for k, v in shards.items()}
+    full = cu.merge_full_state(wrapped, 2, 2, 4, glu=True)
+    ref_split = cu.split_full_state(full, 4, 1, 4, glu=True)
+    for tpr in range(4):
+        got = torch.load(
+            out / "release" / f"mp_rank_{tpr:02d}" / "model_optim_rng.pt",
+            map_location="cpu", weights_only=False,
+        )["model"]["language_model"]
+        ref = ref_split[(tpr, 0)]
+        assert set(got.keys()) == set(ref.keys())
+        for k in ref:
+            assert torch.equal(got[k], ref[k]), k
