@@ -62,6 +62,8 @@ def parse_args():
                    choices=["bf16", "fp16", "fp8"])
     p.add_argument("--fp8-wgrad", action="store_true",
                    help="with --dtype fp8: run the wgrad GEMM in e4m3 too")
+    p.add_argument("--loss-chunk-size", type=int, default=None,
+                   help="override the chunked LM-head loss size (0 = off)")
     return p.parse_args()
 
 
@@ -188,7 +190,9 @@ def main():
         model_name=spec["model_name"],
         sliding_window_size=spec.get("sliding_window_size"),
         rope_scaling_factor=spec.get("rope_scaling_factor", 1.0),
-        loss_chunk_size=spec.get("loss_chunk_size", 0),
+        loss_chunk_size=(args.loss_chunk_size
+                         if args.loss_chunk_size is not None
+                         else spec.get("loss_chunk_size", 0)),
         rank=rank, world_size=world_size, local_rank=local_rank,
         **dtype_flags,
     )
