@@ -403,19 +403,41 @@ __global__ __launch_bounds__(NW * 64) void fa_fwd_kernel(
     __syncthreads();
   }
 
+  // epilogue: pack the 16 x D wave tile through the (now idle) P strip so
+  // the global stores are 16-byte dwordx4 — the scalar bf16 tail is
+  // store-issue-bound (32 instructions/lane)
+  float inv_l[4];
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     int row = qrow0 + kgroup * 4 + r;
-    float inv_l = l_run[r] > 0.f ? 1.0f / l_run[r] : 0.f;
-    if (row < Sq) {
+    inv_l[r] = l_run[r] > 0.f ? 1.0f / l_run[r] : 0.f;
+    if (row < Sq && row_in_tile == 0) {
+      lse[((long)b * Hq + h) * Sq + row] =
+          m_run[r] + logf(fmaxf(l_run[r], 1e-30f));
+    }
+  }
 #pragma unroll
-      for (int t = 0; t < DTILES; ++t) {
-        out[o_base + (long)row * os.ss + t * 16 + row_in_tile] =
-            __float2bfloat16(o_acc[t][r] * inv_l);
+  for (int half = 0; half < D / 64; ++half) {
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        p_lds[wave][(kgroup * 4 + r) * kStrip + t * 16 + row_in_tile] =
+            __float2bfloat16(o_acc[half * 4 + t][r] * inv_l[r]);
       }
-      if (row_in_tile == 0) {
-        lse[((long)b * Hq + h) * Sq + row] =
-            m_run[r] + logf(fmaxf(l_run[r], 1e-30f));
+    }
+    // wave-private strip: in-order LDS, no barrier
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      int e = j * 64 + lane;
+      int srow = e / 8;
+      int scol = (e % 8) * 8;
+      int row = qrow0 + srow;
+      if (row < Sq) {
+        *reinterpret_cast<uint4*>(out + o_base + (long)row * os.ss +
+                                  half * 64 + scol) =
+            *reinterpret_cast<const uint4*>(
+                &p_lds[wave][srow * kStrip + scol]);
       }
     }
   }
@@ -1149,18 +1171,42 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dkdv_kernel(
     __syncthreads();
   }
 
+  // epilogue: pack dk through pt_lds and dv through dst_lds (both idle by
+  // now) for dwordx4 stores instead of 2x32 scalar bf16 per lane
+  {
+    const long dk_base = (long)b * dks.bs + (long)hkv * dks.hs;
+    const long dv_base = (long)b * dvs.bs + (long)hkv * dvs.hs;
+    const int key0w = kb * BN + wave * 16;
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    int key = kb * BN + wave * 16 + kgroup * 4 + r;
-    if (key < Sk) {
-      const long dk_base = (long)b * dks.bs + (long)hkv * dks.hs;
-      const long dv_base = (long)b * dvs.bs + (long)hkv * dvs.hs;
+    for (int half = 0; half < D / 64; ++half) {
 #pragma unroll
-      for (int t = 0; t < DTILES; ++t) {
-        dk[dk_base + (long)key * dks.ss + t * 16 + row_in_tile] =
-            __float2bfloat16(dk_acc[t][r]);
-        dv[dv_base + (long)key * dvs.ss + t * 16 + row_in_tile] =
-            __float2bfloat16(dv_acc[t][r]);
+      for (int t = 0; t < 4; ++t) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int srow = kgroup * 4 + r;
+          int scol = t * 16 + row_in_tile;
+          pt_lds[wave][srow * kStrip + scol] =
+              __float2bfloat16(dk_acc[half * 4 + t][r]);
+          dst_lds[wave][srow * kStrip + scol] =
+              __float2bfloat16(dv_acc[half * 4 + t][r]);
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        int e = j * 64 + lane;
+        int srow = e / 8;
+        int scol = (e % 8) * 8;
+        int key = key0w + srow;
+        if (key < Sk) {
+          *reinterpret_cast<uint4*>(dk + dk_base + (long)key * dks.ss +
+                                    half * 64 + scol) =
+              *reinterpret_cast<const uint4*>(
+                  &pt_lds[wave][srow * kStrip + scol]);
+          *reinterpret_cast<uint4*>(dv + dv_base + (long)key * dvs.ss +
+                                    half * 64 + scol) =
+              *reinterpret_cast<const uint4*>(
+                  &dst_lds[wave][srow * kStrip + scol]);
+        }
       }
     }
   }
@@ -1337,14 +1383,31 @@ __global__ __launch_bounds__(NW * 64) void fa_bwd_dq_kernel(
     __syncthreads();
   }
 
+  // epilogue: pack dq through the idle dS strip for dwordx4 stores
+  {
+    const long dq_base = (long)b * dqs.bs + (long)h * dqs.hs;
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    int row = qrow0 + kgroup * 4 + r;
-    if (row < Sq) {
+    for (int half = 0; half < D / 64; ++half) {
 #pragma unroll
-      for (int t = 0; t < DTILES; ++t) {
-        dq[(long)b * dqs.bs + (long)h * dqs.hs + (long)row * dqs.ss +
-           t * 16 + row_in_tile] = __float2bfloat16(dq_acc[t][r]);
+      for (int t = 0; t < 4; ++t) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          ds_lds[wave][(kgroup * 4 + r) * kStrip + t * 16 + row_in_tile] =
+              __float2bfloat16(dq_acc[half * 4 + t][r]);
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        int e = j * 64 + lane;
+        int srow = e / 8;
+        int scol = (e % 8) * 8;
+        int row = qrow0 + srow;
+        if (row < Sq) {
+          *reinterpret_cast<uint4*>(dq + dq_base + (long)row * dqs.ss +
+                                    half * 64 + scol) =
+              *reinterpret_cast<const uint4*>(
+                  &ds_lds[wave][srow * kStrip + scol]);
+        }
       }
     }
   }
