@@ -120,7 +120,10 @@ def _communicate(tensor_send_next, tensor_send_prev, recv_prev, recv_next,
     if dtype_ is not None:
         dtype = dtype_
 
-    requires_grad = True
+    # split path: the raw chunk must NOT require grad (it is all-gathered
+    # in-place across the TP group first — autograd forbids that on a
+    # grad-tracking buffer); grad is re-attached on the gathered tensor
+    requires_grad = not split
     device = torch.cuda.current_device() if torch.cuda.is_available() else "cpu"
     if recv_prev:
         tensor_recv_prev = torch.empty(

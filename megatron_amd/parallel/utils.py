@@ -50,15 +50,22 @@ def split_tensor_into_1d_equal_chunks(tensor: torch.Tensor, new_buffer=False):
 
 
 def gather_split_1d_tensor(tensor: torch.Tensor) -> torch.Tensor:
-    """All-gather a tensor split with split_tensor_into_1d_equal_chunks."""
+    """All-gather a tensor split with split_tensor_into_1d_equal_chunks.
+
+    Runs under no_grad on a detached input: the collective writes the
+    output in place (and flattens via views), which autograd rejects in a
+    grad-enabled context — the TP>1 x PP>1 1F1B steady state calls this
+    with grad mode on (caught by test_tp2_pp2_train_step)."""
     numel_gathered = tensor.numel() * ps.get_tensor_model_parallel_world_size()
-    gathered = torch.empty(
-        numel_gathered, dtype=tensor.dtype,
-        device=tensor.device, requires_grad=False,
-    )
-    torch.distributed.all_gather_into_tensor(
-        gathered, tensor, group=ps.get_tensor_model_parallel_group()
-    )
+    with torch.no_grad():
+        gathered = torch.empty(
+            numel_gathered, dtype=tensor.dtype,
+            device=tensor.device, requires_grad=False,
+        )
+        torch.distributed.all_gather_into_tensor(
+            gathered, tensor.detach(),
+            group=ps.get_tensor_model_parallel_group(),
+        )
     return gathered
 
 

@@ -495,3 +495,108 @@ REPO_PIPE = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 def test_pp2_generation_matches_merged():
     mp.spawn(_worker_pp2_generation, args=(29681,), nprocs=WORLD, join=True)
+
+
+# ---------------------------------------------------------------------------
+# full 3D geometry: TP2 x PP2 (world 4) — the rank topology of the 8-GPU
+# 70B config (tp-inner, pp-outer), one 1F1B training step on gloo
+
+
+def _worker_tp2pp2(rank, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = "4"
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo", rank=rank, world_size=4)
+    try:
+        _body_tp2pp2(rank)
+    finally:
+        dist.barrier()
+        from megatron_amd import parallel as mpu
+
+        mpu.destroy_model_parallel()
+        dist.destroy_process_group()
+
+
+def _body_tp2pp2(rank):
+    import functools
+
+    from megatron_amd import parallel as mpu
+    from megatron_amd.config import TrainingConfig, set_config
+    from megatron_amd.microbatches import setup_microbatch_calculator
+    from megatron_amd.models import LlamaModel, ModelType
+    from megatron_amd.optim import get_megatron_optimizer
+    from megatron_amd.parallel.ddp import DistributedDataParallel as LocalDDP
+    from megatron_amd.parallel.schedules import (
+        forward_backward_pipelining_without_interleaving,
+    )
+    from megatron_amd.utils import get_ltor_masks_and_position_ids
+
+    mpu.initialize_model_parallel(2, 2)
+    mpu.model_parallel_cuda_manual_seed(1234)
+    torch.manual_seed(1234)
+    assert mpu.get_tensor_model_parallel_world_size() == 2
+    assert mpu.get_pipeline_model_parallel_world_size() == 2
+
+    cfg = TrainingConfig(
+        num_layers=4, hidden_size=64, num_attention_heads=4,
+        num_attention_heads_kv=2, seq_length=32, max_position_embeddings=64,
+        micro_batch_size=1, global_batch_size=2, hidden_dropout=0.0,
+        attention_dropout=0.0, use_cpu_initialization=True,
+        use_flash_attn=False, tensor_model_parallel_size=2,
+        pipeline_model_parallel_size=2, world_size=4, lr=1e-3, clip_grad=1.0,
+        no_async_tensor_model_parallel_allreduce=True,
+    )
+    cfg.finalize()
+    cfg.pad_vocab_size(96)
+    set_config(cfg)
+    setup_microbatch_calculator(cfg)
+
+    pre = mpu.is_pipeline_first_stage()
+    post = mpu.is_pipeline_last_stage()
+    m = LlamaModel(cfg, pre_process=pre, post_process=post)
+    m.model_type = ModelType.encoder_or_decoder
+    ddp = LocalDDP(m, True, True)
+    opt = get_megatron_optimizer([ddp], cfg)
+
+    tokens = torch.randint(0, 90, (1, 33))
+    torch.distributed.broadcast(tokens, 0)
+
+    def fwd_step(it, model):
+        inp = tokens[:, :-1].contiguous()
+        labels = tokens[:, 1:].contiguous()
+        am, _, pids = get_ltor_masks_and_position_ids(inp, 0, False, False,
+                                                      False)
+        out = model(inp, pids, am, labels=labels)
+
+        def loss_fn(o):
+            loss = o.float().mean()
+            return loss, {"lm loss": loss.detach()}
+
+        return out, loss_fn
+
+    for step in range(2):
+        ddp.zero_grad_buffer()
+        opt.zero_grad()
+        store = forward_backward_pipelining_without_interleaving(
+            fwd_step, None, [ddp], opt, cfg, None, False,
+        )
+        opt.reduce_model_grads()
+        ok, grad_norm, _ = opt.step()
+        assert ok
+        if post:
+            loss = store[0]["lm loss"].item()
+            assert loss == loss and abs(loss) < 1e4
+            # loss identical on both TP ranks of the last stage
+            t = torch.tensor([loss])
+            torch.distributed.broadcast(
+                t, mpu.get_tensor_model_parallel_src_rank(),
+                group=mpu.get_tensor_model_parallel_group(),
+            )
+            assert abs(t.item() - loss) < 1e-5
+
+
+def test_tp2_pp2_train_step():
+    mp.spawn(_worker_tp2pp2, args=(29641,), nprocs=4, join=True)
