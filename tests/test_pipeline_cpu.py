@@ -502,7 +502,7 @@ def test_pp2_generation_matches_merged():
 # 70B config (tp-inner, pp-outer), one 1F1B training step on gloo
 
 
-def _worker_tp2pp2(rank, port):
+def _worker_tp2pp2(rank, port, sp=False):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     os.environ["RANK"] = str(rank)
@@ -511,7 +511,7 @@ def _worker_tp2pp2(rank, port):
 
     dist.init_process_group("gloo", rank=rank, world_size=4)
     try:
-        _body_tp2pp2(rank)
+        _body_tp2pp2(rank, sp)
     finally:
         dist.barrier()
         from megatron_amd import parallel as mpu
@@ -520,7 +520,7 @@ def _worker_tp2pp2(rank, port):
         dist.destroy_process_group()
 
 
-def _body_tp2pp2(rank):
+def _body_tp2pp2(rank, sp=False):
     import functools
 
     from megatron_amd import parallel as mpu
@@ -548,6 +548,7 @@ def _body_tp2pp2(rank):
         use_flash_attn=False, tensor_model_parallel_size=2,
         pipeline_model_parallel_size=2, world_size=4, lr=1e-3, clip_grad=1.0,
         no_async_tensor_model_parallel_allreduce=True,
+        sequence_parallel=sp,
     )
     cfg.finalize()
     cfg.pad_vocab_size(96)
@@ -600,3 +601,9 @@ def _body_tp2pp2(rank):
 
 def test_tp2_pp2_train_step():
     mp.spawn(_worker_tp2pp2, args=(29641,), nprocs=4, join=True)
+
+
+def test_tp2_pp2_sp_train_step():
+    # the bench/70B path: TP2 x PP2 WITH sequence parallelism (seq-split
+    # p2p tensors, LN-grad TP all-reduce in reduce_model_grads)
+    mp.spawn(_worker_tp2pp2, args=(29642, True), nprocs=4, join=True)
