@@ -10,6 +10,7 @@ import os
 
 import pytest
 import torch
+import functools
 import torch.multiprocessing as mp
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
@@ -803,3 +804,100 @@ def _body_async_allreduce_backward_parity(rank):
 
 def test_async_allreduce_backward_parity():
     _spawn("_body_async_allreduce_backward_parity", 29613)
+
+
+# ---------------------------------------------------------------------------
+# TP2 x DP2 (world 4): the grad path must both TP-replicate and DP-average;
+# run with the plain mixed-precision optimizer AND the ZeRO-1 distributed
+# optimizer (combinations the 2-rank tests cannot reach)
+
+
+def _worker4(rank, fn_name, port, args=()):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = "4"
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo", rank=rank, world_size=4)
+    fn = globals()[fn_name]
+    try:
+        fn(rank, *args)
+    finally:
+        dist.barrier()
+        from megatron_amd import parallel as mpu
+
+        mpu.destroy_model_parallel()
+        dist.destroy_process_group()
+
+
+def _body_tp2_dp2(rank, use_dist_opt):
+    from megatron_amd import parallel as mpu
+    from megatron_amd.config import TrainingConfig, set_config
+    from megatron_amd.models import LlamaModel, ModelType
+    from megatron_amd.optim import get_megatron_optimizer
+    from megatron_amd.parallel.ddp import DistributedDataParallel as LocalDDP
+    from megatron_amd.utils import get_ltor_masks_and_position_ids
+
+    mpu.initialize_model_parallel(2, 1)
+    mpu.model_parallel_cuda_manual_seed(1234)
+    torch.manual_seed(1234)
+    assert mpu.get_data_parallel_world_size() == 2
+
+    cfg = TrainingConfig(
+        num_layers=2, hidden_size=64, num_attention_heads=4,
+        num_attention_heads_kv=2, seq_length=16, max_position_embeddings=32,
+        micro_batch_size=1, hidden_dropout=0.0, attention_dropout=0.0,
+        use_cpu_initialization=True, use_flash_attn=False,
+        tensor_model_parallel_size=2, world_size=4, lr=1e-3, clip_grad=1.0,
+        use_distributed_optimizer=use_dist_opt,
+        no_async_tensor_model_parallel_allreduce=True,
+    )
+    cfg.finalize()
+    cfg.pad_vocab_size(96)
+    set_config(cfg)
+
+    m = LlamaModel(cfg)
+    m.model_type = ModelType.encoder_or_decoder
+    ddp = LocalDDP(m, True, True)
+    ddp.broadcast_params()
+    opt = get_megatron_optimizer([ddp], cfg)
+
+    # each DP rank gets different data; TP pair shares it
+    tokens_all = torch.randint(0, 90, (2, 17))
+    torch.distributed.broadcast(tokens_all, 0)
+    dp_rank = mpu.get_data_parallel_rank()
+    inp = tokens_all[dp_rank:dp_rank + 1, :-1].contiguous()
+    labels = tokens_all[dp_rank:dp_rank + 1, 1:].contiguous()
+    am, _, pids = get_ltor_masks_and_position_ids(inp, 0, False, False, False)
+
+    for _ in range(2):
+        ddp.zero_grad_buffer()
+        opt.zero_grad()
+        out = ddp(inp, pids, am, labels=labels)
+        loss = out.float().mean()
+        loss.backward()
+        opt.reduce_model_grads()
+        ok, _, _ = opt.step()
+        assert ok
+        if hasattr(opt, "gather_model_params"):
+            opt.gather_model_params()
+
+    # updated params must be identical across the DP group (same TP rank)
+    for n, p in m.named_parameters():
+        ref = p.data.clone()
+        torch.distributed.broadcast(
+            ref, mpu.get_data_parallel_src_rank(),
+            group=mpu.get_data_parallel_group(),
+        )
+        assert torch.allclose(p.data, ref, atol=1e-6), n
+
+
+def test_tp2_dp2_plain_optimizer():
+    mp.spawn(functools.partial(_worker4, args=(False,)),
+             args=("_body_tp2_dp2", 29721), nprocs=4, join=True)
+
+
+def test_tp2_dp2_zero1_optimizer():
+    mp.spawn(functools.partial(_worker4, args=(True,)),
+             args=("_body_tp2_dp2", 29722), nprocs=4, join=True)
