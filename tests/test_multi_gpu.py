@@ -10,6 +10,7 @@ Reference parity: megatron/mpu/tests/test_layers.py + tests/tensor_parallel/*
 """
 
 import os
+import sys
 
 import pytest
 import torch
@@ -335,3 +336,177 @@ def _body_zero1_step(rank):
 
 def test_zero1_step():
     _spawn("_body_zero1_step", 29705)
+
+
+# ---------------------------------------------------------------------------
+# world-4 combinations (need >= 4 GPUs): the full-3D paths that caught the
+# scatter-gather p2p grad bug on gloo, on real RCCL
+
+WORLD4 = 4
+
+_skip4 = pytest.mark.skipif(
+    not _FORCE_CPU and (
+        not torch.cuda.is_available() or torch.cuda.device_count() < WORLD4
+    ),
+    reason="needs >= 4 GPUs",
+)
+
+
+def _worker4(rank, fn_name, port, args):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(WORLD4)
+    os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    import torch.distributed as dist
+
+    if _FORCE_CPU:
+        dist.init_process_group("gloo", rank=rank, world_size=WORLD4)
+    else:
+        torch.cuda.set_device(rank)
+        dist.init_process_group("nccl", rank=rank, world_size=WORLD4)
+    fn = globals()[fn_name]
+    try:
+        fn(rank, *args)
+    finally:
+        dist.barrier()
+        from megatron_amd import parallel as mpu
+
+        mpu.destroy_model_parallel()
+        dist.destroy_process_group()
+
+
+def _body4_tp2pp2(rank, sp):
+    """TP2 x PP2 1F1B training steps over RCCL (the 70B rank topology);
+    sp=True adds sequence parallelism (the bench path)."""
+    from megatron_amd import parallel as mpu
+    from megatron_amd.config import set_config
+    from megatron_amd.microbatches import setup_microbatch_calculator
+    from megatron_amd.models import LlamaModel, ModelType
+    from megatron_amd.optim import get_megatron_optimizer
+    from megatron_amd.parallel.ddp import DistributedDataParallel as LocalDDP
+    from megatron_amd.parallel.schedules import (
+        forward_backward_pipelining_without_interleaving,
+    )
+    from megatron_amd.utils import get_ltor_masks_and_position_ids
+
+    mpu.initialize_model_parallel(2, 2)
+    mpu.model_parallel_cuda_manual_seed(1234)
+    torch.manual_seed(1234)
+    cfg = _mk_cfg(num_layers=4, tensor_model_parallel_size=2,
+                  pipeline_model_parallel_size=2, world_size=4,
+                  global_batch_size=2, sequence_parallel=sp,
+                  no_async_tensor_model_parallel_allreduce=not sp,
+                  clip_grad=1.0)
+    setup_microbatch_calculator(cfg)
+
+    pre = mpu.is_pipeline_first_stage()
+    post = mpu.is_pipeline_last_stage()
+    m = LlamaModel(cfg, pre_process=pre, post_process=post)
+    m = m.to(_dev()).bfloat16()
+    m.model_type = ModelType.encoder_or_decoder
+    ddp = LocalDDP(m, True, True)
+    opt = get_megatron_optimizer([ddp], cfg)
+
+    tokens = torch.randint(0, 500, (1, cfg.seq_length + 1), device=_dev())
+    torch.distributed.broadcast(tokens, 0)
+
+    def fwd_step(it, model):
+        inp = tokens[:, :-1].contiguous()
+        labels = tokens[:, 1:].contiguous()
+        am, _, pids = get_ltor_masks_and_position_ids(inp, 0, False, False,
+                                                      False)
+        out = model(inp, pids, am, labels=labels)
+
+        def loss_fn(o):
+            loss = o.float().mean()
+            return loss, {"lm loss": loss.detach()}
+
+        return out, loss_fn
+
+    for _ in range(2):
+        ddp.zero_grad_buffer()
+        opt.zero_grad()
+        store = forward_backward_pipelining_without_interleaving(
+            fwd_step, None, [ddp], opt, cfg, None, False,
+        )
+        opt.reduce_model_grads()
+        ok, _, _ = opt.step()
+        assert ok
+        if post:
+            loss = store[0]["lm loss"].item()
+            assert loss == loss and abs(loss) < 1e4
+            # loss replicated across the last stage's TP pair
+            t = torch.tensor([loss], device=_dev())
+            torch.distributed.broadcast(
+                t, mpu.get_tensor_model_parallel_src_rank(),
+                group=mpu.get_tensor_model_parallel_group(),
+            )
+            assert abs(t.item() - loss) < 1e-3
+
+
+def _body4_tp2dp2_zero1(rank):
+    """TP2 x DP2 with the ZeRO-1 distributed optimizer over RCCL: params
+    stay DP-replicated after the sharded update + all-gather."""
+    from megatron_amd import parallel as mpu
+    from megatron_amd.config import set_config
+    from megatron_amd.models import LlamaModel, ModelType
+    from megatron_amd.optim import get_megatron_optimizer
+    from megatron_amd.parallel.ddp import DistributedDataParallel as LocalDDP
+    from megatron_amd.utils import get_ltor_masks_and_position_ids
+
+    mpu.initialize_model_parallel(2, 1)
+    mpu.model_parallel_cuda_manual_seed(1234)
+    torch.manual_seed(1234)
+    assert mpu.get_data_parallel_world_size() == 2
+    cfg = _mk_cfg(tensor_model_parallel_size=2, world_size=4,
+                  use_distributed_optimizer=True, clip_grad=1.0)
+
+    m = LlamaModel(cfg).to(_dev()).bfloat16()
+    m.model_type = ModelType.encoder_or_decoder
+    ddp = LocalDDP(m, True, True)
+    ddp.broadcast_params()
+    opt = get_megatron_optimizer([ddp], cfg)
+
+    tokens_all = torch.randint(0, 500, (2, cfg.seq_length + 1), device=_dev())
+    torch.distributed.broadcast(tokens_all, 0)
+    dp_rank = mpu.get_data_parallel_rank()
+    inp = tokens_all[dp_rank:dp_rank + 1, :-1].contiguous()
+    labels = tokens_all[dp_rank:dp_rank + 1, 1:].contiguous()
+    am, _, pids = get_ltor_masks_and_position_ids(inp, 0, False, False, False)
+
+    for _ in range(2):
+        ddp.zero_grad_buffer()
+        opt.zero_grad()
+        out = ddp(inp, pids, am, labels=labels)
+        out.float().mean().backward()
+        opt.reduce_model_grads()
+        ok, _, _ = opt.step()
+        assert ok
+        opt.gather_model_params()
+
+    for n, p in m.named_parameters():
+        ref = p.data.clone()
+        torch.distributed.broadcast(
+            ref, mpu.get_data_parallel_src_rank(),
+            group=mpu.get_data_parallel_group(),
+        )
+        assert torch.allclose(p.data.float(), ref.float(), atol=1e-5), n
+
+
+@_skip4
+def test_tp2_pp2_rccl():
+    mp.spawn(_worker4, args=("_body4_tp2pp2", 29731, (False,)),
+             nprocs=WORLD4, join=True)
+
+
+@_skip4
+def test_tp2_pp2_sp_rccl():
+    mp.spawn(_worker4, args=("_body4_tp2pp2", 29732, (True,)),
+             nprocs=WORLD4, join=True)
+
+
+@_skip4
+def test_tp2_dp2_zero1_rccl():
+    mp.spawn(_worker4, args=("_body4_tp2dp2_zero1", 29733, ()),
+             nprocs=WORLD4, join=True)
