@@ -607,3 +607,97 @@ def test_tp2_pp2_sp_train_step():
     # the bench/70B path: TP2 x PP2 WITH sequence parallelism (seq-split
     # p2p tensors, LN-grad TP all-reduce in reduce_model_grads)
     mp.spawn(_worker_tp2pp2, args=(29642, True), nprocs=4, join=True)
+
+
+# ---------------------------------------------------------------------------
+# TP2 x PP4-interleaved (world 8): tensor parallelism combined with the
+# virtual-pipeline schedule — the deepest 3D combination expressible on
+# gloo (interleaving requires pp > 2, matching the reference)
+
+
+def _worker_tp2_ivl(rank, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = "8"
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo", rank=rank, world_size=8)
+    try:
+        _body_tp2_ivl(rank)
+    finally:
+        dist.barrier()
+        from megatron_amd import parallel as mpu
+
+        mpu.destroy_model_parallel()
+        dist.destroy_process_group()
+
+
+def _body_tp2_ivl(rank):
+    import functools
+
+    from megatron_amd import global_state
+    from megatron_amd import parallel as mpu
+    from megatron_amd.config import TrainingConfig, set_config
+    from megatron_amd.microbatches import setup_microbatch_calculator
+    from megatron_amd.models import LlamaModel, ModelType
+    from megatron_amd.optim import (
+        get_megatron_optimizer, get_optimizer_param_scheduler,
+    )
+    from megatron_amd.parallel.ddp import DistributedDataParallel as LocalDDP
+    from megatron_amd.training import train_step
+    from megatron_amd.utils import get_ltor_masks_and_position_ids
+
+    global_state.init_timers()
+    mpu.initialize_model_parallel(2, 4, virtual_pipeline_model_parallel_size=2)
+    mpu.model_parallel_cuda_manual_seed(1234)
+    cfg = TrainingConfig(
+        num_layers=8, hidden_size=64, num_attention_heads=4,
+        num_attention_heads_kv=2, seq_length=32, max_position_embeddings=64,
+        micro_batch_size=1, global_batch_size=4,
+        tensor_model_parallel_size=2, pipeline_model_parallel_size=4,
+        world_size=8, num_layers_per_virtual_pipeline_stage=1,
+        hidden_dropout=0.0, attention_dropout=0.0,
+        use_cpu_initialization=True, lr=1e-3, train_iters=4, clip_grad=1.0,
+        no_async_tensor_model_parallel_allreduce=True,
+    )
+    cfg.finalize()
+    cfg.pad_vocab_size(128)
+    set_config(cfg)
+    setup_microbatch_calculator(cfg)
+    models = []
+    for i in range(2):
+        mpu.set_virtual_pipeline_model_parallel_rank(i)
+        m = LlamaModel(cfg, pre_process=mpu.is_pipeline_first_stage(),
+                       post_process=mpu.is_pipeline_last_stage())
+        m.model_type = ModelType.encoder_or_decoder
+        models.append(LocalDDP(m, True, True))
+    opt = get_megatron_optimizer(models, cfg)
+    sched = get_optimizer_param_scheduler(opt, cfg)
+    torch.manual_seed(55)
+    batches = [torch.randint(0, 128, (1, 33)) for _ in range(32)]
+    its = [iter(batches[:16]), iter(batches[16:])]
+
+    def fsf(it, model):
+        data = next(it)
+        tokens = data[:, :-1].contiguous()
+        labels = data[:, 1:].contiguous()
+        am, lm, pids = get_ltor_masks_and_position_ids(tokens, 0, False,
+                                                       False, False)
+        out = model(tokens, pids, am, labels=labels)
+
+        def loss_fn(lm, o):
+            loss = (o.float().view(-1) * lm.view(-1).float()).sum() / lm.sum()
+            return loss, {"lm loss": loss.detach()}
+
+        return out, functools.partial(loss_fn, lm)
+
+    for _ in range(2):
+        ld, skipped, gn, _ = train_step(fsf, its, models, opt, sched, cfg)
+        assert skipped == 0
+    if mpu.is_pipeline_last_stage(ignore_virtual=True):
+        assert ld["lm loss"].item() > 0
+
+
+def test_tp2_pp4_interleaved_trains():
+    mp.spawn(_worker_tp2_ivl, args=(29645,), nprocs=8, join=True)
