@@ -535,12 +535,13 @@ class TestGemv:
 class TestDecodeAttn:
     @pytest.mark.parametrize("gqa", [1, 4])
     @pytest.mark.parametrize("h", [128, 64])
-    def test_matches_eager(self, gqa, h):
-        """Fused decode attention vs the eager fp32 cache-softmax chain."""
+    @pytest.mark.parametrize("L,pos_val", [(96, 57), (2048, 1791)])
+    def test_matches_eager(self, gqa, h, L, pos_val):
+        """Fused decode attention vs the eager fp32 cache-softmax chain
+        (L >= 1024 exercises the flash-decoding split path)."""
         ext = _ext()
-        b, n, L = 2, 8, 96
+        b, n = 2, 8
         nkv = n // gqa
-        pos_val = 57
         q = torch.randn(b, n, h, device="cuda", dtype=torch.bfloat16)
         kc = torch.randn(L, b, nkv, h, device="cuda", dtype=torch.bfloat16)
         vc = torch.randn(L, b, nkv, h, device="cuda", dtype=torch.bfloat16)
@@ -593,12 +594,12 @@ class TestDecodeRopeAppend:
 
 
 class TestDecodeAttnWindow:
-    def test_sliding_window(self):
+    @pytest.mark.parametrize("L,pos_val,w", [(64, 40, 8), (4096, 3500, 512)])
+    def test_sliding_window(self, L, pos_val, w):
         """decode_attn with window w attends only the last w cache rows
-        (incl. current) — HF/Mistral convention."""
+        (incl. current) — HF/Mistral convention; long-L = split path."""
         ext = _ext()
-        b, n, h, L, w = 1, 2, 128, 64, 8
-        pos_val = 40
+        b, n, h = 1, 2, 128
         q = torch.randn(b, n, h, device="cuda", dtype=torch.bfloat16)
         kc = torch.randn(L, b, n, h, device="cuda", dtype=torch.bfloat16)
         vc = torch.randn(L, b, n, h, device="cuda", dtype=torch.bfloat16)
